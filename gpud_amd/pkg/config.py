@@ -1,0 +1,107 @@
+"""Daemon configuration (reference: pkg/config/config.go:18-98).
+
+A dataclass covering the reference Config surface: listen address, data dir,
+state file, retention periods, compact period, pprof, auto-update, command
+overrides, NFS group configs, plugin specs file, component enable/disable
+lists — plus AMD-specific knobs (expected GPU count, RAS thresholds).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import os
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+import yaml
+
+DEFAULT_ADDRESS = "localhost:15132"  # reference: cmd/gpud/main.go:15
+DEFAULT_DATA_DIR = "/var/lib/gpud"  # reference: pkg/config/default.go:18
+DEFAULT_RETENTION_DAYS = 3
+DEFAULT_COMPACT_PERIOD_HOURS = 24
+
+
+@dataclass
+class Config:
+    address: str = DEFAULT_ADDRESS
+    data_dir: str = DEFAULT_DATA_DIR
+    state_file: str = ""  # default: <data_dir>/gpud.state
+    events_retention_days: float = DEFAULT_RETENTION_DAYS
+    metrics_retention_days: float = DEFAULT_RETENTION_DAYS
+    compact_period_hours: float = DEFAULT_COMPACT_PERIOD_HOURS
+    pprof: bool = False
+    enable_auto_update: bool = False
+    auto_update_exit_code: int = -1
+    poll_interval_seconds: float = 60.0
+    # component selection (reference: component enable/disable list)
+    enabled_components: List[str] = field(default_factory=list)
+    disabled_components: List[str] = field(default_factory=list)
+    # nsenter-style command overrides (test seams)
+    reboot_command: str = ""
+    findmnt_command: str = ""
+    lsblk_command: str = ""
+    df_command: str = ""
+    lspci_command: str = ""
+    containerd_address: str = "/run/containerd/containerd.sock"
+    docker_socket: str = "/var/run/docker.sock"
+    nfs_host_root: str = ""
+    plugin_specs_file: str = ""
+    kernel_modules_to_check: List[str] = field(default_factory=list)
+    libraries_to_check: Dict[str, List[str]] = field(default_factory=dict)
+    mount_points: List[str] = field(default_factory=lambda: ["/"])
+    # accelerator knobs
+    expected_gpu_count: int = 0
+    expected_xgmi_link_count: int = 0  # 7 per GPU on an 8-GPU MI355X node
+    temperature_margin_threshold_c: float = 10.0
+    zombie_degraded_threshold: int = 1000
+    zombie_unhealthy_threshold: int = 2000
+    # control plane
+    endpoint: str = ""
+    token: str = ""
+    machine_id: str = ""
+    session_protocol: str = "v1"
+
+    @property
+    def state_path(self) -> str:
+        return self.state_file or os.path.join(self.data_dir, "gpud.state")
+
+    @property
+    def fifo_path(self) -> str:
+        return os.path.join(self.data_dir, "gpud.fifo")
+
+    @property
+    def packages_dir(self) -> str:
+        return os.path.join(self.data_dir, "packages")
+
+    @property
+    def target_version_path(self) -> str:
+        return os.path.join(self.data_dir, "target_version")
+
+    def component_enabled(self, name: str, tags: Optional[List[str]] = None) -> bool:
+        if self.disabled_components and name in self.disabled_components:
+            return False
+        if self.enabled_components:
+            return name in self.enabled_components
+        return True
+
+    def to_dict(self) -> Dict[str, Any]:
+        return dataclasses.asdict(self)
+
+    @staticmethod
+    def from_dict(d: Dict[str, Any]) -> "Config":
+        known = {f.name for f in dataclasses.fields(Config)}
+        return Config(**{k: v for k, v in d.items() if k in known})
+
+    @staticmethod
+    def load(path: str) -> "Config":
+        with open(path) as f:
+            return Config.from_dict(yaml.safe_load(f) or {})
+
+    def save(self, path: str) -> None:
+        os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+        with open(path, "w") as f:
+            yaml.safe_dump(self.to_dict(), f, sort_keys=True)
+
+
+def default_config(data_dir: str = DEFAULT_DATA_DIR) -> Config:
+    return Config(data_dir=data_dir)
