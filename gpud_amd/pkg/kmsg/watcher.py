@@ -1,0 +1,122 @@
+"""/dev/kmsg watcher: one-shot ring read + live follow.
+
+Reference: pkg/kmsg/watcher.go — ``ReadAll`` (seek-to-start one-shot,
+watcher.go:86-188) and ``Watch`` (follow channel, watcher.go:223-290). The
+kernel delivers one record per read(2); EPIPE means the ring overwrote our
+position (continue), EINVAL on a too-small buffer is impossible at 8 KiB
+(kernel caps records well below that).
+
+The watcher is shared: the RAS/error components register match callbacks
+on ONE watcher instance so the daemon reads /dev/kmsg once, not once per
+component (shared data sources keep overhead flat).
+"""
+
+from __future__ import annotations
+
+import errno
+import os
+import threading
+from typing import Callable, List, Optional
+
+from ..log import logger
+from .parser import Message, boot_wall_time, parse_line
+
+KMSG_PATH = "/dev/kmsg"
+_READ_SIZE = 8192
+
+
+class Watcher:
+    def __init__(self, path: str = KMSG_PATH):
+        self.path = path
+        self._follow_fd: Optional[int] = None
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._callbacks: List[Callable[[Message], None]] = []
+        self._cb_lock = threading.Lock()
+
+    # -- one-shot ring read (reference watcher.go ReadAll) ------------------
+
+    def read_all(self, limit: int = 100_000) -> List[Message]:
+        boot = boot_wall_time()
+        out: List[Message] = []
+        try:
+            fd = os.open(self.path, os.O_RDONLY | os.O_NONBLOCK)
+        except OSError as e:
+            logger.warning("cannot open %s: %s", self.path, e)
+            return out
+        try:
+            while len(out) < limit:
+                try:
+                    data = os.read(fd, _READ_SIZE)
+                except OSError as e:
+                    if e.errno == errno.EPIPE:
+                        continue  # ring overwrote our position; keep reading
+                    if e.errno == errno.EAGAIN:
+                        break  # drained
+                    raise
+                if not data:
+                    break
+                m = parse_line(data.decode("utf-8", "replace"), boot)
+                if m is not None:
+                    out.append(m)
+        finally:
+            os.close(fd)
+        return out
+
+    # -- live follow (reference watcher.go Watch/readFollow) ----------------
+
+    def register(self, cb: Callable[[Message], None]) -> None:
+        with self._cb_lock:
+            self._callbacks.append(cb)
+
+    def start(self, from_start: bool = False) -> None:
+        if self._thread is not None:
+            return
+        try:
+            # blocking fd for the follow loop
+            self._follow_fd = os.open(self.path, os.O_RDONLY)
+            if not from_start:
+                os.lseek(self._follow_fd, 0, os.SEEK_END)
+        except OSError as e:
+            logger.warning("kmsg follow unavailable (%s): %s", self.path, e)
+            return
+        self._thread = threading.Thread(
+            target=self._follow_loop, daemon=True, name="gpud-kmsg-watch"
+        )
+        self._thread.start()
+
+    def _follow_loop(self) -> None:
+        boot = boot_wall_time()
+        fd = self._follow_fd
+        assert fd is not None
+        while not self._stop.is_set():
+            try:
+                data = os.read(fd, _READ_SIZE)
+            except OSError as e:
+                if e.errno == errno.EPIPE:
+                    continue
+                if e.errno in (errno.EBADF, errno.EINVAL):
+                    return  # closed
+                logger.warning("kmsg read error: %s", e)
+                return
+            if not data:
+                continue
+            m = parse_line(data.decode("utf-8", "replace"), boot)
+            if m is None:
+                continue
+            with self._cb_lock:
+                cbs = list(self._callbacks)
+            for cb in cbs:
+                try:
+                    cb(m)
+                except Exception:
+                    logger.exception("kmsg callback failed")
+
+    def close(self) -> None:
+        self._stop.set()
+        fd, self._follow_fd = self._follow_fd, None
+        if fd is not None:
+            try:
+                os.close(fd)
+            except OSError:
+                pass

@@ -1,0 +1,103 @@
+"""Bash-script process runner (reference: pkg/process/process.go:21,
+runner.go:14-20).
+
+Backend for custom plugins, bootstrap scripts, diagnostics and the reboot
+method: run a bash script to completion with a timeout, capturing combined
+output; or stream a long-running process.
+"""
+
+from __future__ import annotations
+
+import os
+import signal
+import subprocess
+import tempfile
+import threading
+from dataclasses import dataclass
+from typing import Optional
+
+
+@dataclass
+class RunResult:
+    exit_code: int
+    output: str
+    timed_out: bool = False
+    error: str = ""
+
+
+class Runner:
+    """Serializes script runs like the reference's exclusive Runner."""
+
+    def __init__(self) -> None:
+        self._lock = threading.Lock()
+
+    def run_until_completion(
+        self,
+        bash_script: str,
+        timeout_seconds: float = 60.0,
+        max_output_bytes: int = 256 * 1024,
+        env: Optional[dict] = None,
+    ) -> RunResult:
+        acquired = self._lock.acquire(timeout=timeout_seconds)
+        if not acquired:
+            return RunResult(
+                exit_code=-1, output="", error="another script is running"
+            )
+        try:
+            return run_bash(
+                bash_script,
+                timeout_seconds=timeout_seconds,
+                max_output_bytes=max_output_bytes,
+                env=env,
+            )
+        finally:
+            self._lock.release()
+
+
+def run_bash(
+    bash_script: str,
+    timeout_seconds: float = 60.0,
+    max_output_bytes: int = 256 * 1024,
+    env: Optional[dict] = None,
+) -> RunResult:
+    """Run a bash script in its own process group; kill the group on timeout."""
+    with tempfile.NamedTemporaryFile(
+        "w", suffix=".sh", prefix="gpud-", delete=False
+    ) as f:
+        f.write(bash_script)
+        path = f.name
+    merged_env = dict(os.environ)
+    if env:
+        merged_env.update(env)
+    try:
+        proc = subprocess.Popen(
+            ["bash", path],
+            stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT,
+            start_new_session=True,
+            env=merged_env,
+        )
+        try:
+            out, _ = proc.communicate(timeout=timeout_seconds)
+            timed_out = False
+        except subprocess.TimeoutExpired:
+            try:
+                os.killpg(proc.pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+            out, _ = proc.communicate()
+            timed_out = True
+        text = (out or b"").decode("utf-8", "replace")
+        if len(text) > max_output_bytes:
+            text = text[-max_output_bytes:]
+        return RunResult(
+            exit_code=proc.returncode if not timed_out else -1,
+            output=text,
+            timed_out=timed_out,
+            error="script timed out" if timed_out else "",
+        )
+    finally:
+        try:
+            os.unlink(path)
+        except OSError:
+            pass

@@ -1,0 +1,273 @@
+"""kmsg parsing, RAS catalog matching, fault injection, pstore, host."""
+
+import datetime
+import os
+
+import pytest
+
+from gpud_amd.apiv1.types import EventType, RepairActionType, utcnow
+from gpud_amd.pkg import ras_catalog
+from gpud_amd.pkg.eventstore import Store
+from gpud_amd.pkg.fault_injector import Injector, Request, SMIFailureInjector
+from gpud_amd.pkg.kmsg.parser import Message, parse_line
+from gpud_amd.pkg.kmsg.syncer import MatchResult, Syncer
+from gpud_amd.pkg.kmsg.watcher import Watcher
+from gpud_amd.pkg.kmsg.writer import NoopWriter, build_line
+
+
+# ---------------------------------------------------------------------------
+# kmsg parser
+# ---------------------------------------------------------------------------
+
+def test_parse_line_basic():
+    m = parse_line("6,1234,5000000,-;amdgpu 0000:0a:00.0: hello", boot_time_epoch=1000.0)
+    assert m is not None
+    assert m.priority == 6 and m.severity == 6 and m.facility == 0
+    assert m.sequence == 1234
+    assert m.timestamp_us == 5_000_000
+    assert m.message == "amdgpu 0000:0a:00.0: hello"
+    assert abs(m.time.timestamp() - 1005.0) < 1e-6
+
+
+def test_parse_line_rejects_continuation_and_garbage():
+    assert parse_line(" SUBSYSTEM=pci") is None
+    assert parse_line("not a kmsg line") is None
+    assert parse_line("") is None
+    assert parse_line("a,b,c;msg") is None
+
+
+def test_priority_severity_split():
+    # facility 3 (daemon), severity 2 (crit) -> priority 26
+    m = parse_line("26,1,0,-;x", boot_time_epoch=0.0)
+    assert m.facility == 3 and m.severity == 2
+    assert m.described_severity() == "crit"
+
+
+def test_build_kmsg_line():
+    assert build_line(2, "hello") == b"<2>hello"
+
+
+# ---------------------------------------------------------------------------
+# RAS catalog
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize(
+    "line,expected_name",
+    [
+        (
+            "[drm:amdgpu_job_timedout [amdgpu]] *ERROR* ring gfx_0.0.0 timeout, signaled seq=5",
+            "amdgpu_ring_timeout",
+        ),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: [gfxhub] no-retry page fault (src_id:0 ring:24 vmid:3 pasid:32770)",
+            "amdgpu_page_fault",
+        ),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: uncorrectable hardware error(ERREVENT_ATHUB_INTERRUPT) detected!",
+            "amdgpu_ras_uncorrectable",
+        ),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: 2 correctable hardware errors detected in UMC block",
+            "amdgpu_ras_corrected_error",
+        ),
+        ("amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!", "amdgpu_gpu_reset_begin"),
+        ("amdgpu 0000:0a:00.0: amdgpu: GPU reset(2) failed", "amdgpu_gpu_reset_failed"),
+        ("amdgpu 0000:0a:00.0: amdgpu: GPU reset(1) succeeded, trying to resume", "amdgpu_gpu_reset_succeeded"),
+        (
+            "python[999]: segfault at 10 ip 00007f12 sp 00007ffc error 4 in librccl.so.1.0[7f1234+100]",
+            "amd_rccl_segfault_in_librccl",
+        ),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: XGMI: link 3 error detected, fabric degraded",
+            "amdgpu_xgmi_error",
+        ),
+        ("kfd kfd: amdgpu: Failed to evict process queues", "kfd_evict_failed"),
+        ("Out of memory: Killed process 4242 (python3)", "memory_oom_kill"),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: GPU has fallen off the bus",
+            "amdgpu_fallen_off_bus",
+        ),
+        (
+            "amdgpu 0000:0a:00.0: amdgpu: SMU: I'm not done with your previous command: SMN_C2PMSG_66",
+            "amdgpu_smu_error",
+        ),
+    ],
+)
+def test_catalog_matches(line, expected_name):
+    res = ras_catalog.match(line)
+    assert res is not None, f"no match for {line!r}"
+    detail, groups = res
+    assert detail.name == expected_name
+
+
+def test_catalog_no_match_on_benign():
+    assert ras_catalog.match("usb 1-1: new high-speed USB device") is None
+    assert ras_catalog.match("amdgpu: loading firmware amdgpu/gfx950_sdma.bin") is None
+
+
+def test_catalog_page_fault_captures_pasid():
+    res = ras_catalog.match(
+        "amdgpu 0000:0a:00.0: amdgpu: [gfxhub] page fault (src_id:0 ring:24 vmid:3 pasid:32770)"
+    )
+    detail, groups = res
+    assert groups.get("hub") == "gfxhub"
+
+
+def test_catalog_severity_and_actions():
+    d = ras_catalog.lookup("amdgpu_ras_uncorrectable")
+    assert d.event_type == EventType.FATAL
+    assert RepairActionType.HARDWARE_INSPECTION in d.repair_actions
+    assert d.critical
+    sa = d.suggested_actions()
+    assert sa is not None and sa.repair_actions
+
+
+def test_every_injectable_matches_its_own_catalog_entry():
+    # each canned injectable line must be matched back to the same entry
+    for name, msg in ras_catalog.INJECTABLE.items():
+        res = ras_catalog.match(msg)
+        assert res is not None, f"injectable {name} does not match catalog"
+        assert res[0].name == name, f"{name} matched {res[0].name}"
+
+
+# ---------------------------------------------------------------------------
+# syncer: watcher -> match -> eventstore with dedup
+# ---------------------------------------------------------------------------
+
+def _ras_match(line):
+    res = ras_catalog.match(line)
+    if res is None:
+        return None
+    d, groups = res
+    return MatchResult(name=d.name, event_type=d.event_type, message=line, extra_info=groups or None)
+
+
+def test_syncer_inserts_and_dedups(mem_db):
+    rw, ro = mem_db
+    store = Store(rw, ro)
+    bucket = store.bucket("accelerator-amd-error-ras", disable_purge=True)
+    watcher = Watcher(path="/nonexistent")  # no real fd needed; we feed replay
+    syncer = Syncer(watcher, _ras_match, bucket)
+
+    now = utcnow()
+    msgs = [
+        Message(message="amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!", time=now),
+        Message(message="amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!", time=now),  # dup
+        Message(message="usb 1-1: boring", time=now),
+    ]
+    syncer.replay(msgs)
+    evs = bucket.get(now - datetime.timedelta(minutes=1))
+    assert len(evs) == 1
+    assert evs[0].name == "amdgpu_gpu_reset_begin"
+    # outside the dedup window it inserts again
+    later = now + datetime.timedelta(minutes=10)
+    syncer.replay([Message(message="amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!", time=later)])
+    evs = bucket.get(now - datetime.timedelta(minutes=1))
+    assert len(evs) == 2
+    store.close()
+
+
+# ---------------------------------------------------------------------------
+# fault injector
+# ---------------------------------------------------------------------------
+
+def test_inject_by_catalog_name():
+    w = NoopWriter()
+    inj = Injector(w)
+    err = inj.inject(Request(ras_event_name="amdgpu_ring_timeout"))
+    assert err is None
+    assert len(w.written) == 1
+    assert "ring gfx_0.0.0 timeout" in w.written[0][1]
+
+
+def test_inject_raw_kernel_message():
+    from gpud_amd.pkg.fault_injector import KernelMessage
+
+    w = NoopWriter()
+    inj = Injector(w)
+    err = inj.inject(Request(kernel_message=KernelMessage(message="custom msg", priority=4)))
+    assert err is None
+    assert w.written[0] == (4, "custom msg")
+
+
+def test_inject_unknown_name_errors():
+    inj = Injector(NoopWriter())
+    assert inj.inject(Request(ras_event_name="nope")) is not None
+    assert inj.inject(Request()) is not None
+
+
+def test_smi_failure_injector_flags():
+    fi = SMIFailureInjector()
+    assert not fi.any_active()
+    fi.gpu_lost_uuids.add("uuid-1")
+    assert fi.any_active()
+
+
+# ---------------------------------------------------------------------------
+# host / reboot event store
+# ---------------------------------------------------------------------------
+
+def test_reboot_event_store_records_once(mem_db):
+    from gpud_amd.pkg.host import RebootEventStore
+
+    rw, ro = mem_db
+    store = Store(rw, ro)
+    rs = RebootEventStore(store)
+    ev1 = rs.record_reboot()
+    assert ev1 is not None and ev1.name == "reboot"
+    # second call for the same boot is a no-op
+    assert rs.record_reboot() is None
+    since = ev1.time - datetime.timedelta(days=1)
+    assert rs.reboot_count_since(since) == 1
+    store.close()
+
+
+def test_host_identity_readers():
+    from gpud_amd.pkg import host
+
+    assert host.uptime_seconds() > 0
+    assert host.kernel_version()
+    assert host.hostname()
+    # boot_id may be empty in odd containers but should not raise
+    host.boot_id()
+    host.machine_id()
+    host.os_image()
+
+
+# ---------------------------------------------------------------------------
+# pstore
+# ---------------------------------------------------------------------------
+
+def test_pstore_scan_dedup(tmp_path, mem_db):
+    from gpud_amd.pkg.pstore import Scanner
+
+    rw, ro = mem_db
+    d = tmp_path / "pstore"
+    d.mkdir()
+    (d / "dmesg-efi-1").write_text("Kernel panic - not syncing: Fatal exception")
+    (d / "console-1").write_text("nothing interesting")
+    s = Scanner(rw, ro, pstore_dir=str(d))
+    findings = s.scan()
+    assert len(findings) == 1
+    assert findings[0][0] == "dmesg-efi-1"
+    assert "Kernel panic" in findings[0][1]
+    # second scan: already seen
+    assert s.scan() == []
+
+
+# ---------------------------------------------------------------------------
+# process runner
+# ---------------------------------------------------------------------------
+
+def test_run_bash_success_and_timeout():
+    from gpud_amd.pkg.process_runner import Runner, run_bash
+
+    r = run_bash("echo hello; echo err >&2; exit 3", timeout_seconds=10)
+    assert r.exit_code == 3
+    assert "hello" in r.output and "err" in r.output
+    t = run_bash("sleep 30", timeout_seconds=0.3)
+    assert t.timed_out and t.exit_code == -1
+
+    runner = Runner()
+    ok = runner.run_until_completion("exit 0", timeout_seconds=10)
+    assert ok.exit_code == 0
