@@ -1,0 +1,140 @@
+"""accelerator-amd-gpu-counts — SMI count vs lspci count vs expected.
+
+Reference: components/accelerator/nvidia/gpu-counts (lspci | grep NVIDIA
+vs NVML device count vs expected — gpu-counts/component.go:216-368).
+AMD-first: count AMD display/processing accelerators on the PCI bus
+(vendor 0x1002, Instinct devices report as "Processing accelerators"),
+compare with the amdsmi enumeration and the configured expected count.
+"""
+
+from __future__ import annotations
+
+import subprocess
+from typing import Callable, Optional
+
+from ...apiv1.types import HealthStateType, RepairActionType, SuggestedActions
+from ..base import CheckResult, Component, GPUdInstance, TickerComponent
+from ..metrics_util import ComponentGauges
+from .shared import SmiComponentMixin
+
+NAME = "accelerator-amd-gpu-counts"
+
+
+def count_amd_gpus_lspci(lspci_command: str = "") -> Optional[int]:
+    """Count AMD accelerator/VGA functions via lspci -n (vendor 1002)."""
+    cmd = lspci_command or "lspci"
+    try:
+        out = subprocess.run(
+            [cmd, "-n"], capture_output=True, text=True, timeout=10
+        ).stdout
+    except (OSError, subprocess.TimeoutExpired):
+        return None
+    n = 0
+    for line in out.splitlines():
+        # "0a:00.0 1200: 1002:75a0" — class 0x1200 processing accelerator,
+        # 0x0300/0x0380 display classes; count vendor 1002 in those classes
+        parts = line.split()
+        if len(parts) >= 3 and parts[2].startswith("1002:"):
+            cls = parts[1].rstrip(":")
+            if cls.startswith(("1200", "0300", "0302", "0380")):
+                n += 1
+    return n
+
+
+class GPUCountsComponent(TickerComponent, SmiComponentMixin):
+    def __init__(self, inst: GPUdInstance):
+        super().__init__()
+        self._smi = inst.smi
+        self._gauges = ComponentGauges(NAME, inst.metrics_registry)
+        self.expected = inst.expected_gpu_count
+        self._lspci_command = inst.lspci_command
+        self.count_lspci: Callable[[], Optional[int]] = lambda: count_amd_gpus_lspci(
+            self._lspci_command
+        )
+
+    @property
+    def name(self) -> str:
+        return NAME
+
+    def tags(self) -> list:
+        return ["accelerator", "amd", "gpu", NAME]
+
+    def is_supported(self) -> bool:
+        return True  # meaningful even when SMI is missing (counts mismatch!)
+
+    def check(self) -> CheckResult:
+        smi_count = 0
+        smi_err = ""
+        if self._smi is not None and self._smi.exists:
+            try:
+                smi_count = self._smi.device_count()
+            except Exception as e:
+                smi_err = str(e)
+        pci_count = self.count_lspci()
+        self._gauges.set(
+            "accelerator_amd_gpu_counts_smi",
+            "GPUs enumerated by amdsmi",
+            smi_count,
+        )
+        if pci_count is not None:
+            self._gauges.set(
+                "accelerator_amd_gpu_counts_lspci",
+                "AMD accelerators on the PCI bus",
+                pci_count,
+            )
+        extra = {
+            "smi_count": str(smi_count),
+            "lspci_count": "" if pci_count is None else str(pci_count),
+            "expected": str(self.expected),
+        }
+        if smi_err:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason="amdsmi device enumeration failed",
+                error=smi_err,
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="GPU enumeration failure",
+                    repair_actions=[RepairActionType.REBOOT_SYSTEM],
+                ),
+            )
+        if self.expected > 0 and smi_count < self.expected:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason=f"expected {self.expected} GPU(s), amdsmi sees {smi_count}",
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="missing GPU(s)",
+                    repair_actions=[
+                        RepairActionType.REBOOT_SYSTEM,
+                        RepairActionType.HARDWARE_INSPECTION,
+                    ],
+                ),
+            )
+        if pci_count is not None and smi_count and pci_count != smi_count:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason=(
+                    f"lspci sees {pci_count} AMD accelerator(s) but amdsmi "
+                    f"enumerates {smi_count} — a GPU may have dropped off the driver"
+                ),
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="PCI/driver GPU count mismatch",
+                    repair_actions=[RepairActionType.REBOOT_SYSTEM],
+                ),
+            )
+        return CheckResult(
+            NAME,
+            reason=f"gpu counts consistent (smi={smi_count}"
+            + (f", lspci={pci_count}" if pci_count is not None else "")
+            + ")",
+            extra_info=extra,
+        )
+
+
+def new(inst: GPUdInstance) -> Component:
+    return GPUCountsComponent(inst)

@@ -1,0 +1,192 @@
+"""accelerator-amd-xgmi — inter-GPU xGMI fabric health.
+
+The single-node MI355X replacement for the reference's nvlink +
+fabric-manager + infiniband trio (reference:
+components/accelerator/nvidia/nvlink/nvlink.go:86-93 per-link counters;
+fabric-manager probes; infiniband port state/flap store — SURVEY.md §5
+distributed-backend note): per-GPU link states (7 point-to-point links ×
+≈153 GB/s on an 8-GPU node), per-link traffic counters from amdsmi link
+metrics, the device xGMI error status, and an expected-link-count rule.
+
+Link-flap detection: a link observed DOWN that was previously UP records an
+event (the infiniband flap-store analog, simplified to the event bucket).
+"""
+
+from __future__ import annotations
+
+import datetime
+from typing import Callable, Dict, List
+
+from ...apiv1.types import (
+    Event,
+    EventType,
+    HealthStateType,
+    RepairActionType,
+    SuggestedActions,
+    utcnow,
+)
+from ..base import CheckResult, Component, GPUdInstance, TickerComponent
+from ..metrics_util import ComponentGauges
+from .shared import SmiComponentMixin
+
+NAME = "accelerator-amd-xgmi"
+
+LINK_UP = 1
+LINK_DOWN = 0
+
+
+class XGMIComponent(TickerComponent, SmiComponentMixin):
+    def __init__(self, inst: GPUdInstance):
+        super().__init__()
+        self._smi = inst.smi
+        self._shared = inst.shared_snapshots
+        self._gauges = ComponentGauges(NAME, inst.metrics_registry)
+        self._bucket = (
+            inst.event_store.bucket(NAME) if inst.event_store is not None else None
+        )
+        cfg = inst.config
+        self.expected_links = (
+            getattr(cfg, "expected_xgmi_link_count", 0) if cfg else 0
+        )
+        self._last_states: Dict[str, List[int]] = {}
+        self.get_snapshots: Callable = (
+            self._shared.get if self._shared is not None else lambda: {}
+        )
+        self.get_devices: Callable = (
+            self._smi.devices if self._smi is not None else dict
+        )
+
+    @property
+    def name(self) -> str:
+        return NAME
+
+    def tags(self) -> list:
+        return ["accelerator", "amd", "gpu", "fabric", NAME]
+
+    def is_supported(self) -> bool:
+        return self._smi is not None and self._smi.exists
+
+    def events(self, since: datetime.datetime):
+        return self._bucket.get(since) if self._bucket is not None else []
+
+    def check(self) -> CheckResult:
+        guard = self.smi_guard()
+        if guard is not None:
+            return guard
+        snaps = self.get_snapshots()
+        down_by_uuid: Dict[str, List[int]] = {}
+        err_by_uuid: Dict[str, int] = {}
+        missing_links: List[str] = []
+        extra: Dict[str, str] = {}
+        any_links = False
+        for uuid, snap in snaps.items():
+            x = snap.get("xgmi_link_status")
+            if x:
+                any_links = True
+                states = list(x.get("states") or [])
+                up = sum(1 for s in states if s == LINK_UP)
+                down = [i for i, s in enumerate(states) if s == LINK_DOWN]
+                self._gauges.set(
+                    "accelerator_amd_xgmi_links_up",
+                    "xGMI links in UP state",
+                    up,
+                    uuid=uuid,
+                )
+                self._gauges.set(
+                    "accelerator_amd_xgmi_links_total",
+                    "Total xGMI links",
+                    len(states),
+                    uuid=uuid,
+                )
+                extra[f"{uuid}.links"] = f"{up}/{len(states)} up"
+                if down:
+                    down_by_uuid[uuid] = down
+                # flap detection vs previous observation
+                prev = self._last_states.get(uuid)
+                if prev is not None and self._bucket is not None:
+                    for i, s in enumerate(states):
+                        if i < len(prev) and prev[i] == LINK_UP and s == LINK_DOWN:
+                            self._bucket.insert(
+                                Event(
+                                    time=utcnow(),
+                                    component=NAME,
+                                    name="amd_xgmi_link_down",
+                                    type=EventType.CRITICAL,
+                                    message=f"xGMI link {i} on {uuid} went DOWN",
+                                )
+                            )
+                self._last_states[uuid] = states
+                if self.expected_links > 0 and up < self.expected_links:
+                    missing_links.append(uuid)
+            err = snap.get("xgmi_error_status")
+            if err is not None and int(err) != 0:
+                err_by_uuid[uuid] = int(err)
+                self._gauges.set(
+                    "accelerator_amd_xgmi_error_status",
+                    "xGMI error status (0=no errors)",
+                    int(err),
+                    uuid=uuid,
+                )
+        # per-link traffic counters (separate, heavier SMI call)
+        try:
+            for uuid, dev in self.get_devices().items():
+                lm = dev.link_metrics()
+                for i, link in enumerate(lm.get("links", [])):
+                    if int(link.get("link_type", 0)) != 2:  # XGMI only
+                        continue
+                    self._gauges.set(
+                        "accelerator_amd_xgmi_read_kb_total",
+                        "Accumulated xGMI read traffic (KB)",
+                        float(link.get("read_kb", 0)),
+                        uuid=uuid,
+                        link=str(i),
+                    )
+                    self._gauges.set(
+                        "accelerator_amd_xgmi_write_kb_total",
+                        "Accumulated xGMI write traffic (KB)",
+                        float(link.get("write_kb", 0)),
+                        uuid=uuid,
+                        link=str(i),
+                    )
+        except Exception:
+            pass  # traffic counters are best-effort
+
+        if down_by_uuid or err_by_uuid or missing_links:
+            parts = []
+            if down_by_uuid:
+                parts.append(
+                    "links down: "
+                    + "; ".join(f"{u} {links}" for u, links in down_by_uuid.items())
+                )
+            if err_by_uuid:
+                parts.append(
+                    "xgmi errors on: " + ", ".join(err_by_uuid.keys())
+                )
+            if missing_links:
+                parts.append(
+                    f"fewer than {self.expected_links} links up on: "
+                    + ", ".join(missing_links)
+                )
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason="; ".join(parts),
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="degraded xGMI fabric",
+                    repair_actions=[
+                        RepairActionType.REBOOT_SYSTEM,
+                        RepairActionType.HARDWARE_INSPECTION,
+                    ],
+                ),
+            )
+        reason = (
+            f"all xGMI links healthy on {len(snaps)} GPU(s)"
+            if any_links
+            else "no xGMI links reported (single GPU?)"
+        )
+        return CheckResult(NAME, reason=reason, extra_info=extra or None)
+
+
+def new(inst: GPUdInstance) -> Component:
+    return XGMIComponent(inst)

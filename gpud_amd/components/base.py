@@ -237,6 +237,7 @@ class GPUdInstance:
     """
 
     smi: Any = None  # gpud_amd.smi.Instance (or None on GPU-less hosts)
+    shared_snapshots: Any = None  # accelerator.shared.SharedSnapshots
     db_rw: Any = None  # sqlite3 connection pool (read-write)
     db_ro: Any = None  # sqlite3 connection pool (read-only)
     event_store: Any = None  # pkg.eventstore.Store

@@ -1,0 +1,84 @@
+"""pci — ACS (Access Control Services) status on PCI bridges.
+
+Reference: components/pci (lspci ACS status on bridges; ACS-on hurts GPU
+P2P — pci/component.go:26). On MI355X nodes ACS on the bridges above the
+GPUs forces P2P/xGMI-adjacent PCIe DMA through the root complex.
+"""
+
+from __future__ import annotations
+
+import subprocess
+from typing import Callable, List, Optional
+
+from ...apiv1.types import HealthStateType
+from ..base import CheckResult, Component, GPUdInstance, TickerComponent
+
+NAME = "pci"
+
+
+def bridges_with_acs_enabled(lspci_command: str = "") -> Optional[List[str]]:
+    cmd = lspci_command or "lspci"
+    try:
+        out = subprocess.run(
+            [cmd, "-vvv"], capture_output=True, text=True, timeout=30
+        )
+        if out.returncode != 0:
+            return None
+    except (OSError, subprocess.TimeoutExpired):
+        return None
+    enabled = []
+    current_dev = ""
+    is_bridge = False
+    in_acs_cap = False
+    for line in out.stdout.splitlines():
+        if line and not line[0].isspace():
+            current_dev = line.split(" ", 1)[0]
+            is_bridge = "PCI bridge" in line
+            in_acs_cap = False
+            continue
+        if not is_bridge:
+            continue
+        s = line.strip()
+        if s.startswith("Capabilities:") and "Access Control Services" in s:
+            in_acs_cap = True
+            continue
+        if in_acs_cap and s.startswith("ACSCtl:"):
+            # e.g. "ACSCtl: SrcValid+ TransBlk- ReqRedir+ ..."
+            if "SrcValid+" in s:
+                enabled.append(current_dev)
+            in_acs_cap = False
+    return enabled
+
+
+class PCIComponent(TickerComponent):
+    def __init__(self, inst: GPUdInstance):
+        super().__init__()
+        self._lspci_command = inst.lspci_command
+        self.get_acs_bridges: Callable = lambda: bridges_with_acs_enabled(
+            self._lspci_command
+        )
+
+    @property
+    def name(self) -> str:
+        return NAME
+
+    def tags(self) -> list:
+        return [NAME]
+
+    def check(self) -> CheckResult:
+        bridges = self.get_acs_bridges()
+        if bridges is None:
+            return CheckResult(NAME, reason="lspci unavailable; ACS check skipped")
+        if bridges:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.DEGRADED,
+                reason=f"ACS enabled on {len(bridges)} PCI bridge(s) — "
+                "disable ACS for full GPU P2P bandwidth",
+                extra_info={"bridges": ",".join(bridges[:16])},
+            )
+        return CheckResult(NAME, reason="no PCI bridges with ACS enabled")
+
+
+def new(inst: GPUdInstance) -> Component:
+    return PCIComponent(inst)

@@ -1,0 +1,45 @@
+"""Prometheus gauge helpers for components.
+
+Metric naming follows the reference convention: no
+namespace, subsystem = component name with underscores (e.g.
+``accelerator_amd_temperature_current_celsius``), every metric carrying the
+curried ``gpud_component`` label plus ``uuid`` for per-GPU gauges
+(reference: components/accelerator/nvidia/temperature/metrics.go:10-35).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+from prometheus_client import CollectorRegistry, Gauge
+
+from ..pkg.metrics.registry import LABEL_COMPONENT
+
+
+class ComponentGauges:
+    """Lazily-created gauges bound to one component and one registry."""
+
+    def __init__(self, component: str, registry: Optional[CollectorRegistry]):
+        self.component = component
+        self.registry = registry
+        self._gauges: Dict[str, Gauge] = {}
+
+    def gauge(self, name: str, doc: str, extra_labels: List[str] = ()) -> Optional[Gauge]:
+        if self.registry is None:
+            return None
+        g = self._gauges.get(name)
+        if g is None:
+            g = Gauge(
+                name,
+                doc,
+                [LABEL_COMPONENT, *extra_labels],
+                registry=self.registry,
+            )
+            self._gauges[name] = g
+        return g
+
+    def set(self, name: str, doc: str, value: float, **labels: str) -> None:
+        g = self.gauge(name, doc, sorted(labels.keys()))
+        if g is None:
+            return
+        g.labels(**{LABEL_COMPONENT: self.component, **labels}).set(value)
