@@ -1,0 +1,234 @@
+"""gpud_amd flagship benchmark: the daemon poll cycle on real MI355X GPUs.
+
+BASELINE.json metric: "daemon CPU% + p50 poll-cycle ms at 8 MI355X; diag
+MFMA TFLOPS/GPU" — the reference (leptonai/gpud) publishes no numbers
+(BASELINE.md), so this measures our daemon directly:
+
+  * one "step" = one complete accelerator poll cycle for the GPUs this
+    rank monitors: a native amdsmi telemetry sweep (temperature, power,
+    clocks, activity, VRAM, ECC, throttle, xGMI, bad pages) followed by
+    every accelerator component Check() evaluating health rules and
+    updating Prometheus gauges — exactly what the daemon does per tick;
+  * weak scaling: rank r monitors GPU r, so per-GPU work is fixed as N
+    grows (the reference daemon monitors all GPUs from one process; flat
+    scaling to 8 GPUs is the design goal — SURVEY.md §7);
+  * headline value = p50 poll-cycle milliseconds (max over ranks);
+  * extras: daemon CPU%% during the timed region and the per-GPU MFMA
+    bf16 diag TFLOPS (measured after the timed region).
+
+Launch: python bench.py --gpus N --steps K --warmup W
+(N>1 via torch.distributed.run, one rank per GPU over RCCL).
+"""
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+class _RankFilteredSMI:
+    """Expose only this rank's GPU through the Instance snapshot API."""
+
+    def __init__(self, inst, index: int):
+        self._inst = inst
+        self.index = index
+        uuids = inst.device_uuids()
+        self.uuid = uuids[index % len(uuids)] if uuids else ""
+        self.failure_injector = inst.failure_injector
+
+    @property
+    def exists(self):
+        return self._inst.exists
+
+    def init_error(self):
+        return self._inst.init_error()
+
+    def devices(self):
+        devs = self._inst.devices()
+        return {self.uuid: devs[self.uuid]} if self.uuid in devs else {}
+
+    def device_uuids(self):
+        return [self.uuid] if self.uuid else []
+
+    def device_count(self):
+        return 1 if self.uuid else 0
+
+    @property
+    def product_name(self):
+        return self._inst.product_name
+
+    def snapshot_all(self):
+        devs = self._inst.devices()
+        if self.uuid not in devs:
+            return {}
+        return {self.uuid: devs[self.uuid].snapshot()}
+
+    def shutdown(self):
+        pass
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--mock", action="store_true", help="CPU-only: mock SMI backend")
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    if args.mock:
+        os.environ["GPUD_AMDSMI_MOCK"] = "1"
+
+    import torch
+    import torch.distributed as dist
+
+    use_cuda = torch.cuda.is_available() and not args.mock
+    if use_cuda:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+
+    distributed = world > 1
+    if distributed:
+        backend = "nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from gpud_amd import smi as smi_pkg
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+    import psutil
+
+    cfg = Config()
+    core = build_core(
+        cfg,
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    inst = core.smi_instance
+    if inst.exists and inst.device_count() > 0:
+        filtered = _RankFilteredSMI(inst, local_rank)
+        core.shared_snapshots.smi = filtered
+        core.gpud_instance.smi = filtered
+        n_gpus_seen = 1
+        data_source = "mock" if args.mock else "amdsmi"
+    else:
+        if not args.mock:
+            print(
+                json.dumps(
+                    {
+                        "error": "no AMD GPU visible and --mock not set",
+                        "init_error": inst.init_error(),
+                    }
+                ),
+                file=sys.stderr,
+            )
+            return 1
+        n_gpus_seen = inst.device_count()
+        data_source = "mock"
+
+    accel_components = [
+        c
+        for c in core.registry.all_components()
+        if c.name.startswith("accelerator-amd-")
+        and "diag" not in c.name
+    ]
+
+    def one_cycle() -> None:
+        core.shared_snapshots.refresh()
+        for c in accel_components:
+            c.trigger_check()
+
+    def barrier_sync() -> None:
+        if distributed:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    proc = psutil.Process()
+
+    # warmup
+    for _ in range(args.warmup):
+        one_cycle()
+
+    barrier_sync()
+    cpu0 = proc.cpu_times()
+    wall0 = time.monotonic()
+    cycle_ms = []
+    for _ in range(args.steps):
+        t0 = time.perf_counter()
+        one_cycle()
+        cycle_ms.append((time.perf_counter() - t0) * 1000.0)
+    barrier_sync()
+    wall1 = time.monotonic()
+    cpu1 = proc.cpu_times()
+
+    wall = wall1 - wall0
+    cpu_used = (cpu1.user - cpu0.user) + (cpu1.system - cpu0.system)
+    cpu_pct = 100.0 * cpu_used / wall if wall > 0 else 0.0
+    p50 = statistics.median(cycle_ms)
+    mean_ms = sum(cycle_ms) / len(cycle_ms)
+    p99 = sorted(cycle_ms)[max(0, int(len(cycle_ms) * 0.99) - 1)]
+
+    # diag MFMA TFLOPS (after the timed region; real GPU only)
+    mfma_tflops = None
+    if use_cuda:
+        try:
+            from gpud_amd.diag import _diag
+
+            _diag.set_device(local_rank % max(torch.cuda.device_count(), 1))
+            res = _diag.mfma_stress_bf16(iters=1024, workgroups=1024)
+            if res["verified"]:
+                mfma_tflops = round(float(res["tflops"]), 1)
+        except Exception:
+            mfma_tflops = None
+
+    # MAX over ranks for the time metrics (the whole-job aggregate)
+    if distributed:
+        t = torch.tensor([p50, mean_ms, cpu_pct, p99], dtype=torch.float64)
+        if use_cuda:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        p50, mean_ms, cpu_pct_max, p99 = [float(x) for x in t.tolist()]
+    else:
+        cpu_pct_max = cpu_pct
+
+    if rank == 0:
+        out = {
+            "metric": "poll_cycle_p50_ms",
+            "value": round(p50, 4),
+            "unit": "ms",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(mean_ms, 4),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic" if data_source == "mock" else "live-telemetry",
+            "config": {
+                "model": "gpud-amd accelerator poll cycle",
+                "components_per_cycle": len(accel_components),
+                "gpus_per_rank": 1 if data_source == "amdsmi" else n_gpus_seen,
+                "data_source": data_source,
+                "poll_cycle_p99_ms": round(p99, 4),
+                "daemon_cpu_percent": round(cpu_pct_max, 2),
+                "mfma_bf16_tflops_per_gpu": mfma_tflops,
+            },
+        }
+        print(json.dumps(out))
+
+    core.close()
+    if distributed:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -42,15 +42,30 @@ def count_amd_gpus_lspci(lspci_command: str = "") -> Optional[int]:
 
 
 class GPUCountsComponent(TickerComponent, SmiComponentMixin):
+    # The PCI topology changes only on hotplug/driver rebind, and a GPU
+    # dropping off the driver is caught every cycle by the amdsmi
+    # enumeration below — so the lspci crosscheck (a fork+exec) is cached
+    # for 5 minutes instead of spawned per check.
+    LSPCI_TTL_SECONDS = 300.0
+
     def __init__(self, inst: GPUdInstance):
         super().__init__()
         self._smi = inst.smi
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
         self.expected = inst.expected_gpu_count
         self._lspci_command = inst.lspci_command
-        self.count_lspci: Callable[[], Optional[int]] = lambda: count_amd_gpus_lspci(
-            self._lspci_command
-        )
+        self._lspci_cache: Optional[int] = None
+        self._lspci_cached_at = 0.0
+        self.count_lspci: Callable[[], Optional[int]] = self._cached_lspci
+
+    def _cached_lspci(self) -> Optional[int]:
+        import time
+
+        now = time.monotonic()
+        if now - self._lspci_cached_at > self.LSPCI_TTL_SECONDS:
+            self._lspci_cache = count_amd_gpus_lspci(self._lspci_command)
+            self._lspci_cached_at = now
+        return self._lspci_cache
 
     @property
     def name(self) -> str:

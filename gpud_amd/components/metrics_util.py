@@ -17,12 +17,19 @@ from ..pkg.metrics.registry import LABEL_COMPONENT
 
 
 class ComponentGauges:
-    """Lazily-created gauges bound to one component and one registry."""
+    """Lazily-created gauges bound to one component and one registry.
+
+    Child (labelled) gauge objects are cached: prometheus_client's
+    ``labels()`` re-validates and re-hashes label tuples on every call,
+    which dominates a tight poll cycle with hundreds of per-GPU gauge
+    updates — the cache turns each update into a dict hit + ``set()``.
+    """
 
     def __init__(self, component: str, registry: Optional[CollectorRegistry]):
         self.component = component
         self.registry = registry
         self._gauges: Dict[str, Gauge] = {}
+        self._children: Dict[tuple, object] = {}
 
     def gauge(self, name: str, doc: str, extra_labels: List[str] = ()) -> Optional[Gauge]:
         if self.registry is None:
@@ -39,7 +46,14 @@ class ComponentGauges:
         return g
 
     def set(self, name: str, doc: str, value: float, **labels: str) -> None:
-        g = self.gauge(name, doc, sorted(labels.keys()))
-        if g is None:
+        if self.registry is None:
             return
-        g.labels(**{LABEL_COMPONENT: self.component, **labels}).set(value)
+        key = (name, *sorted(labels.items()))
+        child = self._children.get(key)
+        if child is None:
+            g = self.gauge(name, doc, sorted(labels.keys()))
+            if g is None:
+                return
+            child = g.labels(**{LABEL_COMPONENT: self.component, **labels})
+            self._children[key] = child
+        child.set(value)
