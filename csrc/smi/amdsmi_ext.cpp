@@ -19,6 +19,7 @@
 
 #include <amd_smi/amdsmi.h>
 
+#include <chrono>
 #include <cstring>
 #include <mutex>
 #include <stdexcept>
@@ -556,7 +557,6 @@ struct Snapshot {
   amdsmi_engine_usage_t act;
   amdsmi_vram_usage_t vram;
   amdsmi_error_count_t ecc;
-  amdsmi_violation_status_t viol;
   amdsmi_xgmi_link_status_t xgmi;
   int xgmi_err = -1;
   uint32_t throttle_status = 0;
@@ -566,17 +566,27 @@ struct Snapshot {
   uint32_t bp_total = 0, bp_reserved = 0, bp_pending = 0, bp_unreservable = 0;
   uint32_t bp_threshold = 0;
   bool ok_bp_threshold = false;
+  // throttle residency accumulators from gpu_metrics (violation analog)
+  uint64_t thr_acc_counter = 0, thr_prochot = 0, thr_ppt = 0, thr_socket = 0,
+           thr_vr = 0, thr_hbm = 0;
 };
 
 void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
-  // temperatures (current + limits)
-  s.ok_temp =
+  // temperatures (current + limits) — some SKUs/partitions expose only a
+  // subset of sensors (observed: no EDGE on an MI355X 0x75a3 box), so the
+  // block is reported when ANY current-temperature sensor answers
+  bool ok_edge =
       amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_EDGE, AMDSMI_TEMP_CURRENT,
                              &s.temp_edge) == AMDSMI_STATUS_SUCCESS;
-  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT, AMDSMI_TEMP_CURRENT,
-                         &s.temp_hotspot);
-  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_VRAM, AMDSMI_TEMP_CURRENT,
-                         &s.temp_vram);
+  bool ok_hot =
+      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
+                             AMDSMI_TEMP_CURRENT,
+                             &s.temp_hotspot) == AMDSMI_STATUS_SUCCESS;
+  bool ok_vram =
+      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_VRAM, AMDSMI_TEMP_CURRENT,
+                             &s.temp_vram) == AMDSMI_STATUS_SUCCESS;
+  s.ok_temp = ok_edge || ok_hot || ok_vram;
+  if (!ok_edge && ok_hot) s.temp_edge = s.temp_hotspot;
   amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_EDGE, AMDSMI_TEMP_CRITICAL,
                          &s.temp_edge_limit);
   amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT, AMDSMI_TEMP_CRITICAL,
@@ -603,9 +613,11 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
   std::memset(&s.ecc, 0, sizeof(s.ecc));
   s.ok_ecc =
       amdsmi_get_gpu_total_ecc_count(h, &s.ecc) == AMDSMI_STATUS_SUCCESS;
-  std::memset(&s.viol, 0, sizeof(s.viol));
-  s.ok_throttle =
-      amdsmi_get_violation_status(h, &s.viol) == AMDSMI_STATUS_SUCCESS;
+  // NOTE: amdsmi_get_violation_status is NOT called here — it blocks ~100 ms
+  // per GPU (double-samples internally to compute per_* rates). The same
+  // throttle residency accumulators come from gpu_metrics below at ~0.15 ms;
+  // the throttle component derives activity from deltas across its own
+  // polls. The explicit violation_status() getter remains for manual diags.
   std::memset(&s.xgmi, 0, sizeof(s.xgmi));
   s.ok_xgmi =
       amdsmi_get_gpu_xgmi_link_status(h, &s.xgmi) == AMDSMI_STATUS_SUCCESS;
@@ -644,6 +656,13 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
     s.cur_gfxclk = gm.current_gfxclk;
     s.cur_uclk = gm.current_uclk;
     s.avg_socket_power = gm.average_socket_power;
+    s.ok_throttle = true;
+    s.thr_acc_counter = gm.accumulation_counter;
+    s.thr_prochot = gm.prochot_residency_acc;
+    s.thr_ppt = gm.ppt_residency_acc;
+    s.thr_socket = gm.socket_thm_residency_acc;
+    s.thr_vr = gm.vr_thm_residency_acc;
+    s.thr_hbm = gm.hbm_thm_residency_acc;
   }
 }
 
@@ -703,18 +722,15 @@ py::dict snapshot_to_dict(const Snapshot& s) {
     d["ecc"] = e;
   }
   if (s.ok_throttle) {
+    // residency accumulators from gpu_metrics; the throttle component
+    // derives activity/rates from deltas between its own polls
     py::dict v;
-    v["acc_counter"] = s.viol.acc_counter;
-    v["acc_prochot_thrm"] = s.viol.acc_prochot_thrm;
-    v["acc_ppt_pwr"] = s.viol.acc_ppt_pwr;
-    v["acc_socket_thrm"] = s.viol.acc_socket_thrm;
-    v["acc_vr_thrm"] = s.viol.acc_vr_thrm;
-    v["acc_hbm_thrm"] = s.viol.acc_hbm_thrm;
-    v["active_prochot_thrm"] = static_cast<int>(s.viol.active_prochot_thrm);
-    v["active_ppt_pwr"] = static_cast<int>(s.viol.active_ppt_pwr);
-    v["active_socket_thrm"] = static_cast<int>(s.viol.active_socket_thrm);
-    v["active_vr_thrm"] = static_cast<int>(s.viol.active_vr_thrm);
-    v["active_hbm_thrm"] = static_cast<int>(s.viol.active_hbm_thrm);
+    v["acc_counter"] = s.thr_acc_counter;
+    v["acc_prochot_thrm"] = s.thr_prochot;
+    v["acc_ppt_pwr"] = s.thr_ppt;
+    v["acc_socket_thrm"] = s.thr_socket;
+    v["acc_vr_thrm"] = s.thr_vr;
+    v["acc_hbm_thrm"] = s.thr_hbm;
     d["violation"] = v;
   }
   if (s.ok_xgmi) {
@@ -757,6 +773,65 @@ py::dict metrics_snapshot(int index) {
     take_snapshot(h, s);
   }
   return snapshot_to_dict(s);
+}
+
+// per-call wall-time breakdown of one snapshot — the poll-latency
+// instrument used to pick which SMI calls stay on the fast path
+py::dict snapshot_timings(int index) {
+  auto h = handle_at(index);
+  py::dict out;
+  auto time_call = [&](const char* name, auto&& fn) {
+    auto t0 = std::chrono::steady_clock::now();
+    fn();
+    auto us = std::chrono::duration_cast<std::chrono::microseconds>(
+                  std::chrono::steady_clock::now() - t0)
+                  .count();
+    out[name] = (double)us;
+  };
+  {
+    int64_t tv;
+    time_call("temp_edge", [&] {
+      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_EDGE,
+                             AMDSMI_TEMP_CURRENT, &tv);
+    });
+    time_call("temp_hotspot", [&] {
+      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
+                             AMDSMI_TEMP_CURRENT, &tv);
+    });
+    amdsmi_power_info_t pw;
+    time_call("power_info", [&] { amdsmi_get_power_info(h, &pw); });
+    amdsmi_power_cap_info_t pc;
+    time_call("power_cap", [&] { amdsmi_get_power_cap_info(h, 0, &pc); });
+    amdsmi_clk_info_t ci;
+    time_call("clock_gfx", [&] {
+      amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_GFX, &ci);
+    });
+    time_call("clock_mem", [&] {
+      amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_MEM, &ci);
+    });
+    amdsmi_engine_usage_t eu;
+    time_call("activity", [&] { amdsmi_get_gpu_activity(h, &eu); });
+    amdsmi_vram_usage_t vu;
+    time_call("vram_usage", [&] { amdsmi_get_gpu_vram_usage(h, &vu); });
+    amdsmi_error_count_t ec;
+    time_call("ecc_total", [&] { amdsmi_get_gpu_total_ecc_count(h, &ec); });
+    amdsmi_violation_status_t vs;
+    time_call("violation", [&] { amdsmi_get_violation_status(h, &vs); });
+    amdsmi_xgmi_link_status_t xs;
+    time_call("xgmi_link_status",
+              [&] { amdsmi_get_gpu_xgmi_link_status(h, &xs); });
+    amdsmi_xgmi_status_t xe;
+    time_call("xgmi_error", [&] { amdsmi_gpu_xgmi_error_status(h, &xe); });
+    uint32_t bp = 0;
+    time_call("bad_pages_count",
+              [&] { amdsmi_get_gpu_bad_page_info(h, &bp, nullptr); });
+    uint32_t thr = 0;
+    time_call("bad_page_threshold",
+              [&] { amdsmi_get_gpu_bad_page_threshold(h, &thr); });
+    amdsmi_gpu_metrics_t gm;
+    time_call("gpu_metrics", [&] { amdsmi_get_gpu_metrics_info(h, &gm); });
+  }
+  return out;
 }
 
 py::list metrics_snapshot_all() {
@@ -817,6 +892,8 @@ PYBIND11_MODULE(_amdsmi, m) {
   m.def("energy_count", &energy_count, py::arg("index"));
   m.def("metrics_snapshot", &metrics_snapshot, py::arg("index"),
         "Full telemetry snapshot for one GPU in a single native call");
+  m.def("snapshot_timings", &snapshot_timings, py::arg("index"),
+        "Per-SMI-call microsecond timing of one snapshot (poll tuning)");
   m.def("metrics_snapshot_all", &metrics_snapshot_all,
         "Telemetry snapshots for every GPU, GIL released for the whole sweep");
 

@@ -54,6 +54,11 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
             self._shared.get if self._shared is not None else lambda: {}
         )
         self.get_now: Callable = utcnow
+        # last residency accumulators per uuid: activity = positive delta
+        # between this component's consecutive polls (the fast-path snapshot
+        # reads gpu_metrics accumulators; amdsmi's own violation_status call
+        # blocks ~100 ms/GPU double-sampling, so we difference ourselves)
+        self._last_acc: Dict[str, Dict[str, int]] = {}
 
     @property
     def name(self) -> str:
@@ -68,11 +73,18 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
     def events(self, since: datetime.datetime):
         return self._bucket.get(since) if self._bucket is not None else []
 
-    def _active_throttlers(self, v: Dict) -> List[Tuple[str, str]]:
+    def _active_throttlers(self, uuid: str, v: Dict) -> List[Tuple[str, str]]:
         out = []
+        prev = self._last_acc.get(uuid, {})
+        cur: Dict[str, int] = {}
         for key, desc in _THROTTLERS:
-            if int(v.get(f"active_{key}", 0)):
+            acc = int(v.get(f"acc_{key}", 0))
+            cur[key] = acc
+            explicit = int(v.get(f"active_{key}", 0))
+            rising = key in prev and acc > prev[key]
+            if explicit or rising:
                 out.append((key, desc))
+        self._last_acc[uuid] = cur
         return out
 
     def check(self) -> CheckResult:
@@ -94,7 +106,7 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
                     float(v.get(f"acc_{key}", 0)),
                     uuid=uuid,
                 )
-            active = self._active_throttlers(v)
+            active = self._active_throttlers(uuid, v)
             if active:
                 descs = [d for _k, d in active]
                 active_by_uuid[uuid] = descs
