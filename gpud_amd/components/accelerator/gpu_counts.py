@@ -128,19 +128,36 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
                     ],
                 ),
             )
-        if pci_count is not None and smi_count and pci_count != smi_count:
+        if pci_count is not None and smi_count and pci_count < smi_count:
+            # fewer devices on the bus than the driver claims — real trouble
             return CheckResult(
                 NAME,
                 health=HealthStateType.UNHEALTHY,
                 reason=(
                     f"lspci sees {pci_count} AMD accelerator(s) but amdsmi "
-                    f"enumerates {smi_count} — a GPU may have dropped off the driver"
+                    f"enumerates {smi_count} — a GPU may have dropped off the bus"
                 ),
                 extra_info=extra,
                 suggested_actions=SuggestedActions(
                     description="PCI/driver GPU count mismatch",
                     repair_actions=[RepairActionType.REBOOT_SYSTEM],
                 ),
+            )
+        if pci_count is not None and smi_count and pci_count > smi_count:
+            # more on the bus than the driver exposes: either a GPU fell off
+            # the driver, or this process runs in a container whose device
+            # cgroup hides some render nodes while sysfs shows the host bus.
+            # Without expected_gpu_count we cannot tell, so flag Degraded;
+            # set expected_gpu_count for a hard rule.
+            return CheckResult(
+                NAME,
+                health=HealthStateType.DEGRADED,
+                reason=(
+                    f"lspci sees {pci_count} AMD accelerator(s) but amdsmi "
+                    f"exposes {smi_count} — GPU missing from the driver, or "
+                    "container device visibility is restricted"
+                ),
+                extra_info=extra,
             )
         return CheckResult(
             NAME,

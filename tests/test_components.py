@@ -167,9 +167,12 @@ def test_gpu_counts_with_expected(mock_core):
     assert cr.health == HealthStateType.UNHEALTHY
     assert "expected 8" in cr.reason
     comp.expected = 4
-    comp.count_lspci = lambda: 3  # pci/driver mismatch
+    comp.count_lspci = lambda: 3  # bus shows fewer than the driver: hard fail
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.UNHEALTHY
+    comp.count_lspci = lambda: 8  # bus shows more: container visibility case
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
 
 
 def test_error_ras_state_machine(mock_core):
