@@ -128,6 +128,11 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
                     ],
                 ),
             )
+        from ... import smi as smi_pkg
+
+        if smi_pkg.mock_enabled():
+            # mock backend: the real PCI bus is unrelated to the mocked GPUs
+            pci_count = None
         if pci_count is not None and smi_count and pci_count < smi_count:
             # fewer devices on the bus than the driver claims — real trouble
             return CheckResult(

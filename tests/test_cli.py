@@ -1,0 +1,82 @@
+"""CLI tests via typer's CliRunner (reference: cmd/gpud command tests)."""
+
+import json
+import os
+
+import pytest
+from typer.testing import CliRunner
+
+from gpud_amd.cli.main import app
+
+runner = CliRunner()
+
+
+@pytest.fixture(autouse=True)
+def mock_smi(monkeypatch):
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK_GPUS", "2")
+
+
+def test_version():
+    res = runner.invoke(app, ["version"])
+    assert res.exit_code == 0
+    assert "gpud-amd" in res.output
+
+
+def test_scan_mock():
+    res = runner.invoke(app, ["scan", "--mock"])
+    assert res.exit_code == 0, res.output
+    assert "accelerator-amd-temperature" in res.output
+    assert "overall: Healthy" in res.output
+
+
+def test_scan_mock_expected_count_mismatch():
+    res = runner.invoke(app, ["scan", "--mock", "--expected-gpu-count", "8"])
+    assert res.exit_code == 1
+    assert "Unhealthy" in res.output
+
+
+def test_machine_info_mock():
+    res = runner.invoke(app, ["machine-info", "--mock"])
+    assert res.exit_code == 0, res.output
+    mi = json.loads(res.output)
+    assert mi["gpuInfo"]["product"] == "AMD Instinct MI355X"
+
+
+def test_compact_and_metadata(tmp_path):
+    from gpud_amd.pkg import metadata
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.sqlite_util import open_rw
+
+    cfg = Config(data_dir=str(tmp_path))
+    conn = open_rw(cfg.state_path)
+    metadata.create_table(conn)
+    metadata.set_value(conn, metadata.KEY_MACHINE_ID, "m-42")
+    metadata.set_value(conn, metadata.KEY_TOKEN, "secret")
+    conn.close()
+    res = runner.invoke(app, ["compact", "--data-dir", str(tmp_path)])
+    assert res.exit_code == 0 and "compacted" in res.output
+    res = runner.invoke(app, ["metadata", "--data-dir", str(tmp_path)])
+    assert res.exit_code == 0
+    assert "m-42" in res.output
+    assert "secret" not in res.output  # token redacted
+
+
+def test_list_plugins_and_run_group(tmp_path):
+    specs = tmp_path / "plugins.yaml"
+    specs.write_text(
+        """
+- plugin_name: quick
+  plugin_type: component
+  tags: [grp]
+  health_state_plugin:
+    steps:
+      - run_bash_script:
+          script: echo fine
+"""
+    )
+    res = runner.invoke(app, ["list-plugins", str(specs)])
+    assert res.exit_code == 0 and "quick" in res.output
+    res = runner.invoke(app, ["run-plugin-group", str(specs), "--tag", "grp"])
+    assert res.exit_code == 0
+    assert "custom-plugin-quick: Healthy" in res.output
