@@ -61,6 +61,11 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
     def _cached_lspci(self) -> Optional[int]:
         import time
 
+        from ... import smi as smi_pkg
+
+        if smi_pkg.mock_enabled():
+            # mock backend: the real PCI bus is unrelated to the mocked GPUs
+            return None
         now = time.monotonic()
         if now - self._lspci_cached_at > self.LSPCI_TTL_SECONDS:
             self._lspci_cache = count_amd_gpus_lspci(self._lspci_command)
@@ -128,11 +133,6 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
                     ],
                 ),
             )
-        from ... import smi as smi_pkg
-
-        if smi_pkg.mock_enabled():
-            # mock backend: the real PCI bus is unrelated to the mocked GPUs
-            pci_count = None
         if pci_count is not None and smi_count and pci_count < smi_count:
             # fewer devices on the bus than the driver claims — real trouble
             return CheckResult(
