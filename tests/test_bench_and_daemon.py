@@ -1,0 +1,135 @@
+"""bench.py contract and full daemon boot via the CLI entry point.
+
+The round-end driver launches bench.py through torch.distributed.run with
+one rank per GPU; the 2-rank gloo mock run here exercises exactly that path
+on CPU (init_process_group, barriers, MAX all-reduce, single JSON line from
+rank 0).
+"""
+
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def test_bench_single_rank_mock():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--mock", "--steps", "20", "--warmup", "2"],
+        capture_output=True,
+        text=True,
+        timeout=180,
+        cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-800:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["metric"] == "poll_cycle_p50_ms"
+    assert d["higher_is_better"] is False
+    assert d["scaling"] == "weak"
+    assert d["n_gpus"] == 1
+    assert d["steps"] == 20 and d["warmup"] == 2
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["data"] == "synthetic"
+
+
+def test_bench_two_ranks_gloo_mock():
+    port = _free_port()
+    out = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "torch.distributed.run",
+            "--nnodes=1",
+            "--nproc-per-node",
+            "2",
+            "--master-addr",
+            "127.0.0.1",
+            "--master-port",
+            str(port),
+            "bench.py",
+            "--mock",
+            "--gpus",
+            "2",
+            "--steps",
+            "10",
+            "--warmup",
+            "2",
+        ],
+        capture_output=True,
+        text=True,
+        timeout=300,
+        cwd=REPO,
+        env={**os.environ, "MASTER_ADDR": "127.0.0.1"},
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    json_lines = [
+        l for l in out.stdout.strip().splitlines() if l.startswith("{")
+    ]
+    assert len(json_lines) == 1, "exactly one JSON line from rank 0"
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 2
+    assert d["value"] > 0
+
+
+def test_daemon_boot_via_cli():
+    """Boot the full daemon via `python -m gpud_amd run` and poke it."""
+    port = _free_port()
+    env = {
+        **os.environ,
+        "GPUD_AMDSMI_MOCK": "1",
+        "GPUD_AMDSMI_MOCK_GPUS": "2",
+        "PYTHONPATH": REPO,
+    }
+    proc = subprocess.Popen(
+        [
+            sys.executable,
+            "-m",
+            "gpud_amd",
+            "run",
+            "--in-memory-db",
+            "--address",
+            f"127.0.0.1:{port}",
+            "--log-level",
+            "warning",
+        ],
+        cwd=REPO,
+        env=env,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(30), "daemon did not become healthy"
+        comps = client.get_components()
+        assert "accelerator-amd-temperature" in comps
+        states = client.get_health_states(components=["cpu"])
+        assert states["cpu"][0].health in ("Healthy", "Degraded", "Initializing")
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            os.killpg(proc.pid, signal.SIGKILL)
+            proc.wait(timeout=5)
