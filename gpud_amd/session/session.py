@@ -374,3 +374,43 @@ class Session:
 
     def _m_delete(self, payload: dict) -> dict:
         return self._m_logout(payload)
+
+    # -- kap-mTLS (reference: pkg/kapmtls + session methods) ------------------
+
+    def _kapmtls(self):
+        from ..pkg.kapmtls import Manager
+
+        import os
+
+        return Manager(os.path.join(self.core.config.data_dir, "kapmtls"))
+
+    def _m_kapMTLSStatus(self, payload: dict) -> dict:
+        return self._kapmtls().status()
+
+    def _m_updateKAPMTLSCredentials(self, payload: dict) -> dict:
+        cert = base64.b64decode(payload.get("cert", ""))
+        key = base64.b64decode(payload.get("key", ""))
+        if not cert or not key:
+            return {"error": "cert and key (base64) required"}
+        version = self._kapmtls().stage(cert, key, payload.get("version", ""))
+        return {"staged_version": version}
+
+    def _m_activateKAPMTLS(self, payload: dict) -> dict:
+        mgr = self._kapmtls()
+        err = mgr.activate(payload.get("version", ""))
+        if err:
+            return {"error": err}
+        return {"active_version": mgr.active_version()}
+
+    def _m_nodeCredentials(self, payload: dict) -> dict:
+        if self.core.db_ro is None:
+            return {"error": "no metadata store"}
+        return {
+            "machine_id": metadata.get_value(
+                self.core.db_ro, metadata.KEY_MACHINE_ID
+            ),
+            "endpoint": metadata.get_value(self.core.db_ro, metadata.KEY_ENDPOINT),
+            "has_token": bool(
+                metadata.get_value(self.core.db_ro, metadata.KEY_TOKEN)
+            ),
+        }
