@@ -21,6 +21,7 @@
 
 #include <chrono>
 #include <cstring>
+#include <map>
 #include <mutex>
 #include <stdexcept>
 #include <string>
@@ -544,24 +545,23 @@ py::dict energy_count(int index) {
 
 struct Snapshot {
   // filled flags let Python distinguish "not supported" from zero
-  bool ok_temp = false, ok_power = false, ok_clock_gfx = false,
-       ok_clock_mem = false, ok_activity = false, ok_vram = false,
-       ok_ecc = false, ok_throttle = false, ok_xgmi = false, ok_metrics = false;
+  bool ok_temp = false, ok_power = false, ok_clock = false,
+       ok_activity = false, ok_vram = false, ok_ecc = false,
+       ok_throttle = false, ok_xgmi = false, ok_metrics = false;
   int64_t temp_edge = 0, temp_hotspot = 0, temp_vram = 0;
   int64_t temp_edge_limit = 0, temp_hotspot_limit = 0, temp_vram_limit = 0;
   int64_t temp_hotspot_shutdown = 0;
-  amdsmi_power_info_t power;
-  amdsmi_power_cap_info_t power_cap;
-  bool ok_power_cap = false;
-  amdsmi_clk_info_t clk_gfx, clk_mem;
-  amdsmi_engine_usage_t act;
+  uint32_t power_w = 0, avg_power_w = 0, power_limit_w = 0;
+  uint64_t power_cap_uw = 0;
+  uint16_t gfx_mhz = 0, mem_mhz = 0, gfx_max_mhz = 0, mem_max_mhz = 0;
+  uint16_t gfx_activity = 0, umc_activity = 0, mm_activity = 0;
   amdsmi_vram_usage_t vram;
   amdsmi_error_count_t ecc;
-  amdsmi_xgmi_link_status_t xgmi;
+  uint32_t xgmi_total_links = 0;
+  int xgmi_states[AMDSMI_MAX_NUM_XGMI_LINKS] = {0};
   int xgmi_err = -1;
   uint32_t throttle_status = 0;
   uint64_t indep_throttle_status = 0;
-  uint16_t cur_gfxclk = 0, cur_uclk = 0, avg_socket_power = 0;
   bool ok_bad_pages = false;
   uint32_t bp_total = 0, bp_reserved = 0, bp_pending = 0, bp_unreservable = 0;
   uint32_t bp_threshold = 0;
@@ -571,43 +571,155 @@ struct Snapshot {
            thr_vr = 0, thr_hbm = 0;
 };
 
-void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
-  // temperatures (current + limits) — some SKUs/partitions expose only a
-  // subset of sensors (observed: no EDGE on an MI355X 0x75a3 box), so the
-  // block is reported when ANY current-temperature sensor answers
-  bool ok_edge =
-      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_EDGE, AMDSMI_TEMP_CURRENT,
-                             &s.temp_edge) == AMDSMI_STATUS_SUCCESS;
-  bool ok_hot =
-      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
-                             AMDSMI_TEMP_CURRENT,
-                             &s.temp_hotspot) == AMDSMI_STATUS_SUCCESS;
-  bool ok_vram =
-      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_VRAM, AMDSMI_TEMP_CURRENT,
-                             &s.temp_vram) == AMDSMI_STATUS_SUCCESS;
-  s.ok_temp = ok_edge || ok_hot || ok_vram;
-  if (!ok_edge && ok_hot) s.temp_edge = s.temp_hotspot;
+// Static per-device values (temperature limits, power caps, max clocks, the
+// bad-page threshold) never change at runtime — they are fetched once per
+// handle and cached, keeping the per-cycle snapshot down to the dynamic
+// ioctls only (~1 ms/GPU instead of ~3 ms; amdsmi_get_violation_status at
+// ~100 ms stays off this path entirely, see NOTE below).
+struct StaticInfo {
+  bool filled = false;
+  int64_t temp_edge_limit = 0, temp_hotspot_limit = 0, temp_vram_limit = 0;
+  int64_t temp_hotspot_shutdown = 0;
+  uint32_t power_limit_w = 0;
+  uint64_t power_cap_uw = 0;
+  uint16_t gfx_max_mhz = 0, mem_max_mhz = 0;
+  uint32_t bp_threshold = 0;
+  bool has_bp_threshold = false;
+};
+
+std::mutex g_static_mu;
+std::map<amdsmi_processor_handle, StaticInfo> g_static;
+
+const StaticInfo& static_info_for(amdsmi_processor_handle h) {
+  {
+    std::lock_guard<std::mutex> lk(g_static_mu);
+    auto it = g_static.find(h);
+    if (it != g_static.end() && it->second.filled) return it->second;
+  }
+  StaticInfo si;
   amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_EDGE, AMDSMI_TEMP_CRITICAL,
-                         &s.temp_edge_limit);
-  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT, AMDSMI_TEMP_CRITICAL,
-                         &s.temp_hotspot_limit);
+                         &si.temp_edge_limit);
+  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
+                         AMDSMI_TEMP_CRITICAL, &si.temp_hotspot_limit);
   amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_VRAM, AMDSMI_TEMP_CRITICAL,
-                         &s.temp_vram_limit);
-  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT, AMDSMI_TEMP_SHUTDOWN,
-                         &s.temp_hotspot_shutdown);
-  std::memset(&s.power, 0, sizeof(s.power));
-  s.ok_power = amdsmi_get_power_info(h, &s.power) == AMDSMI_STATUS_SUCCESS;
-  std::memset(&s.power_cap, 0, sizeof(s.power_cap));
-  s.ok_power_cap =
-      amdsmi_get_power_cap_info(h, 0, &s.power_cap) == AMDSMI_STATUS_SUCCESS;
-  std::memset(&s.clk_gfx, 0, sizeof(s.clk_gfx));
-  s.ok_clock_gfx = amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_GFX, &s.clk_gfx) ==
-                   AMDSMI_STATUS_SUCCESS;
-  std::memset(&s.clk_mem, 0, sizeof(s.clk_mem));
-  s.ok_clock_mem = amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_MEM, &s.clk_mem) ==
-                   AMDSMI_STATUS_SUCCESS;
-  std::memset(&s.act, 0, sizeof(s.act));
-  s.ok_activity = amdsmi_get_gpu_activity(h, &s.act) == AMDSMI_STATUS_SUCCESS;
+                         &si.temp_vram_limit);
+  amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
+                         AMDSMI_TEMP_SHUTDOWN, &si.temp_hotspot_shutdown);
+  amdsmi_power_cap_info_t cap;
+  std::memset(&cap, 0, sizeof(cap));
+  if (amdsmi_get_power_cap_info(h, 0, &cap) == AMDSMI_STATUS_SUCCESS)
+    si.power_cap_uw = cap.power_cap;
+  amdsmi_power_info_t pw;
+  std::memset(&pw, 0, sizeof(pw));
+  if (amdsmi_get_power_info(h, &pw) == AMDSMI_STATUS_SUCCESS)
+    si.power_limit_w = pw.power_limit;
+  amdsmi_clk_info_t ci;
+  std::memset(&ci, 0, sizeof(ci));
+  if (amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_GFX, &ci) ==
+      AMDSMI_STATUS_SUCCESS)
+    si.gfx_max_mhz = (uint16_t)ci.max_clk;
+  std::memset(&ci, 0, sizeof(ci));
+  if (amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_MEM, &ci) ==
+      AMDSMI_STATUS_SUCCESS)
+    si.mem_max_mhz = (uint16_t)ci.max_clk;
+  si.has_bp_threshold = amdsmi_get_gpu_bad_page_threshold(
+                            h, &si.bp_threshold) == AMDSMI_STATUS_SUCCESS;
+  si.filled = true;
+  std::lock_guard<std::mutex> lk(g_static_mu);
+  auto& slot = g_static[h];
+  slot = si;
+  return slot;
+}
+
+inline bool gm_valid16(uint16_t v) { return v != 0 && v != 0xFFFF; }
+
+void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
+  const StaticInfo& si = static_info_for(h);
+  s.temp_edge_limit = si.temp_edge_limit;
+  s.temp_hotspot_limit = si.temp_hotspot_limit;
+  s.temp_vram_limit = si.temp_vram_limit;
+  s.temp_hotspot_shutdown = si.temp_hotspot_shutdown;
+  s.power_limit_w = si.power_limit_w;
+  s.power_cap_uw = si.power_cap_uw;
+  s.gfx_max_mhz = si.gfx_max_mhz;
+  s.mem_max_mhz = si.mem_max_mhz;
+
+  // ONE gpu_metrics ioctl covers temps, activity, power, current clocks,
+  // throttle residencies and xGMI link state
+  amdsmi_gpu_metrics_t gm;
+  std::memset(&gm, 0, sizeof(gm));
+  if (amdsmi_get_gpu_metrics_info(h, &gm) == AMDSMI_STATUS_SUCCESS) {
+    s.ok_metrics = true;
+    s.throttle_status = gm.throttle_status;
+    s.indep_throttle_status = gm.indep_throttle_status;
+    if (gm_valid16(gm.temperature_hotspot) || gm_valid16(gm.temperature_edge) ||
+        gm_valid16(gm.temperature_mem)) {
+      s.ok_temp = true;
+      s.temp_edge = gm_valid16(gm.temperature_edge) ? gm.temperature_edge
+                                                    : gm.temperature_hotspot;
+      s.temp_hotspot = gm.temperature_hotspot;
+      s.temp_vram = gm.temperature_mem;
+    }
+    s.ok_activity = true;
+    s.gfx_activity = gm.average_gfx_activity;
+    s.umc_activity = gm.average_umc_activity;
+    s.mm_activity = gm.average_mm_activity;
+    s.ok_power = true;
+    s.power_w = gm_valid16(gm.current_socket_power)
+                    ? gm.current_socket_power
+                    : gm.average_socket_power;
+    s.avg_power_w = gm.average_socket_power;
+    s.ok_clock = true;
+    s.gfx_mhz = gm_valid16(gm.current_gfxclk) ? gm.current_gfxclk
+                                              : gm.average_gfxclk_frequency;
+    s.mem_mhz = gm_valid16(gm.current_uclk) ? gm.current_uclk
+                                            : gm.average_uclk_frequency;
+    s.ok_throttle = true;
+    s.thr_acc_counter = gm.accumulation_counter;
+    s.thr_prochot = gm.prochot_residency_acc;
+    s.thr_ppt = gm.ppt_residency_acc;
+    s.thr_socket = gm.socket_thm_residency_acc;
+    s.thr_vr = gm.vr_thm_residency_acc;
+    s.thr_hbm = gm.hbm_thm_residency_acc;
+    // xGMI link state straight from the metrics table (up/down per link)
+    uint32_t nlinks = 0;
+    for (uint32_t i = 0; i < AMDSMI_MAX_NUM_XGMI_LINKS; ++i) {
+      uint16_t st = gm.xgmi_link_status[i];
+      if (st == 0xFFFF) break;
+      s.xgmi_states[nlinks++] = (int)st;
+    }
+    if (nlinks > 0) {
+      s.ok_xgmi = true;
+      s.xgmi_total_links = nlinks;
+    }
+  }
+  // fallbacks for boards whose metrics table lacks a field
+  if (!s.ok_temp) {
+    int64_t t = 0;
+    if (amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_HOTSPOT,
+                               AMDSMI_TEMP_CURRENT, &t) ==
+        AMDSMI_STATUS_SUCCESS) {
+      s.ok_temp = true;
+      s.temp_hotspot = t;
+      s.temp_edge = t;
+      amdsmi_get_temp_metric(h, AMDSMI_TEMPERATURE_TYPE_VRAM,
+                             AMDSMI_TEMP_CURRENT, &s.temp_vram);
+    }
+  }
+  if (!s.ok_xgmi) {
+    amdsmi_xgmi_link_status_t xs;
+    std::memset(&xs, 0, sizeof(xs));
+    if (amdsmi_get_gpu_xgmi_link_status(h, &xs) == AMDSMI_STATUS_SUCCESS) {
+      s.ok_xgmi = true;
+      s.xgmi_total_links = xs.total_links;
+      for (uint32_t i = 0;
+           i < xs.total_links && i < AMDSMI_MAX_NUM_XGMI_LINKS; ++i)
+        s.xgmi_states[i] = (int)xs.status[i];
+    }
+  }
+  amdsmi_xgmi_status_t xe;
+  if (amdsmi_gpu_xgmi_error_status(h, &xe) == AMDSMI_STATUS_SUCCESS)
+    s.xgmi_err = static_cast<int>(xe);
   std::memset(&s.vram, 0, sizeof(s.vram));
   s.ok_vram = amdsmi_get_gpu_vram_usage(h, &s.vram) == AMDSMI_STATUS_SUCCESS;
   std::memset(&s.ecc, 0, sizeof(s.ecc));
@@ -615,15 +727,10 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
       amdsmi_get_gpu_total_ecc_count(h, &s.ecc) == AMDSMI_STATUS_SUCCESS;
   // NOTE: amdsmi_get_violation_status is NOT called here — it blocks ~100 ms
   // per GPU (double-samples internally to compute per_* rates). The same
-  // throttle residency accumulators come from gpu_metrics below at ~0.15 ms;
-  // the throttle component derives activity from deltas across its own
-  // polls. The explicit violation_status() getter remains for manual diags.
-  std::memset(&s.xgmi, 0, sizeof(s.xgmi));
-  s.ok_xgmi =
-      amdsmi_get_gpu_xgmi_link_status(h, &s.xgmi) == AMDSMI_STATUS_SUCCESS;
-  amdsmi_xgmi_status_t xe;
-  if (amdsmi_gpu_xgmi_error_status(h, &xe) == AMDSMI_STATUS_SUCCESS)
-    s.xgmi_err = static_cast<int>(xe);
+  // throttle residency accumulators come from gpu_metrics above; the
+  // throttle component derives activity from deltas across its own polls.
+  // The explicit violation_status() getter remains for manual diags.
+
   // bad pages (retired HBM rows — the remapped-rows analog)
   uint32_t bp_num = 0;
   if (amdsmi_get_gpu_bad_page_info(h, &bp_num, nullptr) ==
@@ -643,26 +750,8 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
         }
       }
     }
-    s.ok_bp_threshold = amdsmi_get_gpu_bad_page_threshold(h, &s.bp_threshold) ==
-                        AMDSMI_STATUS_SUCCESS;
-  }
-  // gpu_metrics: throttle bits + current clocks in one ioctl
-  amdsmi_gpu_metrics_t gm;
-  std::memset(&gm, 0, sizeof(gm));
-  if (amdsmi_get_gpu_metrics_info(h, &gm) == AMDSMI_STATUS_SUCCESS) {
-    s.ok_metrics = true;
-    s.throttle_status = gm.throttle_status;
-    s.indep_throttle_status = gm.indep_throttle_status;
-    s.cur_gfxclk = gm.current_gfxclk;
-    s.cur_uclk = gm.current_uclk;
-    s.avg_socket_power = gm.average_socket_power;
-    s.ok_throttle = true;
-    s.thr_acc_counter = gm.accumulation_counter;
-    s.thr_prochot = gm.prochot_residency_acc;
-    s.thr_ppt = gm.ppt_residency_acc;
-    s.thr_socket = gm.socket_thm_residency_acc;
-    s.thr_vr = gm.vr_thm_residency_acc;
-    s.thr_hbm = gm.hbm_thm_residency_acc;
+    s.ok_bp_threshold = si.has_bp_threshold;
+    s.bp_threshold = si.bp_threshold;
   }
 }
 
@@ -681,31 +770,26 @@ py::dict snapshot_to_dict(const Snapshot& s) {
   }
   if (s.ok_power) {
     py::dict p;
-    p["socket_power_w"] = s.power.socket_power;
-    p["current_socket_power_w"] = s.power.current_socket_power;
-    p["average_socket_power_w"] = s.power.average_socket_power;
-    p["power_limit_w"] = s.power.power_limit;
-    if (s.ok_power_cap) p["power_cap_uw"] = s.power_cap.power_cap;
+    p["socket_power_w"] = s.power_w;
+    p["current_socket_power_w"] = s.power_w;
+    p["average_socket_power_w"] = s.avg_power_w;
+    p["power_limit_w"] = s.power_limit_w;
+    if (s.power_cap_uw) p["power_cap_uw"] = s.power_cap_uw;
     d["power"] = p;
   }
-  if (s.ok_clock_gfx || s.ok_clock_mem) {
+  if (s.ok_clock) {
     py::dict c;
-    if (s.ok_clock_gfx) {
-      c["gfx_mhz"] = s.clk_gfx.clk;
-      c["gfx_max_mhz"] = s.clk_gfx.max_clk;
-      c["gfx_deep_sleep"] = static_cast<int>(s.clk_gfx.clk_deep_sleep);
-    }
-    if (s.ok_clock_mem) {
-      c["mem_mhz"] = s.clk_mem.clk;
-      c["mem_max_mhz"] = s.clk_mem.max_clk;
-    }
+    c["gfx_mhz"] = s.gfx_mhz;
+    c["gfx_max_mhz"] = s.gfx_max_mhz;
+    c["mem_mhz"] = s.mem_mhz;
+    c["mem_max_mhz"] = s.mem_max_mhz;
     d["clock"] = c;
   }
   if (s.ok_activity) {
     py::dict a;
-    a["gfx_activity_pct"] = s.act.gfx_activity;
-    a["umc_activity_pct"] = s.act.umc_activity;
-    a["mm_activity_pct"] = s.act.mm_activity;
+    a["gfx_activity_pct"] = s.gfx_activity;
+    a["umc_activity_pct"] = s.umc_activity;
+    a["mm_activity_pct"] = s.mm_activity;
     d["activity"] = a;
   }
   if (s.ok_vram) {
@@ -735,11 +819,10 @@ py::dict snapshot_to_dict(const Snapshot& s) {
   }
   if (s.ok_xgmi) {
     py::dict x;
-    x["total_links"] = s.xgmi.total_links;
+    x["total_links"] = s.xgmi_total_links;
     py::list states;
-    for (uint32_t i = 0;
-         i < s.xgmi.total_links && i < AMDSMI_MAX_NUM_XGMI_LINKS; ++i)
-      states.append(static_cast<int>(s.xgmi.status[i]));
+    for (uint32_t i = 0; i < s.xgmi_total_links; ++i)
+      states.append(s.xgmi_states[i]);
     x["states"] = states;
     d["xgmi_link_status"] = x;
   }
@@ -757,9 +840,9 @@ py::dict snapshot_to_dict(const Snapshot& s) {
     py::dict m;
     m["throttle_status"] = s.throttle_status;
     m["indep_throttle_status"] = s.indep_throttle_status;
-    m["current_gfxclk_mhz"] = s.cur_gfxclk;
-    m["current_uclk_mhz"] = s.cur_uclk;
-    m["average_socket_power_w"] = s.avg_socket_power;
+    m["current_gfxclk_mhz"] = s.gfx_mhz;
+    m["current_uclk_mhz"] = s.mem_mhz;
+    m["average_socket_power_w"] = s.avg_power_w;
     d["gpu_metrics"] = m;
   }
   return d;
