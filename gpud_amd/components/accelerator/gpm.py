@@ -61,10 +61,24 @@ class GPMComponent(TickerComponent, SmiComponentMixin):
                     float(gm.get("current_uclk_mhz", 0)),
                     uuid=uuid,
                 )
-        # per-process CU occupancy (the SM-occupancy analog on 256-CU CDNA4)
+        # per-process CU occupancy (the SM-occupancy analog on 256-CU CDNA4);
+        # process lists are memoized per cycle in the shared cache
         try:
-            for uuid, dev in self.get_devices().items():
-                procs = dev.process_list()
+            def fetch():
+                out = {}
+                for uuid, dev in self.get_devices().items():
+                    try:
+                        out[uuid] = dev.process_list()
+                    except Exception:
+                        out[uuid] = []
+                return out
+
+            plists = (
+                self._shared.get_aux("process_list", fetch)
+                if self._shared is not None
+                else fetch()
+            )
+            for uuid, procs in plists.items():
                 occ = sum(int(p.get("cu_occupancy", 0)) for p in procs)
                 self._gauges.set(
                     "accelerator_amd_gpm_cu_occupancy",

@@ -20,11 +20,26 @@ class ProcessesComponent(TickerComponent, SmiComponentMixin):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
         self._smi = inst.smi
+        self._shared = inst.shared_snapshots
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
         # test seam: devices getter
         self.get_devices: Callable = (
             self._smi.devices if self._smi is not None else dict
         )
+
+    def _process_lists(self):
+        def fetch():
+            out = {}
+            for uuid, dev in self.get_devices().items():
+                try:
+                    out[uuid] = dev.process_list()
+                except Exception:
+                    out[uuid] = []
+            return out
+
+        if self._shared is not None:
+            return self._shared.get_aux("process_list", fetch)
+        return fetch()
 
     @property
     def name(self) -> str:
@@ -42,11 +57,7 @@ class ProcessesComponent(TickerComponent, SmiComponentMixin):
             return guard
         total = 0
         extra = {}
-        for uuid, dev in self.get_devices().items():
-            try:
-                procs = dev.process_list()
-            except Exception:
-                procs = []
+        for uuid, procs in self._process_lists().items():
             total += len(procs)
             self._gauges.set(
                 "accelerator_amd_processes_count",

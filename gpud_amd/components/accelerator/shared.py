@@ -25,6 +25,10 @@ class SharedSnapshots:
         self._lock = threading.Lock()
         self._snapshots: Dict[str, Dict[str, Any]] = {}
         self._taken_at: float = 0.0
+        # per-cycle memo for heavier SMI calls several components need
+        # (process_list is used by both processes and gpm; link_metrics by
+        # xgmi) — cleared on every refresh so data stays cycle-fresh
+        self._aux: Dict[str, Any] = {}
 
     def get(self, max_age: Optional[float] = None) -> Dict[str, Dict[str, Any]]:
         """Snapshots keyed by uuid, refreshed when older than the TTL."""
@@ -33,14 +37,26 @@ class SharedSnapshots:
             now = time.monotonic()
             if now - self._taken_at > ttl:
                 self._snapshots = self.smi.snapshot_all()
+                self._aux = {}
                 self._taken_at = time.monotonic()
             return self._snapshots
 
     def refresh(self) -> Dict[str, Dict[str, Any]]:
         with self._lock:
             self._snapshots = self.smi.snapshot_all()
+            self._aux = {}
             self._taken_at = time.monotonic()
             return self._snapshots
+
+    def get_aux(self, key: str, fn) -> Any:
+        """Memoize ``fn()`` for the current poll cycle under ``key``."""
+        with self._lock:
+            if key in self._aux:
+                return self._aux[key]
+        val = fn()
+        with self._lock:
+            self._aux[key] = val
+        return val
 
 
 class SmiComponentMixin:
