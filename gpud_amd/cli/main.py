@@ -107,7 +107,36 @@ def run(
         session = Session(core, endpoint=endpoint, token=token)
         session.start()
 
+    # package manager reconcile loops (reference: cmd/gpud/run:425-431)
+    from ..pkg.gpud_manager import PackageController
+
+    pkg_controller = PackageController(cfg)
+    pkg_controller.start()
+
     stop = {"flag": False}
+
+    # version-file-triggered update+exit (reference: server.go:815
+    # updateFromVersionFile — systemd Restart=always picks up the new build)
+    def _version_watch():
+        from ..pkg.update import check_version_file
+
+        while not stop["flag"]:
+            time.sleep(60.0)
+            try:
+                pending = check_version_file(cfg)
+            except Exception:
+                continue
+            if pending:
+                typer.echo(
+                    f"target version {pending} requested; exiting for restart",
+                    err=True,
+                )
+                stop["flag"] = True
+                return
+
+    import threading as _threading
+
+    _threading.Thread(target=_version_watch, daemon=True).start()
 
     def _sig(_s, _f):
         stop["flag"] = True
@@ -118,6 +147,7 @@ def run(
         while not stop["flag"]:
             time.sleep(0.5)
     finally:
+        pkg_controller.stop()
         if session is not None:
             session.stop()
         server.stop()
@@ -285,6 +315,62 @@ def run_plugin_group(
             if cr.health != "Healthy":
                 failed += 1
     raise typer.Exit(code=1 if failed else 0)
+
+
+@app.command()
+def notify(
+    event: str = typer.Argument(..., help="startup | shutdown"),
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR),
+):
+    """Notify the control plane of daemon startup/shutdown
+    (reference: gpud notify)."""
+    import httpx
+
+    from ..pkg import metadata as md
+    from ..pkg.sqlite_util import open_ro
+
+    cfg = Config(data_dir=data_dir)
+    try:
+        conn = open_ro(cfg.state_path)
+        endpoint = md.get_value(conn, md.KEY_ENDPOINT)
+        machine_id = md.get_value(conn, md.KEY_MACHINE_ID)
+        token = md.get_value(conn, md.KEY_TOKEN)
+        conn.close()
+    except Exception as e:
+        typer.echo(f"cannot read metadata: {e}", err=True)
+        raise typer.Exit(code=1)
+    if not endpoint:
+        typer.echo("no control-plane endpoint configured; nothing to notify")
+        raise typer.Exit(code=0)
+    try:
+        r = httpx.post(
+            endpoint.rstrip("/") + "/api/v1/notify",
+            json={"machineID": machine_id, "event": event},
+            headers={"token": token},
+            timeout=10,
+            verify=False,
+        )
+        typer.echo(f"notify {event}: HTTP {r.status_code}")
+    except httpx.HTTPError as e:
+        typer.echo(f"notify failed: {e}", err=True)
+        raise typer.Exit(code=1)
+
+
+@app.command()
+def update(
+    version: str = typer.Argument(...),
+    data_dir: str = typer.Option(DEFAULT_DATA_DIR),
+    base_url: str = typer.Option("https://pkg.gpud.dev/packages"),
+):
+    """Self-update to a version (reference: gpud update)."""
+    from ..pkg.update import update_to_version
+
+    cfg = Config(data_dir=data_dir)
+    err = update_to_version(cfg, version, base_url=base_url)
+    if err:
+        typer.echo(f"update failed: {err}", err=True)
+        raise typer.Exit(code=1)
+    typer.echo(f"updated to {version}")
 
 
 @app.command()
