@@ -27,6 +27,7 @@
 // Built for gfx950 only — no multi-arch dispatch (csrc/build.sh).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <pybind11/pybind11.h>
 
 #include <cstdint>
@@ -162,6 +163,273 @@ py::dict mfma_stress_bf16(int iters, int workgroups) {
 }
 py::dict mfma_stress_fp8(int iters, int workgroups) {
   return run_mfma_stress(mfma_stress_fp8_kernel, iters, workgroups, "fp8_e4m3");
+}
+
+// ---------------------------------------------------------------------------
+// MX-scaled fp8 stress — the only path to the ~5 PF dense fp8 peak on CDNA4
+// (non-scaled fp8 runs at the bf16 rate; cdna_hip_programming.md §3/§4:
+// mfma_scale_*_f8f6f4 with per-32-element E8M0 block scales, K=64 for the
+// 32x32 shape). All-ones fp8 (0x38) with unit scales (E8M0 0x7F = 2^0)
+// keeps the exact-verification property: each MFMA adds K=64 per element.
+// ---------------------------------------------------------------------------
+
+typedef int i32x8 __attribute__((ext_vector_type(8)));
+
+__global__ __launch_bounds__(256, 4) void mfma_stress_mxfp8_kernel(
+    float* __restrict__ out, int iters) {
+  i32x8 a, b;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    a[i] = 0x38383838;  // 4x fp8 e4m3 1.0
+    b[i] = 0x38383838;
+  }
+  const int unit_scale = 0x7F7F7F7F;  // E8M0 exponent-bias 127 = 1.0
+  f32x16 acc[kAccums] = {};
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int u = 0; u < kInnerUnroll; ++u) {
+#pragma unroll
+      for (int j = 0; j < kAccums; ++j) {
+        acc[j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a, b, acc[j], /*cbsz fp8*/ 0, /*blgp fp8*/ 0,
+            /*opsel_a*/ 0, unit_scale, /*opsel_b*/ 0, unit_scale);
+      }
+    }
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < kAccums; ++j) s += acc[j][0];
+  out[blockIdx.x * blockDim.x + threadIdx.x] = s;
+}
+
+py::dict mfma_stress_mxfp8(int iters, int workgroups) {
+  if (iters <= 0 || iters > (1 << 17)) throw std::invalid_argument("iters");
+  const int threads = 256;
+  float* d_out = nullptr;
+  const size_t out_elems = (size_t)workgroups * threads;
+  HIP_CHECK(hipMalloc(&d_out, out_elems * sizeof(float)));
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(mfma_stress_mxfp8_kernel, dim3(workgroups), dim3(threads),
+                     0, 0, d_out, 16);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  hipLaunchKernelGGL(mfma_stress_mxfp8_kernel, dim3(workgroups), dim3(threads),
+                     0, 0, d_out, iters);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  std::vector<float> host(out_elems);
+  HIP_CHECK(hipMemcpy(host.data(), d_out, out_elems * sizeof(float),
+                      hipMemcpyDeviceToHost));
+  const double expect = (double)kAccums * 64.0 * iters * kInnerUnroll;
+  size_t bad = 0;
+  for (size_t i = 0; i < out_elems; ++i)
+    if (host[i] != (float)expect) bad++;
+  HIP_CHECK(hipFree(d_out));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double waves = (double)workgroups * threads / 64.0;
+  const double mfmas = waves * (double)iters * kInnerUnroll * kAccums;
+  const double flops = mfmas * 2.0 * 32 * 32 * 64;
+  py::dict d;
+  d["dtype"] = "mxfp8_e4m3";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds"] = ms * 1e-3;
+  d["flops"] = flops;
+  d["workgroups"] = workgroups;
+  d["iters"] = iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
+// ---------------------------------------------------------------------------
+// GEMM-shaped stress: C[M,N] = A[M,K] · Bt[N,K]^T in bf16, LDS-tiled with
+// direct global->LDS DMA (the cdna_hip_programming.md §5 step-3 recipe:
+// 128x128 block tile, BK=64, double-buffered 16-byte global_load_lds).
+// Stresses MFMA + LDS + HBM together (the register stress above isolates
+// the matrix pipe). Asymmetric operands (A varies by row, B by column)
+// with exactly-representable products catch row/col index bugs
+// (guide §3: a symmetric B would hide a transposed C-write).
+// ---------------------------------------------------------------------------
+
+constexpr int BM = 128, BN = 128, BK = 64;
+
+__device__ __forceinline__ float a_val(int i) {
+  return 0.25f * ((i % 5) + 1);
+}
+__device__ __forceinline__ float b_val(int j) {
+  return 0.125f * ((j % 7) + 1);
+}
+
+__global__ __launch_bounds__(256) void gemm_fill_kernel(
+    __hip_bfloat16* __restrict__ A, __hip_bfloat16* __restrict__ Bt, int M,
+    int N, int K) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t na = (size_t)M * K, nb = (size_t)N * K;
+  for (size_t x = i; x < na; x += stride)
+    A[x] = __hip_bfloat16(a_val((int)(x / K)));
+  for (size_t x = i; x < nb; x += stride)
+    Bt[x] = __hip_bfloat16(b_val((int)(x / K)));
+}
+
+__global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  // ONE __shared__ object (guide §5 trap 4a): [2 buffers][A|B][128*64] bf16
+  __shared__ __hip_bfloat16 lds[2][2][BM * BK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;     // 4 waves
+  const int lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 64x64 tiles
+  const int tiles_n = (N + BN - 1) / BN;
+  const int brow = (blockIdx.x / tiles_n) * BM;
+  const int bcol = (blockIdx.x % tiles_n) * BN;
+  const int ntiles = K / BK;
+
+  // glds staging: each wave DMAs 8 KiB: rows [wave*32, wave*32+32) of the
+  // A tile and of the Bt tile; one 1 KiB instruction covers 8 rows
+  // (lane l -> row l/8, 16 B at col (l%8)*8), LDS image row-major [128][64]
+  auto stage = [&](int buf, int kt) {
+    const int k0 = kt * BK;
+    const __hip_bfloat16* gA = A + (size_t)(brow + wave * 32) * K + k0;
+    const __hip_bfloat16* gB = Bt + (size_t)(bcol + wave * 32) * K + k0;
+    __hip_bfloat16* lA = &lds[buf][0][wave * 32 * BK];
+    __hip_bfloat16* lB = &lds[buf][1][wave * 32 * BK];
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int row = g * 8 + (lane >> 3);
+      const int col = (lane & 7) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(gA +
+              (size_t)row * K + col),
+          (__attribute__((address_space(3))) unsigned int*)(lA + g * 8 * BK),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(gB +
+              (size_t)row * K + col),
+          (__attribute__((address_space(3))) unsigned int*)(lB + g * 8 * BK),
+          16, 0, 0);
+    }
+  };
+
+  f32x16 acc[2][2] = {};
+  stage(0, 0);
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < ntiles) stage((kt + 1) & 1, kt + 1);
+    // compute on the current buffer: 4 k-substeps of 16
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        // A fragment: row = wr*64 + m*32 + (lane&31),
+        //             k = ks*16 + (lane>>5)*8 + e  (8 contiguous bf16)
+        const int arow = wr * 64 + m * 32 + (lane & 31);
+        const bf16x8 af = *(const bf16x8*)&lds[cur][0][arow * BK + ks * 16 +
+                                                       (lane >> 5) * 8];
+#pragma unroll
+        for (int n = 0; n < 2; ++n) {
+          const int bcolr = wc * 64 + n * 32 + (lane & 31);
+          const bf16x8 bf = *(const bf16x8*)&lds[cur][1][bcolr * BK + ks * 16 +
+                                                         (lane >> 5) * 8];
+          acc[m][n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              (bf16x8)af, (bf16x8)bf, acc[m][n], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();  // hipcc emits vmcnt(0) here while glds is in flight
+  }
+
+  // epilogue: C/D layout for 32x32 MFMA (guide §3):
+  // col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+#pragma unroll
+  for (int m = 0; m < 2; ++m) {
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row =
+            brow + wr * 64 + m * 32 + (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        const int col = bcol + wc * 64 + n * 32 + (lane & 31);
+        C[(size_t)row * N + col] = acc[m][n][reg];
+      }
+    }
+  }
+}
+
+py::dict gemm_stress_bf16(int size, int iters) {
+  if (size % 128 != 0 || size < 256 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 128 in [256,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / BM) * (N / BN);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_bf16_kernel, dim3(blocks), dim3(256), 0, 0, d_a,
+                     d_bt, d_c, M, N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_bf16_kernel, dim3(blocks), dim3(256), 0, 0, d_a,
+                       d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  // verify a strided sample against the exact expected value
+  size_t bad = 0;
+  {
+    const int sample = 257;  // co-prime stride walk
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s) {
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    }
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
 }
 
 // ---------------------------------------------------------------------------
@@ -343,6 +611,12 @@ PYBIND11_MODULE(_diag, m) {
   m.def("mfma_stress_fp8", &mfma_stress_fp8, py::arg("iters") = 4096,
         py::arg("workgroups") = 1024,
         "Register-resident fp8(e4m3) MFMA stress (non-scaled, bf16 rate)");
+  m.def("mfma_stress_mxfp8", &mfma_stress_mxfp8, py::arg("iters") = 2048,
+        py::arg("workgroups") = 1024,
+        "MX-scaled fp8 MFMA stress (K=64 block-scaled, the ~5 PF dense path)");
+  m.def("gemm_stress_bf16", &gemm_stress_bf16, py::arg("size") = 8192,
+        py::arg("iters") = 5,
+        "LDS-tiled bf16 GEMM stress (128x128 tile, BK=64, global_load_lds)");
   m.def("hbm_bandwidth", &hbm_bandwidth, py::arg("buffer_gb") = 4.0,
         py::arg("iters") = 10, "float4 streaming triad + read over HBM3E");
   m.def("lds_bandwidth", &lds_bandwidth, py::arg("iters") = 100000,
