@@ -289,13 +289,24 @@ __global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
   const int lane = tid & 63;
   const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid of 64x64 tiles
   const int tiles_n = (N + BN - 1) / BN;
+  // blockIdx -> tile map: plain row-major. An XCD row-band remap was
+  // measured SLOWER here (931->842 TF @8192: a per-XCD row band's A slice
+  // is 16 MB, far over the 4 MiB XCD L2, while row-major striping already
+  // shares B tiles chip-wide through the 256 MiB L3), so the simple map
+  // stays.
   const int brow = (blockIdx.x / tiles_n) * BM;
   const int bcol = (blockIdx.x % tiles_n) * BN;
   const int ntiles = K / BK;
 
   // glds staging: each wave DMAs 8 KiB: rows [wave*32, wave*32+32) of the
   // A tile and of the Bt tile; one 1 KiB instruction covers 8 rows
-  // (lane l -> row l/8, 16 B at col (l%8)*8), LDS image row-major [128][64]
+  // (lane l -> row l/8, 16 B at col (l%8)*8), LDS image row-major [128][64].
+  // A 64-elem bf16 row is exactly 128 B, so a straight layout makes the
+  // column-wise ds_read_b128 fragment reads hit one bank per 16-lane group
+  // (the guide's "glds K-tile trap", up to 16-way conflict). Fix: XOR-swizzle
+  // the 16-B slot index with the row parity — applied to the per-lane GLOBAL
+  // source address (glds LDS writes stay lane-linear), and undone in the
+  // fragment-read addressing below.
   auto stage = [&](int buf, int kt) {
     const int k0 = kt * BK;
     const __hip_bfloat16* gA = A + (size_t)(brow + wave * 32) * K + k0;
@@ -305,7 +316,8 @@ __global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
 #pragma unroll
     for (int g = 0; g < 4; ++g) {
       const int row = g * 8 + (lane >> 3);
-      const int col = (lane & 7) * 8;
+      const int slot = (lane & 7) ^ (row & 7);  // pre-swizzled source slot
+      const int col = slot * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(gA +
               (size_t)row * K + col),
@@ -327,23 +339,29 @@ __global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
   for (int kt = 0; kt < ntiles; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < ntiles) stage((kt + 1) & 1, kt + 1);
-    // compute on the current buffer: 4 k-substeps of 16
+    // compute on the current buffer: 4 k-substeps of 16; fragment slot =
+    // (ks*2 + (lane>>5)) XOR (row&7), undoing the staged swizzle
 #pragma unroll
     for (int ks = 0; ks < 4; ++ks) {
+      const int kslot = ks * 2 + (lane >> 5);
+      const int arow0 = wr * 64 + (lane & 31);
+      const int bcol0 = wc * 64 + (lane & 31);
+      bf16x8 af[2], bf[2];
 #pragma unroll
       for (int m = 0; m < 2; ++m) {
-        // A fragment: row = wr*64 + m*32 + (lane&31),
-        //             k = ks*16 + (lane>>5)*8 + e  (8 contiguous bf16)
-        const int arow = wr * 64 + m * 32 + (lane & 31);
-        const bf16x8 af = *(const bf16x8*)&lds[cur][0][arow * BK + ks * 16 +
-                                                       (lane >> 5) * 8];
+        const int arow = arow0 + m * 32;
+        af[m] = *(const bf16x8*)&lds[cur][0][arow * BK +
+                                             (kslot ^ (arow & 7)) * 8];
+        const int bcolr = bcol0 + m * 32;
+        bf[m] = *(const bf16x8*)&lds[cur][1][bcolr * BK +
+                                             (kslot ^ (bcolr & 7)) * 8];
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
 #pragma unroll
         for (int n = 0; n < 2; ++n) {
-          const int bcolr = wc * 64 + n * 32 + (lane & 31);
-          const bf16x8 bf = *(const bf16x8*)&lds[cur][1][bcolr * BK + ks * 16 +
-                                                         (lane >> 5) * 8];
           acc[m][n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              (bf16x8)af, (bf16x8)bf, acc[m][n], 0, 0, 0);
+              af[m], bf[n], acc[m][n], 0, 0, 0);
         }
       }
     }

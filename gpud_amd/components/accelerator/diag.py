@@ -39,6 +39,8 @@ NAME_FABRIC = "accelerator-amd-diag-fabric"
 # genuinely sick board trips them.
 DEFAULT_BF16_TFLOPS_FLOOR = 1600.0
 DEFAULT_FP8_TFLOPS_FLOOR = 1600.0  # non-scaled fp8 runs at the bf16 rate
+DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4570 TF
+DEFAULT_GEMM_TFLOPS_FLOOR = 650.0  # LDS-tiled bf16 GEMM, measured ~930-1000
 DEFAULT_HBM_GBPS_FLOOR = 4500.0
 DEFAULT_LDS_TBPS_FLOOR = 60.0
 DEFAULT_XGMI_PAIR_GBPS_FLOOR = 30.0  # per direction, pairwise sendrecv
@@ -89,8 +91,12 @@ class MFMADiagComponent(_ManualDiagComponent):
         super().__init__(inst)
         self.bf16_floor = DEFAULT_BF16_TFLOPS_FLOOR
         self.fp8_floor = DEFAULT_FP8_TFLOPS_FLOOR
+        self.mxfp8_floor = DEFAULT_MXFP8_TFLOPS_FLOOR
+        self.gemm_floor = DEFAULT_GEMM_TFLOPS_FLOOR
         self.iters = 2048
         self.workgroups = 1024
+        self.gemm_size = 4096
+        self.gemm_iters = 3
 
     @property
     def name(self) -> str:
@@ -114,23 +120,38 @@ class MFMADiagComponent(_ManualDiagComponent):
             diag.set_device(dev)
             bf16 = diag.mfma_stress_bf16(iters=self.iters, workgroups=self.workgroups)
             fp8 = diag.mfma_stress_fp8(iters=self.iters, workgroups=self.workgroups)
+            mxfp8 = diag.mfma_stress_mxfp8(
+                iters=self.iters, workgroups=self.workgroups
+            )
+            gemm = diag.gemm_stress_bf16(
+                size=self.gemm_size, iters=self.gemm_iters
+            )
             extra[f"gpu{dev}.bf16_tflops"] = f"{bf16['tflops']:.0f}"
             extra[f"gpu{dev}.fp8_tflops"] = f"{fp8['tflops']:.0f}"
-            if not bf16["verified"] or not fp8["verified"]:
-                failures.append(
-                    f"gpu{dev}: MFMA numeric verification FAILED "
-                    f"(bf16 bad={bf16['verify_failures']}, fp8 bad={fp8['verify_failures']})"
-                )
-            if bf16["tflops"] < self.bf16_floor:
-                failures.append(
-                    f"gpu{dev}: bf16 MFMA {bf16['tflops']:.0f} TF below floor "
-                    f"{self.bf16_floor:.0f}"
-                )
-            if fp8["tflops"] < self.fp8_floor:
-                failures.append(
-                    f"gpu{dev}: fp8 MFMA {fp8['tflops']:.0f} TF below floor "
-                    f"{self.fp8_floor:.0f}"
-                )
+            extra[f"gpu{dev}.mxfp8_tflops"] = f"{mxfp8['tflops']:.0f}"
+            extra[f"gpu{dev}.gemm_bf16_tflops"] = f"{gemm['tflops']:.0f}"
+            for name, res in (
+                ("bf16", bf16),
+                ("fp8", fp8),
+                ("mxfp8", mxfp8),
+                ("gemm", gemm),
+            ):
+                if not res["verified"]:
+                    failures.append(
+                        f"gpu{dev}: {name} MFMA numeric verification FAILED "
+                        f"(bad={res['verify_failures']})"
+                    )
+            for name, res, floor in (
+                ("bf16", bf16, self.bf16_floor),
+                ("fp8", fp8, self.fp8_floor),
+                ("mxfp8", mxfp8, self.mxfp8_floor),
+                ("gemm bf16", gemm, self.gemm_floor),
+            ):
+                if res["tflops"] < floor:
+                    failures.append(
+                        f"gpu{dev}: {name} {res['tflops']:.0f} TF below floor "
+                        f"{floor:.0f}"
+                    )
         if failures:
             return CheckResult(
                 self.name,

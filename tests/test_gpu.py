@@ -76,6 +76,27 @@ def test_diag_mfma_fp8_verified():
     assert res["tflops"] > 1600, res
 
 
+def test_diag_mfma_mxfp8_verified():
+    from gpud_amd.diag import _diag
+
+    _diag.set_device(0)
+    res = _diag.mfma_stress_mxfp8(iters=1024, workgroups=1024)
+    assert res["verified"], res
+    # the MX-scaled path must clearly exceed the non-scaled fp8 rate
+    assert res["tflops"] > 3400, res
+
+
+def test_diag_gemm_bf16_verified():
+    from gpud_amd.diag import _diag
+
+    _diag.set_device(0)
+    res = _diag.gemm_stress_bf16(size=2048, iters=2)
+    assert res["verified"], res
+    res = _diag.gemm_stress_bf16(size=4096, iters=2)
+    assert res["verified"], res
+    assert res["tflops"] > 650, res
+
+
 def test_diag_hbm_bandwidth():
     from gpud_amd.diag import _diag
 
