@@ -219,6 +219,10 @@ def main() -> int:
                 "data_source": data_source,
                 "poll_cycle_p99_ms": round(p99, 4),
                 "daemon_cpu_percent": round(cpu_pct_max, 2),
+                # deployed overhead: cycles run once per 60 s poll interval
+                "projected_cpu_percent_at_60s_interval": round(
+                    100.0 * (mean_ms / 1000.0) / 60.0, 5
+                ),
                 "mfma_bf16_tflops_per_gpu": mfma_tflops,
             },
         }

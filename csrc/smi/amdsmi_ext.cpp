@@ -848,6 +848,16 @@ py::dict snapshot_to_dict(const Snapshot& s) {
   return d;
 }
 
+
+bool power_management_enabled(int index) {
+  auto h = handle_at(index);
+  bool enabled = false;
+  py::gil_scoped_release nogil;
+  check(amdsmi_is_gpu_power_management_enabled(h, &enabled),
+        "amdsmi_is_gpu_power_management_enabled");
+  return enabled;
+}
+
 py::dict metrics_snapshot(int index) {
   auto h = handle_at(index);
   Snapshot s;
@@ -968,6 +978,8 @@ PYBIND11_MODULE(_amdsmi, m) {
   m.def("bad_page_info", &bad_page_info, py::arg("index"));
   m.def("process_list", &process_list, py::arg("index"));
   m.def("violation_status", &violation_status, py::arg("index"));
+  m.def("power_management_enabled", &power_management_enabled,
+        py::arg("index"));
   m.def("xgmi_link_status", &xgmi_link_status, py::arg("index"));
   m.def("xgmi_error_status", &xgmi_error_status, py::arg("index"));
   m.def("xgmi_info", &xgmi_info, py::arg("index"));

@@ -77,6 +77,31 @@ def run(
             raise typer.Exit(code=1)
 
     core = build_core(cfg, in_memory_db=in_memory_db)
+
+    # token FIFO (reference: pkg/server/server.go:640-710 — a named pipe at
+    # <dataDir>/gpud.fifo other local processes write a fresh CP token into)
+    if not in_memory_db:
+        try:
+            if not os.path.exists(cfg.fifo_path):
+                os.mkfifo(cfg.fifo_path)
+
+            def _fifo_watch():
+                from ..pkg import metadata as _md
+
+                while True:
+                    try:
+                        with open(cfg.fifo_path) as f:  # blocks for a writer
+                            new_token = f.read().strip()
+                        if new_token:
+                            _md.set_value(core.db_rw, _md.KEY_TOKEN, new_token)
+                    except OSError:
+                        return
+
+            import threading as _th
+
+            _th.Thread(target=_fifo_watch, daemon=True).start()
+        except OSError:
+            pass
     # register plugin components
     for spec in specs:
         if spec.plugin_type == custom_plugins.PLUGIN_TYPE_INIT:
@@ -390,6 +415,78 @@ def login(
         typer.echo(f"login failed: {err}", err=True)
         raise typer.Exit(code=1)
     typer.echo("login ok")
+
+
+@app.command()
+def logout(data_dir: str = typer.Option(DEFAULT_DATA_DIR)):
+    """Clear control-plane credentials (reference: gpud logout)."""
+    from ..pkg import metadata as md
+    from ..pkg.sqlite_util import open_rw
+
+    cfg = Config(data_dir=data_dir)
+    conn = open_rw(cfg.state_path)
+    md.create_table(conn)
+    md.delete_value(conn, md.KEY_TOKEN)
+    md.delete_value(conn, md.KEY_MACHINE_PROOF)
+    conn.close()
+    typer.echo("logged out")
+
+
+release_app = typer.Typer(help="release signing (reference: pkg/release/distsign)")
+app.add_typer(release_app, name="release")
+
+
+@release_app.command("gen-key")
+def release_gen_key(out_prefix: str = typer.Argument(...)):
+    """Generate an ed25519 keypair: <prefix>.key (seed) + <prefix>.pub."""
+    from ..pkg import distsign
+
+    seed, pub = distsign.generate_keypair()
+    with open(out_prefix + ".key", "wb") as f:
+        f.write(seed)
+    os.chmod(out_prefix + ".key", 0o600)
+    with open(out_prefix + ".pub", "wb") as f:
+        f.write(pub)
+    typer.echo(f"wrote {out_prefix}.key and {out_prefix}.pub")
+
+
+@release_app.command("sign")
+def release_sign(
+    artifact: str = typer.Argument(...),
+    key: str = typer.Option(..., help="path to the .key seed file"),
+):
+    """Sign an artifact; writes <artifact>.sig."""
+    from ..pkg import distsign
+
+    with open(artifact, "rb") as f:
+        data = f.read()
+    with open(key, "rb") as f:
+        seed = f.read()
+    sig = distsign.sign(data, seed)
+    with open(artifact + ".sig", "wb") as f:
+        f.write(sig)
+    typer.echo(f"wrote {artifact}.sig")
+
+
+@release_app.command("verify")
+def release_verify(
+    artifact: str = typer.Argument(...),
+    pub: str = typer.Option(..., help="path to the .pub key file"),
+    sig: str = typer.Option("", help="signature file (default <artifact>.sig)"),
+):
+    from ..pkg import distsign
+
+    with open(artifact, "rb") as f:
+        data = f.read()
+    with open(sig or artifact + ".sig", "rb") as f:
+        signature = f.read()
+    with open(pub, "rb") as f:
+        public = f.read()
+    if distsign.verify(data, signature, public):
+        typer.echo("signature OK")
+    else:
+        typer.echo("signature INVALID", err=True)
+        raise typer.Exit(code=1)
 
 
 @app.command()

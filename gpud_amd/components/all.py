@@ -20,6 +20,7 @@ from .accelerator import (
     gpu_counts,
     gpu_memory,
     power,
+    power_management,
     processes,
     rccl,
     temperature,
@@ -58,6 +59,7 @@ def all_init_funcs() -> List[InitFunc]:
         rccl.new,
         xgmi.new,
         power.new,
+        power_management.new,
         processes.new,
         bad_pages.new,
         temperature.new,

@@ -268,6 +268,44 @@ def create_app(core: DaemonCore, plugin_specs: Optional[list] = None) -> FastAPI
     def admin_config(request: Request):
         return _negotiate(request, core.config.to_dict())
 
+    @app.get("/admin/packages")
+    def admin_packages(request: Request):
+        from ..pkg.gpud_manager import package_statuses
+
+        return _negotiate(
+            request, [p.to_dict() for p in package_statuses(core.config)]
+        )
+
+    # -- profiling (reference: /admin/pprof/* behind --pprof;
+    #    Python analog: live thread stacks + tracemalloc heap) --------------
+
+    @app.get("/admin/pprof/threads")
+    def pprof_threads(request: Request):
+        import sys
+        import traceback
+
+        frames = sys._current_frames()
+        out = {}
+        import threading as _t
+
+        names = {t.ident: t.name for t in _t.enumerate()}
+        for tid, frame in frames.items():
+            out[names.get(tid, str(tid))] = traceback.format_stack(frame)
+        return _negotiate(request, out)
+
+    @app.get("/admin/pprof/heap")
+    def pprof_heap(request: Request):
+        import tracemalloc
+
+        if not tracemalloc.is_tracing():
+            tracemalloc.start()
+            return _negotiate(
+                request, {"status": "tracing started; query again for stats"}
+            )
+        snap = tracemalloc.take_snapshot()
+        top = snap.statistics("lineno")[:50]
+        return _negotiate(request, [str(s) for s in top])
+
     # -- prometheus -----------------------------------------------------------
 
     @app.get("/metrics")

@@ -103,6 +103,21 @@ class Session:
             target=self._serve_loop, daemon=True, name="gpud-session"
         )
         self._thread.start()
+        # keepalive ticker (reference: session_keepalive.go:11) — a periodic
+        # ping frame on the writer stream so the control plane sees liveness
+        self._ka_thread = threading.Thread(
+            target=self._keepalive_loop, daemon=True, name="gpud-session-ka"
+        )
+        self._ka_thread.start()
+
+    def _keepalive_loop(self) -> None:
+        while not self._stop.wait(KEEPALIVE_INTERVAL):
+            try:
+                self._send_response(
+                    {"req_id": "", "method": "ping", "data": {"pong": True}}
+                )
+            except Exception:
+                pass  # reconnect machinery handles the stream
 
     def stop(self) -> None:
         self._stop.set()

@@ -89,6 +89,9 @@ class Device:
     def process_list(self) -> List[Dict[str, Any]]:
         return list(self._b.process_list(self.index))
 
+    def power_management_enabled(self) -> bool:
+        return bool(self._b.power_management_enabled(self.index))
+
     def violation_status(self) -> Dict[str, Any]:
         return self._b.violation_status(self.index)
 

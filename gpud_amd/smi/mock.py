@@ -234,6 +234,10 @@ class MockBackend:
     def process_list(self, i: int) -> List[Dict[str, Any]]:
         return list(self._check(i)["processes"])
 
+    def power_management_enabled(self, i: int) -> bool:
+        self._check(i)
+        return True
+
     def violation_status(self, i: int) -> Dict[str, Any]:
         s = self._check(i)
         base = {

@@ -80,3 +80,36 @@ def test_list_plugins_and_run_group(tmp_path):
     res = runner.invoke(app, ["run-plugin-group", str(specs), "--tag", "grp"])
     assert res.exit_code == 0
     assert "custom-plugin-quick: Healthy" in res.output
+
+
+def test_release_sign_verify(tmp_path):
+    art = tmp_path / "artifact.bin"
+    art.write_bytes(b"release payload")
+    prefix = str(tmp_path / "rootkey")
+    res = runner.invoke(app, ["release", "gen-key", prefix])
+    assert res.exit_code == 0, res.output
+    res = runner.invoke(app, ["release", "sign", str(art), "--key", prefix + ".key"])
+    assert res.exit_code == 0, res.output
+    res = runner.invoke(app, ["release", "verify", str(art), "--pub", prefix + ".pub"])
+    assert res.exit_code == 0 and "OK" in res.output
+    # tamper -> invalid
+    art.write_bytes(b"tampered payload!")
+    res = runner.invoke(app, ["release", "verify", str(art), "--pub", prefix + ".pub"])
+    assert res.exit_code == 1
+
+
+def test_logout_cli(tmp_path):
+    from gpud_amd.pkg import metadata
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.sqlite_util import open_ro, open_rw
+
+    cfg = Config(data_dir=str(tmp_path))
+    conn = open_rw(cfg.state_path)
+    metadata.create_table(conn)
+    metadata.set_value(conn, metadata.KEY_TOKEN, "tok")
+    conn.close()
+    res = runner.invoke(app, ["logout", "--data-dir", str(tmp_path)])
+    assert res.exit_code == 0
+    conn = open_ro(cfg.state_path)
+    assert metadata.get_value(conn, metadata.KEY_TOKEN) == ""
+    conn.close()
