@@ -23,6 +23,7 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <thread>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -937,8 +938,21 @@ py::list metrics_snapshot_all() {
   std::vector<Snapshot> snaps(handles.size());
   {
     py::gil_scoped_release nogil;
-    for (size_t i = 0; i < handles.size(); ++i)
-      take_snapshot(handles[i], snaps[i]);
+    if (handles.size() <= 1) {
+      for (size_t i = 0; i < handles.size(); ++i)
+        take_snapshot(handles[i], snaps[i]);
+    } else {
+      // one thread per GPU: amdsmi getters are thread-safe reads, and the
+      // per-device ioctls dominate, so an 8-GPU sweep costs ~one GPU's
+      // latency instead of 8x (keeps the single-daemon poll cycle flat
+      // to 8 GPUs — SURVEY.md §7 hard parts)
+      std::vector<std::thread> ts;
+      ts.reserve(handles.size());
+      for (size_t i = 0; i < handles.size(); ++i) {
+        ts.emplace_back([&, i] { take_snapshot(handles[i], snaps[i]); });
+      }
+      for (auto& t : ts) t.join();
+    }
   }
   py::list out;
   for (auto& s : snaps) out.append(snapshot_to_dict(s));
