@@ -612,8 +612,15 @@ const StaticInfo& static_info_for(amdsmi_processor_handle h) {
     si.power_cap_uw = cap.power_cap;
   amdsmi_power_info_t pw;
   std::memset(&pw, 0, sizeof(pw));
-  if (amdsmi_get_power_info(h, &pw) == AMDSMI_STATUS_SUCCESS)
-    si.power_limit_w = pw.power_limit;
+  if (amdsmi_get_power_info(h, &pw) == AMDSMI_STATUS_SUCCESS) {
+    // observed on ROCm 7.2/MI355X: power_limit reported in MICROwatts
+    // (1400000000 for a 1400 W board) — normalize to watts
+    si.power_limit_w = pw.power_limit > 100000
+                           ? (uint32_t)(pw.power_limit / 1000000)
+                           : pw.power_limit;
+  }
+  if (si.power_limit_w == 0 && si.power_cap_uw > 0)
+    si.power_limit_w = (uint32_t)(si.power_cap_uw / 1000000);
   amdsmi_clk_info_t ci;
   std::memset(&ci, 0, sizeof(ci));
   if (amdsmi_get_clock_info(h, AMDSMI_CLK_TYPE_GFX, &ci) ==
@@ -669,7 +676,10 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
     s.power_w = gm_valid16(gm.current_socket_power)
                     ? gm.current_socket_power
                     : gm.average_socket_power;
-    s.avg_power_w = gm.average_socket_power;
+    // 0xFFFF = not-supported sentinel in the metrics table
+    s.avg_power_w = gm_valid16(gm.average_socket_power)
+                        ? gm.average_socket_power
+                        : s.power_w;
     s.ok_clock = true;
     s.gfx_mhz = gm_valid16(gm.current_gfxclk) ? gm.current_gfxclk
                                               : gm.average_gfxclk_frequency;
