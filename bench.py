@@ -62,10 +62,16 @@ class _RankFilteredSMI:
         return self._inst.product_name
 
     def snapshot_all(self):
-        devs = self._inst.devices()
+        from gpud_amd.smi import Instance
+
+        devs = self._inst.devices()  # applies gpu-lost filtering
         if self.uuid not in devs:
             return {}
-        return {self.uuid: devs[self.uuid].snapshot()}
+        snap = devs[self.uuid].snapshot()
+        fi = self._inst.failure_injector
+        if fi is not None:
+            snap = Instance._apply_injection(self.uuid, snap, fi)
+        return {self.uuid: snap}
 
     def shutdown(self):
         pass
@@ -77,6 +83,12 @@ def main() -> int:
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--mock", action="store_true", help="CPU-only: mock SMI backend")
+    p.add_argument(
+        "--fault-replay",
+        action="store_true",
+        help="inject SMI-level faults on a schedule during the timed region "
+        "(BASELINE.json config 5) and report detection counts",
+    )
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -139,10 +151,50 @@ def main() -> int:
         and "diag" not in c.name
     ]
 
+    # fault-replay: flip a rotating SMI-level failure on/off every 10 cycles
+    # and count how many cycles observe a non-Healthy accelerator state —
+    # the injected-fault load the north star's baseline asks for.
+    fi = core.smi_failure_injector
+    my_uuids = core.gpud_instance.smi.device_uuids()
+    fault_kinds = ("ecc", "throttle", "xgmi", "bad_pages")
+    replay = {"cycle": 0, "active": "", "detected": 0, "injected_cycles": 0}
+
+    def _apply_fault(kind: str, on: bool) -> None:
+        if not my_uuids:
+            return
+        u = my_uuids[0]
+        target = {
+            "ecc": fi.ecc_uncorrectable_uuids,
+            "throttle": fi.throttle_uuids,
+            "xgmi": fi.xgmi_unhealthy_uuids,
+            "bad_pages": fi.bad_page_pending_uuids,
+        }[kind]
+        (target.add if on else target.discard)(u)
+
     def one_cycle() -> None:
+        if args.fault_replay:
+            c_i = replay["cycle"]
+            if c_i % 10 == 0:
+                if replay["active"]:
+                    _apply_fault(replay["active"], False)
+                replay["active"] = (
+                    fault_kinds[(c_i // 10) % len(fault_kinds)]
+                    if (c_i // 10) % 2 == 0
+                    else ""
+                )
+                if replay["active"]:
+                    _apply_fault(replay["active"], True)
+            replay["cycle"] = c_i + 1
         core.shared_snapshots.refresh()
+        unhealthy_seen = False
         for c in accel_components:
-            c.trigger_check()
+            cr = c.trigger_check()
+            if cr.health != "Healthy":
+                unhealthy_seen = True
+        if args.fault_replay and replay["active"]:
+            replay["injected_cycles"] += 1
+            if unhealthy_seen:
+                replay["detected"] += 1
 
     def barrier_sync() -> None:
         if distributed:
@@ -226,6 +278,11 @@ def main() -> int:
                 "mfma_bf16_tflops_per_gpu": mfma_tflops,
             },
         }
+        if args.fault_replay:
+            out["config"]["fault_replay"] = {
+                "injected_cycles": replay["injected_cycles"],
+                "detected_cycles": replay["detected"],
+            }
         print(json.dumps(out))
 
     core.close()

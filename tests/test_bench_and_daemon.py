@@ -133,3 +133,19 @@ def test_daemon_boot_via_cli():
         except subprocess.TimeoutExpired:
             os.killpg(proc.pid, signal.SIGKILL)
             proc.wait(timeout=5)
+
+
+def test_bench_fault_replay_mock():
+    out = subprocess.run(
+        [
+            sys.executable, "bench.py", "--mock", "--steps", "40",
+            "--warmup", "2", "--fault-replay",
+        ],
+        capture_output=True, text=True, timeout=180, cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-800:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    fr = d["config"]["fault_replay"]
+    assert fr["injected_cycles"] > 0
+    # every injected-fault cycle must be detected by some component
+    assert fr["detected_cycles"] == fr["injected_cycles"]
