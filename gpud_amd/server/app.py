@@ -66,6 +66,10 @@ def _parse_since(request: Request) -> datetime.datetime:
 
 def create_app(core: DaemonCore, plugin_specs: Optional[list] = None) -> FastAPI:
     app = FastAPI(title="gpud", version=__version__, docs_url="/swagger")
+    # gzip via Accept-Encoding (reference: pkg/server/server.go:412)
+    from starlette.middleware.gzip import GZipMiddleware
+
+    app.add_middleware(GZipMiddleware, minimum_size=1024)
     registry = core.registry
     app.state.core = core
     app.state.plugin_specs = plugin_specs or []

@@ -171,3 +171,44 @@ def test_deregister_rules(daemon):
         verify=False,
     )
     assert r.status_code == 404
+
+
+def test_trigger_tag_endpoint(daemon):
+    _core, server, client = daemon
+    import httpx
+
+    r = httpx.get(
+        server.base_url + "/v1/components/trigger-tag",
+        params={"tagName": "network"},
+        verify=False,
+    )
+    assert r.status_code == 200
+    body = r.json()
+    assert "components" in body and "success" in body
+
+
+def test_gzip_negotiation(daemon):
+    _core, server, _client = daemon
+    import httpx
+
+    r = httpx.get(
+        server.base_url + "/v1/states",
+        headers={"accept-encoding": "gzip"},
+        verify=False,
+    )
+    assert r.status_code == 200
+    # httpx transparently decompresses; verify the server really gzipped
+    assert r.headers.get("content-encoding") == "gzip"
+
+
+def test_admin_endpoints(daemon):
+    _core, server, _client = daemon
+    import httpx
+
+    r = httpx.get(server.base_url + "/admin/config", verify=False)
+    assert r.status_code == 200 and "data_dir" in r.json()
+    r = httpx.get(server.base_url + "/admin/pprof/threads", verify=False)
+    assert r.status_code == 200
+    assert any("gpud" in k or "MainThread" in k for k in r.json())
+    r = httpx.get(server.base_url + "/admin/packages", verify=False)
+    assert r.status_code == 200
