@@ -54,6 +54,33 @@ class ECCComponent(TickerComponent, SmiComponentMixin):
     def events(self, since: datetime.datetime):
         return self._bucket.get(since) if self._bucket is not None else []
 
+    # RAS block bits (amdsmi_gpu_block_t): queried only when UEs appear
+    _BLOCKS = (
+        ("UMC", 1 << 0),
+        ("SDMA", 1 << 1),
+        ("GFX", 1 << 2),
+        ("MMHUB", 1 << 3),
+        ("XGMI_WAFL", 1 << 7),
+    )
+
+    def _blame_blocks(self, uuid: str) -> str:
+        try:
+            dev = self._smi.devices().get(uuid)
+            if dev is None:
+                return ""
+            hits = []
+            for name, bit in self._BLOCKS:
+                try:
+                    ec = dev.ecc_count_block(bit)
+                except Exception:
+                    continue
+                ue = int(ec.get("uncorrectable", 0))
+                if ue > 0:
+                    hits.append(f"{name}:{ue}")
+            return ",".join(hits)
+        except Exception:
+            return ""
+
     def check(self) -> CheckResult:
         guard = self.smi_guard()
         if guard is not None:
@@ -89,6 +116,11 @@ class ECCComponent(TickerComponent, SmiComponentMixin):
             extra[f"{uuid}.ecc"] = f"ce={corr},ue={uncorr},de={deferred}"
             if uncorr > 0:
                 unhealthy.append((uuid, uncorr))
+                # attribute the UE to RAS blocks (per-block queries are
+                # extra ioctls, so they run only on the failure path)
+                blocks = self._blame_blocks(uuid)
+                if blocks:
+                    extra[f"{uuid}.ue_blocks"] = blocks
             prev = self._last_correctable.get(uuid)
             if prev is not None and corr > prev and self._bucket is not None:
                 self._bucket.insert(

@@ -169,3 +169,56 @@ def test_diag_components_on_gpu():
         assert cr.health == HealthStateType.HEALTHY, f"{cr.reason} / {cr.error}"
     finally:
         core.close()
+
+
+def test_daemon_boot_live_gpu():
+    """Boot the full daemon against the REAL amdsmi + kmsg stack and poke
+    the HTTP surface — the end-to-end native path on hardware."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+
+    def _free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    port = _free_port()
+    env = {**os.environ, "PYTHONPATH": REPO}
+    env.pop("GPUD_AMDSMI_MOCK", None)
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run",
+            "--in-memory-db", "--address", f"127.0.0.1:{port}",
+            "--log-level", "warning",
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(60), "daemon did not become healthy"
+        states = client.get_health_states()
+        temp = states["accelerator-amd-temperature"][0]
+        assert temp.health in ("Healthy", "Degraded", "Initializing"), temp.reason
+        mi = client.get_machine_info()
+        assert mi.get("gpuInfo", {}).get("gpus"), "live GPU not in machine-info"
+        out = client.trigger_check(component="accelerator-amd-xgmi")
+        assert out["states"], "xgmi trigger-check returned nothing"
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            os.killpg(proc.pid, signal.SIGKILL)
+            proc.wait(timeout=5)
