@@ -385,6 +385,231 @@ __global__ __launch_bounds__(256, 2) void gemm_bf16_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// GEMM v2: the 256² 8-phase structure (cdna_hip_programming.md §5 "The 256²
+// 8-phase template"): 256x256 tile, BK=64, 8 waves (2M×4N, 512 threads),
+// mfma_f32_16x16x32_bf16, 128 KiB double-buffered LDS, st_16x32 swizzle
+// (byte ^= ((byte>>9)&1)<<5 — kills the 8-way bank conflict of 128-B rows),
+// raw s_barrier + s_setprio(1) around the MFMA burst, and K-tile-boundary
+// vmcnt(0) placed where only the CURRENT tile's global_load_lds are
+// outstanding (prefetches for the next tile are issued in phases 0-1, so
+// the drain at the next boundary has had ≥2 phases of cover).
+// ---------------------------------------------------------------------------
+
+typedef f32x4 accfrag_t;
+
+__device__ __forceinline__ int swz(int byte_off) {
+  // st_16x32-style XOR swizzle extended for this kernel's read pattern:
+  // bank-slot bits of a 16-B ds_read_b128 granule are byte bits 4..7; the
+  // guide's bit5^=bit9 alone leaves row bit1 unused (measured exactly
+  // 2-way, SQ_LDS_BANK_CONFLICT/IDX_ACTIVE = 50%), so bit6^=bit8 folds it
+  // in too -> all 16 slots distinct within each 16-lane group.
+  return byte_off ^ (((byte_off >> 9) & 1) << 5) ^
+         (((byte_off >> 8) & 1) << 6);
+}
+
+__global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  // [2 buffers][A image 32 KiB | B image 32 KiB]
+  __shared__ __hip_bfloat16 lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;          // 8 waves
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;       // 0..1 -> output rows [wave_m*128, +128)
+  const int wave_n = wave & 3;        // 0..3 -> output cols [wave_n*64, +64)
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  // staging: 64 slots of 1 KiB cover [A image | B image]; wave w owns
+  // slots [w*8, w*8+8) (waves 0-3 stage A rows, 4-7 stage B cols).
+  // Lane l of slot s writes LDS byte D = s*1024 + l*16 (lane-linear glds);
+  // the content belongs at logical offset L = swz(D) (XOR is an involution),
+  // so the GLOBAL source address is pre-swizzled per lane.
+  auto stage_slots = [&](int buf, int kt, int s0, int nslots) {
+    const int k0 = kt * TK;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (j >= nslots) break;
+      const int s = wave * 8 + s0 + j;
+      const int img = s >> 5;           // 0 = A, 1 = B
+      const int s_img = s & 31;
+      const int D = s_img * 1024 + lane * 16;
+      const int L = swz(D);
+      const int row = L >> 7;           // row of the [256][64] bf16 image
+      const int k = (L & 127) >> 1;
+      const __hip_bfloat16* g =
+          img == 0 ? A + (size_t)(brow + row) * K + k0 + k
+                   : Bt + (size_t)(bcol + row) * K + k0 + k;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)(
+              &lds[buf][img * TM * TK] + s_img * 512),
+          16, 0, 0);
+    }
+  };
+
+  // fragment read for mfma_f32_16x16x32_bf16: lane l holds row
+  // (base + (l&15)), k = (l>>4)*8 + e — 16 B at the swizzled offset of
+  // byte row*128 + kh*64 + (l>>4)*16
+  auto read_a = [&](int buf, int rbase, int kh) -> bf16x8 {
+    const int r = rbase + (lane & 15);
+    const int off = swz(r * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][0] + off);
+  };
+  auto read_b = [&](int buf, int cbase, int kh) -> bf16x8 {
+    const int c = cbase + (lane & 15);
+    const int off = swz(c * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][TM * TK] + off);
+  };
+
+  accfrag_t acc[8][4] = {};  // 8 row-frags x 4 col-frags of 16x16
+  bf16x8 bfrag[4][2];        // per-K-tile B fragments (reused by all phases)
+
+  // prologue: stage tile 0 (all 8 slots per wave)
+  stage_slots(0, 0, 0, 4);
+  stage_slots(0, 0, 4, 4);
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    const bool has_next = kt + 1 < ntiles;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      // loads for this phase
+      if (p == 0) {
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+          for (int kh = 0; kh < 2; ++kh)
+            bfrag[c][kh] = read_b(cur, wave_n * 64 + c * 16, kh);
+        }
+      }
+      bf16x8 afrag[2][2];
+#pragma unroll
+      for (int rr = 0; rr < 2; ++rr) {
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh)
+          afrag[rr][kh] =
+              read_a(cur, wave_m * 128 + p * 32 + rr * 16, kh);
+      }
+      // prefetch next tile early: 4 slots in phase 0, 4 in phase 1
+      // (all 8 in phase 0 congests it: measured 1154 -> 1001 TF @8192;
+      // 2/phase leaves the last pair under-covered at the boundary drain)
+      if (has_next && p < 2) stage_slots(nxt, kt + 1, p * 4, 4);
+      // rendezvous + MFMA burst (template: barrier, lgkmcnt(0), prio 1)
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+      // kh outer: 8 independent MFMAs between accumulator reuses (the
+      // dependent-accumulator latency exceeds the issue interval)
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+#pragma unroll
+          for (int c = 0; c < 4; ++c) {
+            acc[p * 2 + rr][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[rr][kh], bfrag[c][kh], acc[p * 2 + rr][c], 0, 0, 0);
+          }
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      if (p == 3) {
+        // K-tile boundary: only tile kt+1's 8 glds are outstanding, and
+        // they were issued >=2 phases ago — this drain is cheap and no
+        // *later* prefetch gets caught by it (the step-3 ceiling trap)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row =
+            brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
+py::dict gemm_stress_bf16_v2(int size, int iters) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_bf16_8phase_kernel, dim3(blocks), dim3(512), 0, 0,
+                     d_a, d_bt, d_c, M, N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_bf16_8phase_kernel, dim3(blocks), dim3(512), 0, 0,
+                       d_a, d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-8phase";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_bf16(int size, int iters) {
   if (size % 128 != 0 || size < 256 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 128 in [256,16384]");
@@ -635,6 +860,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16", &gemm_stress_bf16, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "LDS-tiled bf16 GEMM stress (128x128 tile, BK=64, global_load_lds)");
+  m.def("gemm_stress_bf16_v2", &gemm_stress_bf16_v2, py::arg("size") = 8192,
+        py::arg("iters") = 5,
+        "bf16 GEMM stress, 256^2 8-phase structure (swizzled LDS, setprio)");
   m.def("hbm_bandwidth", &hbm_bandwidth, py::arg("buffer_gb") = 4.0,
         py::arg("iters") = 10, "float4 streaming triad + read over HBM3E");
   m.def("lds_bandwidth", &lds_bandwidth, py::arg("iters") = 100000,
