@@ -40,7 +40,7 @@ NAME_FABRIC = "accelerator-amd-diag-fabric"
 DEFAULT_BF16_TFLOPS_FLOOR = 1600.0
 DEFAULT_FP8_TFLOPS_FLOOR = 1600.0  # non-scaled fp8 runs at the bf16 rate
 DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4570 TF
-DEFAULT_GEMM_TFLOPS_FLOOR = 650.0  # LDS-tiled bf16 GEMM, measured ~930-1000
+DEFAULT_GEMM_TFLOPS_FLOOR = 800.0  # 8-phase bf16 GEMM, measured ~1100-1150
 DEFAULT_HBM_GBPS_FLOOR = 4500.0
 DEFAULT_LDS_TBPS_FLOOR = 60.0
 DEFAULT_XGMI_PAIR_GBPS_FLOOR = 30.0  # per direction, pairwise sendrecv
@@ -123,7 +123,7 @@ class MFMADiagComponent(_ManualDiagComponent):
             mxfp8 = diag.mfma_stress_mxfp8(
                 iters=self.iters, workgroups=self.workgroups
             )
-            gemm = diag.gemm_stress_bf16(
+            gemm = diag.gemm_stress_bf16_v2(
                 size=self.gemm_size, iters=self.gemm_iters
             )
             extra[f"gpu{dev}.bf16_tflops"] = f"{bf16['tflops']:.0f}"

@@ -95,6 +95,10 @@ def test_diag_gemm_bf16_verified():
     res = _diag.gemm_stress_bf16(size=4096, iters=2)
     assert res["verified"], res
     assert res["tflops"] > 650, res
+    # the 8-phase structure must verify and beat the 2-buffer one
+    res2 = _diag.gemm_stress_bf16_v2(size=4096, iters=2)
+    assert res2["verified"], res2
+    assert res2["tflops"] > 800, res2
 
 
 def test_diag_hbm_bandwidth():
