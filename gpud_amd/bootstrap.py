@@ -10,13 +10,13 @@ server, `scan`, tests and bench.py can all assemble the same daemon core.
 from __future__ import annotations
 
 import datetime
-from dataclasses import dataclass, field
-from typing import Any, List, Optional
+from dataclasses import dataclass
+from typing import Any, Optional
 
 from . import smi as smi_pkg
 from .components.accelerator.shared import SharedSnapshots
 from .components.all import all_init_funcs
-from .components.base import Component, GPUdInstance, Registry
+from .components.base import GPUdInstance, Registry
 from .pkg import metadata
 from .pkg.config import Config
 from .pkg.eventstore import Store as EventStore

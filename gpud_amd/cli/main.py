@@ -4,14 +4,11 @@ metadata/list-plugins/custom-plugins/run-plugin-group/login/logout)."""
 
 from __future__ import annotations
 
-import datetime
 import json
 import os
 import signal
 import sys
 import time
-from typing import List, Optional
-
 import typer
 
 from .. import __version__

@@ -24,10 +24,10 @@ from __future__ import annotations
 import json
 import os
 import subprocess
-from typing import Any, Callable, Dict, List, Optional
+from typing import Any, Callable, Dict, List
 
 from ...apiv1.types import HealthStateType, RepairActionType, RunModeType, SuggestedActions
-from ..base import CheckResult, Component, GPUdInstance, TickerComponent
+from ..base import CheckResult, GPUdInstance, TickerComponent
 
 NAME_MFMA = "accelerator-amd-diag-mfma"
 NAME_BANDWIDTH = "accelerator-amd-diag-bandwidth"

@@ -10,7 +10,7 @@ inspection; a rising correctable rate is surfaced as an event.
 from __future__ import annotations
 
 import datetime
-from typing import Callable, Dict, Optional
+from typing import Callable, Dict
 
 from ...apiv1.types import (
     Event,

@@ -9,7 +9,6 @@ from __future__ import annotations
 
 from typing import Callable
 
-from ...apiv1.types import HealthStateType
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 from ..metrics_util import ComponentGauges
 from .shared import SmiComponentMixin

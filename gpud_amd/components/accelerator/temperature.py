@@ -9,7 +9,7 @@ configured threshold, Unhealthy at/over the limit.
 
 from __future__ import annotations
 
-from typing import Callable, Optional
+from typing import Callable
 
 from ...apiv1.types import HealthStateType
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent

@@ -9,7 +9,6 @@ import os
 
 import psutil
 
-from ...apiv1.types import HealthStateType
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 from ..metrics_util import ComponentGauges
 

@@ -16,7 +16,6 @@ examples) with ``expect.regex`` rules and per-rule suggested actions.
 from __future__ import annotations
 
 import base64
-import datetime
 import json
 import re
 from dataclasses import dataclass
@@ -31,7 +30,7 @@ from ..apiv1.types import (
     RunModeType,
     SuggestedActions,
 )
-from ..components.base import CheckResult, Component, GPUdInstance, TickerComponent
+from ..components.base import CheckResult, Component, TickerComponent
 from .process_runner import run_bash
 
 PLUGIN_TYPE_INIT = "init"

@@ -14,7 +14,7 @@ Two injection surfaces, same as the reference:
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Set
+from typing import Dict, Optional, Set
 
 from . import ras_catalog
 from .kmsg.writer import Writer

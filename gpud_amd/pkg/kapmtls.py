@@ -10,7 +10,6 @@ previous credentials.
 from __future__ import annotations
 
 import os
-import shutil
 import time
 from typing import Dict, Optional
 

@@ -9,8 +9,6 @@ control-plane wire compatibility (apiv1.types.MachineInfo).
 
 from __future__ import annotations
 
-import datetime
-import os
 import platform
 import socket
 from typing import Any, Optional

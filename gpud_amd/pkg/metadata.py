@@ -6,7 +6,7 @@ endpoint, public/private IP, last login success timestamp.
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 from .sqlite_util import Conn
 

@@ -19,7 +19,7 @@ import json
 import random
 import threading
 import time
-from typing import Any, Callable, Dict, Iterator, List, Optional
+from typing import Callable, Dict, Iterator, List, Optional
 
 import httpx
 
