@@ -34,6 +34,12 @@ NAME = "accelerator-amd-xgmi"
 LINK_UP = 1
 LINK_DOWN = 0
 
+# flap auto-clear window (reference: infiniband store
+# flap_auto_clear_window.go): a link that went down recently keeps the
+# component Degraded for this long after recovery, so a flapping link
+# cannot blink the node healthy between polls
+FLAP_AUTO_CLEAR = datetime.timedelta(minutes=10)
+
 
 class XGMIComponent(TickerComponent, SmiComponentMixin):
     def __init__(self, inst: GPUdInstance):
@@ -180,6 +186,23 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
                     ],
                 ),
             )
+        # flap auto-clear: recent down events keep the state Degraded even
+        # after the link recovered (reference infiniband flap store)
+        if self._bucket is not None:
+            recent_flaps = self._bucket.find_by_name_since(
+                "amd_xgmi_link_down", utcnow() - FLAP_AUTO_CLEAR
+            )
+            if recent_flaps:
+                return CheckResult(
+                    NAME,
+                    health=HealthStateType.DEGRADED,
+                    reason=(
+                        f"xGMI link(s) flapped within the last "
+                        f"{int(FLAP_AUTO_CLEAR.total_seconds() // 60)} min "
+                        f"({len(recent_flaps)} down event(s)); links currently up"
+                    ),
+                    extra_info=extra,
+                )
         reason = (
             f"all xGMI links healthy on {len(snaps)} GPU(s)"
             if any_links

@@ -155,3 +155,20 @@ def test_tar_path_traversal_refused(tmp_path):
         cfg, "6.6.6", base_url=f"file://{tmp_path}", install_dir=str(tmp_path / "out")
     )
     assert err is not None and "unsafe path" in err
+
+
+def test_xgmi_flap_auto_clear(mock_core):
+    """A recovered link keeps the component Degraded inside the auto-clear
+    window (reference: infiniband flap store semantics)."""
+    backend = mock_core.smi_instance._b
+    comp = mock_core.registry.get("accelerator-amd-xgmi")
+    comp.trigger_check()  # prime last_states
+    backend.state[0]["xgmi_states"][2] = 0  # link drops
+    mock_core.shared_snapshots.refresh()
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    backend.state[0]["xgmi_states"][2] = 1  # link recovers
+    mock_core.shared_snapshots.refresh()
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED  # still in the window
+    assert "flapped" in cr.reason
