@@ -728,18 +728,14 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
         s.xgmi_states[i] = (int)xs.status[i];
     }
   }
-  // the ECC RAS query is the slowest remaining per-cycle ioctl (~0.7 ms);
-  // run it concurrently with the short tail calls below
-  std::memset(&s.ecc, 0, sizeof(s.ecc));
-  std::thread ecc_thread([h, &s] {
-    s.ok_ecc =
-        amdsmi_get_gpu_total_ecc_count(h, &s.ecc) == AMDSMI_STATUS_SUCCESS;
-  });
   amdsmi_xgmi_status_t xe;
   if (amdsmi_gpu_xgmi_error_status(h, &xe) == AMDSMI_STATUS_SUCCESS)
     s.xgmi_err = static_cast<int>(xe);
   std::memset(&s.vram, 0, sizeof(s.vram));
   s.ok_vram = amdsmi_get_gpu_vram_usage(h, &s.vram) == AMDSMI_STATUS_SUCCESS;
+  std::memset(&s.ecc, 0, sizeof(s.ecc));
+  s.ok_ecc =
+      amdsmi_get_gpu_total_ecc_count(h, &s.ecc) == AMDSMI_STATUS_SUCCESS;
   // NOTE: amdsmi_get_violation_status is NOT called here — it blocks ~100 ms
   // per GPU (double-samples internally to compute per_* rates). The same
   // throttle residency accumulators come from gpu_metrics above; the
@@ -768,7 +764,6 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
     s.ok_bp_threshold = si.has_bp_threshold;
     s.bp_threshold = si.bp_threshold;
   }
-  ecc_thread.join();
 }
 
 py::dict snapshot_to_dict(const Snapshot& s) {
