@@ -1,22 +1,48 @@
-"""memory — host RAM usage.
+"""memory — host RAM usage + OOM/EDAC kernel events.
 
-Reference: components/memory (gopsutil virtual memory — memory/component.go:28).
+Reference: components/memory (gopsutil virtual memory — memory/component.go:28;
+kmsg matcher for oom / oom_cgroup / edac events — memory testdata fixtures).
 """
 
 from __future__ import annotations
 
+import datetime
+import re
+from typing import Optional
+
 import psutil
 
+from ...pkg.kmsg.syncer import MatchResult, Syncer
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 from ..metrics_util import ComponentGauges
 
 NAME = "memory"
+
+_OOM_RULES = (
+    ("memory_oom", re.compile(r"Out of memory: Killed process"), "Warning"),
+    ("memory_oom_cgroup", re.compile(r"Memory cgroup out of memory"), "Warning"),
+    ("memory_oom_kill_constraint", re.compile(r"oom-kill:constraint="), "Warning"),
+    ("memory_edac_correctable", re.compile(r"EDAC MC\d+: \d+ CE"), "Warning"),
+    ("memory_edac_uncorrectable", re.compile(r"EDAC MC\d+: \d+ UE"), "Critical"),
+)
+
+
+def match_memory_kmsg(line: str) -> Optional[MatchResult]:
+    for name, rx, event_type in _OOM_RULES:
+        if rx.search(line):
+            return MatchResult(name=name, event_type=event_type, message=line)
+    return None
 
 
 class MemoryComponent(TickerComponent):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
+        self._bucket = (
+            inst.event_store.bucket(NAME) if inst.event_store is not None else None
+        )
+        self._kmsg = inst.kmsg_reader
+        self._syncer: Optional[Syncer] = None
 
     @property
     def name(self) -> str:
@@ -24,6 +50,14 @@ class MemoryComponent(TickerComponent):
 
     def tags(self) -> list:
         return [NAME]
+
+    def start(self) -> None:
+        if self._kmsg is not None and self._bucket is not None:
+            self._syncer = Syncer(self._kmsg, match_memory_kmsg, self._bucket)
+        super().start()
+
+    def events(self, since: datetime.datetime):
+        return self._bucket.get(since) if self._bucket is not None else []
 
     def check(self) -> CheckResult:
         vm = psutil.virtual_memory()

@@ -308,11 +308,60 @@ CATALOG: List[Detail] = [
         EventType.WARNING,
         _APP,
     ),
+    # ---- GPU self-test failures ------------------------------------------
+    _d(
+        "amdgpu_ib_test_failed",
+        r"amdgpu.*\*ERROR\* (?:IB test failed|ring test failed) on (?P<ring>\S+)",
+        "amdgpu indirect-buffer/ring self-test failed — the engine did not "
+        "execute a trivial command buffer; driver or hardware fault",
+        EventType.CRITICAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_firmware_load_failed",
+        r"amdgpu.*(?:failed to load firmware|Failed to load gpu firmware|"
+        r"Direct firmware load for amdgpu.* failed)",
+        "amdgpu firmware load failure — ROCm/driver installation problem",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
+    # ---- host memory / machine-check (reference: memory component kmsg) ---
+    _d(
+        "memory_edac_uncorrectable",
+        r"EDAC .*\bUE\b|EDAC MC\d+: \d+ UE",
+        "Uncorrectable host (DIMM) memory error reported by EDAC",
+        EventType.FATAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "memory_edac_correctable",
+        r"EDAC .*\bCE\b|EDAC MC\d+: \d+ CE",
+        "Corrected host (DIMM) memory error reported by EDAC",
+        EventType.WARNING,
+        _IGNORE,
+    ),
+    _d(
+        "host_mce",
+        r"mce: \[Hardware Error\]",
+        "Host machine-check (MCE) hardware error logged",
+        EventType.CRITICAL,
+        _HW,
+    ),
     # ---- OOM (host memory pressure killing GPU jobs) -----------------------
     _d(
         "memory_oom_kill",
         r"Out of memory: Killed process (?P<pid>\d+)",
         "The kernel OOM-killer terminated a process",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "memory_oom_cgroup",
+        r"Memory cgroup out of memory: Killed process (?P<pid>\d+)",
+        "A cgroup memory limit killed a process",
         EventType.WARNING,
         _APP,
     ),

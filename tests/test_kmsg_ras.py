@@ -336,3 +336,45 @@ def test_inject_and_readback_real_kmsg():
         pytest.skip("kernel rate-limited the /dev/kmsg writes")
     res = ras_catalog.match(mine[-1].message)
     assert res is not None and res[0].name == "amdgpu_gpu_reset_begin"
+
+
+def test_dmesg_fixture_catalog_regression():
+    """Parse a captured mixed dmesg fixture; the catalog must match exactly
+    the expected error lines and none of the benign ones."""
+    fixture = os.path.join(
+        os.path.dirname(__file__), "testdata", "dmesg_mixed.txt"
+    )
+    hits = []
+    with open(fixture) as f:
+        for raw in f:
+            m = parse_line(raw.rstrip("\n"), boot_time_epoch=0.0)
+            assert m is not None, raw
+            res = ras_catalog.match(m.message)
+            if res:
+                hits.append(res[0].name)
+    assert hits == [
+        "amdgpu_ring_timeout",
+        "amdgpu_gpu_reset_begin",
+        "amdgpu_gpu_reset_succeeded",
+        "amdgpu_ras_uncorrectable",
+        "amdgpu_ras_corrected_error",
+        "amdgpu_page_fault",
+        "kfd_evict_failed",
+        "amd_rccl_segfault_in_librccl",
+        "memory_oom_kill",
+        "memory_edac_correctable",
+        "amdgpu_gpu_reset_failed",
+        "amdgpu_ib_test_failed",
+    ]
+
+
+def test_memory_component_kmsg_matcher():
+    from gpud_amd.components.host.memory import match_memory_kmsg
+
+    r = match_memory_kmsg("Out of memory: Killed process 12 (x)")
+    assert r is not None and r.name == "memory_oom"
+    r = match_memory_kmsg("Memory cgroup out of memory: Killed process 5 (y)")
+    assert r is not None and r.name == "memory_oom_cgroup"
+    r = match_memory_kmsg("EDAC MC0: 2 UE on DIMM_B2")
+    assert r is not None and r.name == "memory_edac_uncorrectable"
+    assert match_memory_kmsg("nothing to see") is None
