@@ -32,6 +32,17 @@ DEFAULT_ZOMBIE_DEGRADED = 1000
 DEFAULT_ZOMBIE_UNHEALTHY = 2000
 
 
+def file_nr() -> tuple:
+    """(allocated, maximum) file handles from /proc/sys/fs/file-nr
+    (reference: os component fd usage)."""
+    try:
+        with open("/proc/sys/fs/file-nr") as f:
+            parts = f.read().split()
+            return int(parts[0]), int(parts[2])
+    except (OSError, IndexError, ValueError):
+        return 0, 0
+
+
 def count_process_states() -> dict:
     zombies = 0
     dstate = 0
@@ -102,6 +113,17 @@ class OSComponent(TickerComponent):
         self._gauges.set(
             "os_uptime_seconds", "Seconds since boot", pkghost.uptime_seconds()
         )
+        fd_alloc, fd_max = file_nr()
+        if fd_max > 0:
+            self._gauges.set(
+                "os_file_handles_allocated", "System-wide allocated file handles",
+                fd_alloc,
+            )
+            self._gauges.set(
+                "os_file_handles_usage_percent",
+                "Allocated file handles as percent of the system maximum",
+                100.0 * fd_alloc / fd_max,
+            )
         # pstore kernel-panic scan (new findings become Fatal events)
         panic_findings = []
         if self._pstore is not None:
