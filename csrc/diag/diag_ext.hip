@@ -468,6 +468,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
 
   accfrag_t acc[8][4] = {};  // 8 row-frags x 4 col-frags of 16x16
   bf16x8 bfrag[4][2];        // per-K-tile B fragments (reused by all phases)
+  bf16x8 afrag[2][2][2];     // ping-pong: [phase&1][row-frag][k-half]
 
   // prologue: stage tile 0 (all 8 slots per wave)
   stage_slots(0, 0, 0, 4);
@@ -481,7 +482,11 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
     const bool has_next = kt + 1 < ntiles;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      // loads for this phase
+      // loads for this phase: B fragments + quadrant-0 A fragments land at
+      // the head of the tile; later quadrants' A fragments were issued
+      // during the PREVIOUS phase and have a whole MFMA burst of cover,
+      // so the partial lgkmcnt(4) below waits only for in-flight
+      // next-phase reads, not this phase's operands
       if (p == 0) {
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -489,22 +494,34 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
           for (int kh = 0; kh < 2; ++kh)
             bfrag[c][kh] = read_b(cur, wave_n * 64 + c * 16, kh);
         }
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+#pragma unroll
+          for (int kh = 0; kh < 2; ++kh)
+            afrag[0][rr][kh] = read_a(cur, wave_m * 128 + rr * 16, kh);
+        }
       }
-      bf16x8 afrag[2][2];
+      if (p < 3) {
+        // pipeline: issue phase p+1's A fragments before this burst
 #pragma unroll
-      for (int rr = 0; rr < 2; ++rr) {
+        for (int rr = 0; rr < 2; ++rr) {
 #pragma unroll
-        for (int kh = 0; kh < 2; ++kh)
-          afrag[rr][kh] =
-              read_a(cur, wave_m * 128 + p * 32 + rr * 16, kh);
+          for (int kh = 0; kh < 2; ++kh)
+            afrag[(p + 1) & 1][rr][kh] =
+                read_a(cur, wave_m * 128 + (p + 1) * 32 + rr * 16, kh);
+        }
       }
       // prefetch next tile early: 4 slots in phase 0, 4 in phase 1
       // (all 8 in phase 0 congests it: measured 1154 -> 1001 TF @8192;
       // 2/phase leaves the last pair under-covered at the boundary drain)
       if (has_next && p < 2) stage_slots(nxt, kt + 1, p * 4, 4);
-      // rendezvous + MFMA burst (template: barrier, lgkmcnt(0), prio 1)
-      __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      // single barrier per phase (trailing only; the leading barrier of
+      // the template measured -8%: 1119 -> 1203 TF @4096 without it)
+      if (p < 3) {
+        asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
       __builtin_amdgcn_s_setprio(1);
       // kh outer: 8 independent MFMAs between accumulator reuses (the
       // dependent-accumulator latency exceeds the issue interval)
@@ -515,7 +532,8 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
 #pragma unroll
           for (int c = 0; c < 4; ++c) {
             acc[p * 2 + rr][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag[rr][kh], bfrag[c][kh], acc[p * 2 + rr][c], 0, 0, 0);
+                afrag[p & 1][rr][kh], bfrag[c][kh], acc[p * 2 + rr][c], 0, 0,
+                0);
           }
         }
       }
