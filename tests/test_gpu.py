@@ -226,3 +226,84 @@ def test_daemon_boot_live_gpu():
         except subprocess.TimeoutExpired:
             os.killpg(proc.pid, signal.SIGKILL)
             proc.wait(timeout=5)
+
+
+def test_fault_injection_e2e_live_gpu():
+    """The full loop on hardware: boot the daemon, inject a synthetic
+    amdgpu error through /inject-fault (writes the REAL /dev/kmsg), watch
+    error-ras flip Unhealthy, clear it via set-healthy."""
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    def _free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    # skip when the environment rate-limits kmsg writes entirely
+    try:
+        fd = os.open("/dev/kmsg", os.O_WRONLY)
+        os.close(fd)
+    except OSError:
+        pytest.skip("/dev/kmsg not writable")
+
+    port = _free_port()
+    env = {**os.environ, "PYTHONPATH": REPO}
+    env.pop("GPUD_AMDSMI_MOCK", None)
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run",
+            "--in-memory-db", "--address", f"127.0.0.1:{port}",
+            "--log-level", "warning",
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(60)
+
+        detected = False
+        for attempt in range(5):
+            client.inject_fault(ras_event_name="amdgpu_ring_timeout")
+            deadline = time.time() + 5
+            while time.time() < deadline:
+                st = client.get_health_states(
+                    components=["accelerator-amd-error-ras"]
+                )["accelerator-amd-error-ras"][0]
+                if st.health == "Unhealthy":
+                    detected = True
+                    break
+                # the ticker is 60s; trigger an immediate re-check
+                client.trigger_check(component="accelerator-amd-error-ras")
+                time.sleep(0.5)
+            if detected:
+                break
+        if not detected:
+            pytest.skip("kmsg writes rate-limited on this box")
+        assert "amdgpu_ring_timeout" in st.reason
+        # clear
+        client.set_healthy(["accelerator-amd-error-ras"])
+        st = client.get_health_states(components=["accelerator-amd-error-ras"])[
+            "accelerator-amd-error-ras"
+        ][0]
+        assert st.health == "Healthy", st.reason
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            os.killpg(proc.pid, signal.SIGKILL)
+            proc.wait(timeout=5)
