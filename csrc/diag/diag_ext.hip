@@ -28,6 +28,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <pybind11/pybind11.h>
 
 #include <cstdint>
@@ -564,6 +565,221 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MX-fp8 GEMM stress: the 8-phase structure at the block-scaled fp8 rate
+// (mfma_scale_f32_16x16x128_f8f6f4, K=128 per instruction, ~2x the bf16
+// matrix rate — cdna_hip_programming.md §3). Unit E8M0 scales keep the
+// exact-verification property while exercising the full MX datapath with
+// LDS-staged operands. Same LDS budget as the bf16 kernel: fp8 rows of
+// 128 elements are again 128 B wide, so the staging slots, the swizzle
+// and the banking analysis carry over unchanged.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void gemm_fill_fp8_kernel(
+    unsigned char* __restrict__ A, unsigned char* __restrict__ Bt, int M,
+    int N, int K) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  size_t stride = (size_t)gridDim.x * blockDim.x;
+  size_t na = (size_t)M * K, nb = (size_t)N * K;
+  for (size_t x = i; x < na; x += stride) {
+    __hip_fp8_e4m3 v(a_val((int)(x / K)));
+    A[x] = v.__x;
+  }
+  for (size_t x = i; x < nb; x += stride) {
+    __hip_fp8_e4m3 v(b_val((int)(x / K)));
+    Bt[x] = v.__x;
+  }
+}
+
+__global__ __launch_bounds__(512, 2) void gemm_mxfp8_8phase_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 128;  // fp8: 128-wide K tile
+  __shared__ unsigned char lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+  const int unit_scale = 0x7F7F7F7F;  // E8M0 1.0 per 32-elem block
+
+  // staging: fp8 row of TK=128 elements is 128 B — identical slot scheme
+  // to the bf16 kernel (64 x 1 KiB slots, 8 per wave), same swizzle
+  auto stage_slots = [&](int buf, int kt, int s0, int nslots) {
+    const int k0 = kt * TK;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (j >= nslots) break;
+      const int s = wave * 8 + s0 + j;
+      const int img = s >> 5;
+      const int s_img = s & 31;
+      const int D = s_img * 1024 + lane * 16;
+      const int L = swz(D);
+      const int row = L >> 7;
+      const int k = L & 127;  // fp8: 1 B per element
+      const unsigned char* g =
+          img == 0 ? A + (size_t)(brow + row) * K + k0 + k
+                   : Bt + (size_t)(bcol + row) * K + k0 + k;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)(
+              &lds[buf][img * TM * TK] + s_img * 1024),
+          16, 0, 0);
+    }
+  };
+
+  // fragment: lane l holds row base+(l&15), k = (l>>4)*32 + e (32 B -> two
+  // 16-B reads at swizzled offsets)
+  auto read_frag = [&](int buf, int img, int rbase) -> i32x8 {
+    const int r = rbase + (lane & 15);
+    const unsigned char* base = &lds[buf][img * TM * TK];
+    const int b0 = swz(r * 128 + (lane >> 4) * 32);
+    const int b1 = swz(r * 128 + (lane >> 4) * 32 + 16);
+    const int4 lo = *(const int4*)(base + b0);
+    const int4 hi = *(const int4*)(base + b1);
+    i32x8 v;
+    v[0] = lo.x; v[1] = lo.y; v[2] = lo.z; v[3] = lo.w;
+    v[4] = hi.x; v[5] = hi.y; v[6] = hi.z; v[7] = hi.w;
+    return v;
+  };
+
+  accfrag_t acc[8][4] = {};
+  i32x8 bfrag[4];
+  i32x8 afrag[2][2];
+
+  stage_slots(0, 0, 0, 4);
+  stage_slots(0, 0, 4, 4);
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    const bool has_next = kt + 1 < ntiles;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if (p == 0) {
+#pragma unroll
+        for (int c = 0; c < 4; ++c)
+          bfrag[c] = read_frag(cur, 1, wave_n * 64 + c * 16);
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
+          afrag[0][rr] = read_frag(cur, 0, wave_m * 128 + rr * 16);
+      }
+      if (p < 3) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
+          afrag[(p + 1) & 1][rr] =
+              read_frag(cur, 0, wave_m * 128 + (p + 1) * 32 + rr * 16);
+      }
+      if (has_next && p < 2) stage_slots(nxt, kt + 1, p * 4, 4);
+      if (p < 3) {
+        asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int rr = 0; rr < 2; ++rr) {
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          acc[p * 2 + rr][c] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              afrag[p & 1][rr], bfrag[c], acc[p * 2 + rr][c],
+              /*cbsz fp8*/ 0, /*blgp fp8*/ 0,
+              /*opsel_a*/ 0, unit_scale, /*opsel_b*/ 0, unit_scale);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      if (p == 3) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row =
+            brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
+py::dict gemm_stress_mxfp8(int size, int iters) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  unsigned char *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_fp8_kernel, dim3(2048), dim3(256), 0, 0, d_a,
+                     d_bt, M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_mxfp8_8phase_kernel, dim3(blocks), dim3(512), 0, 0,
+                     d_a, d_bt, d_c, M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_mxfp8_8phase_kernel, dim3(blocks), dim3(512), 0,
+                       0, d_a, d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "mxfp8_e4m3";
+  d["size"] = size;
+  d["structure"] = "256sq-8phase";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_bf16_v2(int size, int iters) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -881,6 +1097,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2", &gemm_stress_bf16_v2, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, 256^2 8-phase structure (swizzled LDS, setprio)");
+  m.def("gemm_stress_mxfp8", &gemm_stress_mxfp8, py::arg("size") = 8192,
+        py::arg("iters") = 5,
+        "MX-scaled fp8 GEMM stress, 8-phase structure (K=128 MFMAs)");
   m.def("hbm_bandwidth", &hbm_bandwidth, py::arg("buffer_gb") = 4.0,
         py::arg("iters") = 10, "float4 streaming triad + read over HBM3E");
   m.def("lds_bandwidth", &lds_bandwidth, py::arg("iters") = 100000,

@@ -40,7 +40,8 @@ NAME_FABRIC = "accelerator-amd-diag-fabric"
 DEFAULT_BF16_TFLOPS_FLOOR = 1600.0
 DEFAULT_FP8_TFLOPS_FLOOR = 1600.0  # non-scaled fp8 runs at the bf16 rate
 DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4570 TF
-DEFAULT_GEMM_TFLOPS_FLOOR = 800.0  # 8-phase bf16 GEMM, measured ~1100-1150
+DEFAULT_GEMM_TFLOPS_FLOOR = 800.0  # 8-phase bf16 GEMM, measured ~1200
+DEFAULT_GEMM_FP8_TFLOPS_FLOOR = 1400.0  # 8-phase MX-fp8 GEMM, measured ~1940
 DEFAULT_HBM_GBPS_FLOOR = 4500.0
 DEFAULT_LDS_TBPS_FLOOR = 60.0
 DEFAULT_XGMI_PAIR_GBPS_FLOOR = 30.0  # per direction, pairwise sendrecv
@@ -93,6 +94,7 @@ class MFMADiagComponent(_ManualDiagComponent):
         self.fp8_floor = DEFAULT_FP8_TFLOPS_FLOOR
         self.mxfp8_floor = DEFAULT_MXFP8_TFLOPS_FLOOR
         self.gemm_floor = DEFAULT_GEMM_TFLOPS_FLOOR
+        self.gemm_fp8_floor = DEFAULT_GEMM_FP8_TFLOPS_FLOOR
         self.iters = 2048
         self.workgroups = 1024
         self.gemm_size = 4096
@@ -126,15 +128,20 @@ class MFMADiagComponent(_ManualDiagComponent):
             gemm = diag.gemm_stress_bf16_v2(
                 size=self.gemm_size, iters=self.gemm_iters
             )
+            gemm_fp8 = diag.gemm_stress_mxfp8(
+                size=self.gemm_size, iters=self.gemm_iters
+            )
             extra[f"gpu{dev}.bf16_tflops"] = f"{bf16['tflops']:.0f}"
             extra[f"gpu{dev}.fp8_tflops"] = f"{fp8['tflops']:.0f}"
             extra[f"gpu{dev}.mxfp8_tflops"] = f"{mxfp8['tflops']:.0f}"
             extra[f"gpu{dev}.gemm_bf16_tflops"] = f"{gemm['tflops']:.0f}"
+            extra[f"gpu{dev}.gemm_mxfp8_tflops"] = f"{gemm_fp8['tflops']:.0f}"
             for name, res in (
                 ("bf16", bf16),
                 ("fp8", fp8),
                 ("mxfp8", mxfp8),
                 ("gemm", gemm),
+                ("gemm_fp8", gemm_fp8),
             ):
                 if not res["verified"]:
                     failures.append(
@@ -146,6 +153,7 @@ class MFMADiagComponent(_ManualDiagComponent):
                 ("fp8", fp8, self.fp8_floor),
                 ("mxfp8", mxfp8, self.mxfp8_floor),
                 ("gemm bf16", gemm, self.gemm_floor),
+                ("gemm mx-fp8", gemm_fp8, self.gemm_fp8_floor),
             ):
                 if res["tflops"] < floor:
                     failures.append(

@@ -99,6 +99,9 @@ def test_diag_gemm_bf16_verified():
     res2 = _diag.gemm_stress_bf16_v2(size=4096, iters=2)
     assert res2["verified"], res2
     assert res2["tflops"] > 800, res2
+    res3 = _diag.gemm_stress_mxfp8(size=4096, iters=2)
+    assert res3["verified"], res3
+    assert res3["tflops"] > 1400, res3
 
 
 def test_diag_hbm_bandwidth():
