@@ -203,6 +203,86 @@ __global__ __launch_bounds__(256, 4) void mfma_stress_mxfp8_kernel(
   out[blockIdx.x * blockDim.x + threadIdx.x] = s;
 }
 
+__global__ __launch_bounds__(256, 4) void mfma_stress_mxfp4_kernel(
+    float* __restrict__ out, int iters) {
+  // fp4 e2m1 1.0 = 0b0010 per nibble; cbsz/blgp = 4 selects the fp4 format
+  // (the ~10 PF dense path; fp6 runs at the same rate on CDNA4)
+  i32x8 a, b;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    a[i] = 0x22222222;
+    b[i] = 0x22222222;
+  }
+  const int unit_scale = 0x7F7F7F7F;
+  f32x16 acc[kAccums] = {};
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int u = 0; u < kInnerUnroll; ++u) {
+#pragma unroll
+      for (int j = 0; j < kAccums; ++j) {
+        acc[j] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            a, b, acc[j], /*cbsz fp4*/ 4, /*blgp fp4*/ 4,
+            0, unit_scale, 0, unit_scale);
+      }
+    }
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int j = 0; j < kAccums; ++j) s += acc[j][0];
+  out[blockIdx.x * blockDim.x + threadIdx.x] = s;
+}
+
+template <typename Kernel>
+py::dict run_mx_stress(Kernel kernel, int iters, int workgroups,
+                       const char* dtype) {
+  if (iters <= 0 || iters > (1 << 17)) throw std::invalid_argument("iters");
+  const int threads = 256;
+  float* d_out = nullptr;
+  const size_t out_elems = (size_t)workgroups * threads;
+  HIP_CHECK(hipMalloc(&d_out, out_elems * sizeof(float)));
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(kernel, dim3(workgroups), dim3(threads), 0, 0, d_out, 16);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  hipLaunchKernelGGL(kernel, dim3(workgroups), dim3(threads), 0, 0, d_out,
+                     iters);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  std::vector<float> host(out_elems);
+  HIP_CHECK(hipMemcpy(host.data(), d_out, out_elems * sizeof(float),
+                      hipMemcpyDeviceToHost));
+  const double expect = (double)kAccums * 64.0 * iters * kInnerUnroll;
+  size_t bad = 0;
+  for (size_t i = 0; i < out_elems; ++i)
+    if (host[i] != (float)expect) bad++;
+  HIP_CHECK(hipFree(d_out));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double waves = (double)workgroups * threads / 64.0;
+  const double mfmas = waves * (double)iters * kInnerUnroll * kAccums;
+  const double flops = mfmas * 2.0 * 32 * 32 * 64;
+  py::dict d;
+  d["dtype"] = dtype;
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds"] = ms * 1e-3;
+  d["workgroups"] = workgroups;
+  d["iters"] = iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
+py::dict mfma_stress_mxfp4(int iters, int workgroups) {
+  return run_mx_stress(mfma_stress_mxfp4_kernel, iters, workgroups,
+                       "mxfp4_e2m1");
+}
+
 py::dict mfma_stress_mxfp8(int iters, int workgroups) {
   if (iters <= 0 || iters > (1 << 17)) throw std::invalid_argument("iters");
   const int threads = 256;
@@ -1088,6 +1168,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("mfma_stress_fp8", &mfma_stress_fp8, py::arg("iters") = 4096,
         py::arg("workgroups") = 1024,
         "Register-resident fp8(e4m3) MFMA stress (non-scaled, bf16 rate)");
+  m.def("mfma_stress_mxfp4", &mfma_stress_mxfp4, py::arg("iters") = 2048,
+        py::arg("workgroups") = 1024,
+        "MX-scaled fp4 MFMA stress (the ~10 PF dense path)");
   m.def("mfma_stress_mxfp8", &mfma_stress_mxfp8, py::arg("iters") = 2048,
         py::arg("workgroups") = 1024,
         "MX-scaled fp8 MFMA stress (K=64 block-scaled, the ~5 PF dense path)");

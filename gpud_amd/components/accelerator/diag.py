@@ -39,7 +39,8 @@ NAME_FABRIC = "accelerator-amd-diag-fabric"
 # genuinely sick board trips them.
 DEFAULT_BF16_TFLOPS_FLOOR = 1600.0
 DEFAULT_FP8_TFLOPS_FLOOR = 1600.0  # non-scaled fp8 runs at the bf16 rate
-DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4570 TF
+DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4790 TF
+DEFAULT_MXFP4_TFLOPS_FLOOR = 6500.0  # fp4 MX path, measured ~8730 TF
 DEFAULT_GEMM_TFLOPS_FLOOR = 800.0  # 8-phase bf16 GEMM, measured ~1200
 DEFAULT_GEMM_FP8_TFLOPS_FLOOR = 1400.0  # 8-phase MX-fp8 GEMM, measured ~1940
 DEFAULT_HBM_GBPS_FLOOR = 4500.0
@@ -93,6 +94,7 @@ class MFMADiagComponent(_ManualDiagComponent):
         self.bf16_floor = DEFAULT_BF16_TFLOPS_FLOOR
         self.fp8_floor = DEFAULT_FP8_TFLOPS_FLOOR
         self.mxfp8_floor = DEFAULT_MXFP8_TFLOPS_FLOOR
+        self.mxfp4_floor = DEFAULT_MXFP4_TFLOPS_FLOOR
         self.gemm_floor = DEFAULT_GEMM_TFLOPS_FLOOR
         self.gemm_fp8_floor = DEFAULT_GEMM_FP8_TFLOPS_FLOOR
         self.iters = 2048
@@ -125,6 +127,9 @@ class MFMADiagComponent(_ManualDiagComponent):
             mxfp8 = diag.mfma_stress_mxfp8(
                 iters=self.iters, workgroups=self.workgroups
             )
+            mxfp4 = diag.mfma_stress_mxfp4(
+                iters=self.iters, workgroups=self.workgroups
+            )
             gemm = diag.gemm_stress_bf16_v2(
                 size=self.gemm_size, iters=self.gemm_iters
             )
@@ -134,12 +139,14 @@ class MFMADiagComponent(_ManualDiagComponent):
             extra[f"gpu{dev}.bf16_tflops"] = f"{bf16['tflops']:.0f}"
             extra[f"gpu{dev}.fp8_tflops"] = f"{fp8['tflops']:.0f}"
             extra[f"gpu{dev}.mxfp8_tflops"] = f"{mxfp8['tflops']:.0f}"
+            extra[f"gpu{dev}.mxfp4_tflops"] = f"{mxfp4['tflops']:.0f}"
             extra[f"gpu{dev}.gemm_bf16_tflops"] = f"{gemm['tflops']:.0f}"
             extra[f"gpu{dev}.gemm_mxfp8_tflops"] = f"{gemm_fp8['tflops']:.0f}"
             for name, res in (
                 ("bf16", bf16),
                 ("fp8", fp8),
                 ("mxfp8", mxfp8),
+                ("mxfp4", mxfp4),
                 ("gemm", gemm),
                 ("gemm_fp8", gemm_fp8),
             ):
@@ -152,6 +159,7 @@ class MFMADiagComponent(_ManualDiagComponent):
                 ("bf16", bf16, self.bf16_floor),
                 ("fp8", fp8, self.fp8_floor),
                 ("mxfp8", mxfp8, self.mxfp8_floor),
+                ("mxfp4", mxfp4, self.mxfp4_floor),
                 ("gemm bf16", gemm, self.gemm_floor),
                 ("gemm mx-fp8", gemm_fp8, self.gemm_fp8_floor),
             ):

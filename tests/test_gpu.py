@@ -86,6 +86,15 @@ def test_diag_mfma_mxfp8_verified():
     assert res["tflops"] > 3400, res
 
 
+def test_diag_mfma_mxfp4_verified():
+    from gpud_amd.diag import _diag
+
+    _diag.set_device(0)
+    res = _diag.mfma_stress_mxfp4(iters=1024, workgroups=1024)
+    assert res["verified"], res
+    assert res["tflops"] > 6500, res
+
+
 def test_diag_gemm_bf16_verified():
     from gpud_amd.diag import _diag
 
