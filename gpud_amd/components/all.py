@@ -19,6 +19,7 @@ from .accelerator import (
     gpm,
     gpu_counts,
     gpu_memory,
+    peer_mem,
     power,
     power_management,
     processes,
@@ -30,6 +31,7 @@ from .accelerator import (
 )
 from .host import (
     containerd,
+    infiniband,
     cpu,
     disk,
     docker,
@@ -58,6 +60,7 @@ def all_init_funcs() -> List[InitFunc]:
         gpu_memory.new,
         rccl.new,
         xgmi.new,
+        peer_mem.new,
         power.new,
         power_management.new,
         processes.new,
@@ -71,6 +74,7 @@ def all_init_funcs() -> List[InitFunc]:
         disk.new,
         docker.new,
         fuse.new,
+        infiniband.new,
         kernel_module.new,
         library.new,
         memory.new,

@@ -172,3 +172,63 @@ def test_xgmi_flap_auto_clear(mock_core):
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.DEGRADED  # still in the window
     assert "flapped" in cr.reason
+
+
+def test_infiniband_component_with_fixture(tmp_path, mock_core):
+    """sysfs fixture tree (reference pattern: class parser root param)."""
+    root = tmp_path / "infiniband"
+    for dev, port, state, rate, downed in [
+        ("mlx5_0", "1", "4: ACTIVE", "400 Gb/sec (4X NDR)", 0),
+        ("mlx5_1", "1", "1: DOWN", "400 Gb/sec (4X NDR)", 2),
+    ]:
+        pdir = root / dev / "ports" / port
+        (pdir / "counters").mkdir(parents=True)
+        (pdir / "state").write_text(state + "\n")
+        (pdir / "phys_state").write_text("5: LinkUp\n")
+        (pdir / "rate").write_text(rate + "\n")
+        (pdir / "counters" / "link_downed").write_text(f"{downed}\n")
+        (pdir / "counters" / "symbol_error").write_text("0\n")
+
+    comp = mock_core.registry.get("infiniband")
+    assert comp is not None
+    comp.sysfs_root = str(root)
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert "mlx5_1/1" in cr.reason
+
+    # port comes back up, but a rising link_downed counter records a flap
+    (root / "mlx5_1" / "ports" / "1" / "state").write_text("4: ACTIVE\n")
+    (root / "mlx5_1" / "ports" / "1" / "counters" / "link_downed").write_text("3\n")
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "flapped" in cr.reason
+    evs = comp.events(utcnow() - datetime.timedelta(minutes=5))
+    assert any(e.name == "ib_port_flap" for e in evs)
+
+
+def test_infiniband_expected_ports(mock_core, tmp_path):
+    comp = mock_core.registry.get("infiniband")
+    comp.sysfs_root = str(tmp_path / "none")
+    comp.expected_ports = 2
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    comp.expected_ports = 0
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+
+
+def test_peer_mem_component(mock_core):
+    comp = mock_core.registry.get("accelerator-amd-peer-mem")
+    assert comp is not None
+    comp.has_rdma_nics = lambda: True
+    comp.get_providers = lambda: ["amdgpu"]
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "amdgpu" in cr.reason
+    comp.get_providers = lambda: []
+    comp.has_dmabuf = lambda: False
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    comp.has_dmabuf = lambda: True
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
