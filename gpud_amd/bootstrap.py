@@ -178,6 +178,13 @@ def build_core(
         # wire the per-check duration histogram (our addition)
         if hasattr(c, "set_duration_observer"):
             c.set_duration_observer(recorder.observe_check_duration)
+        # apply the configured poll interval to auto components
+        if (
+            cfg.poll_interval_seconds > 0
+            and getattr(c, "run_mode", "auto") != "manual"
+            and hasattr(c, "poll_interval")
+        ):
+            c.poll_interval = cfg.poll_interval_seconds
 
     return DaemonCore(
         config=cfg,
