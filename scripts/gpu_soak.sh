@@ -30,7 +30,10 @@ try:
         states = c.get_health_states()
         c.trigger_check(tag="amd")
         if i % 10 == 0:
-            c.inject_fault(ras_event_name="amdgpu_ring_timeout")
+            try:
+                c.inject_fault(ras_event_name="amdgpu_ring_timeout")
+            except Exception:
+                pass  # /dev/kmsg absent on some boxes -> expected 400
         c.prometheus_metrics()
         i += 1
         time.sleep(1.0)
