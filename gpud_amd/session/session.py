@@ -307,15 +307,31 @@ class Session:
         except Exception as e:
             return {"packages": [], "error": str(e)}
 
+    DIAG_COMPONENTS = (
+        "accelerator-amd-diag-mfma",
+        "accelerator-amd-diag-bandwidth",
+        "accelerator-amd-diag-fabric",
+    )
+
     def _m_diagnostic(self, payload: dict) -> dict:
-        """Run the active diagnostics (the reference uploads
-        nvidia-bug-report.sh; we run the CDNA4 diag components)."""
+        """Run the active diagnostics (the reference runs its diagnostic
+        asynchronously and uploads — diagnostic.go:48; we run the CDNA4
+        diag components). ``{"async": true}`` starts them in the
+        background — results land in each component's cached health state
+        (queryable via the ``states`` method or /v1/states)."""
+        if payload.get("async"):
+            def _run():
+                for name in self.DIAG_COMPONENTS:
+                    comp = self.core.registry.get(name)
+                    if comp is not None:
+                        comp.trigger_check()
+
+            threading.Thread(
+                target=_run, daemon=True, name="gpud-diagnostic"
+            ).start()
+            return {"status": "started", "components": list(self.DIAG_COMPONENTS)}
         results = {}
-        for name in (
-            "accelerator-amd-diag-mfma",
-            "accelerator-amd-diag-bandwidth",
-            "accelerator-amd-diag-fabric",
-        ):
+        for name in self.DIAG_COMPONENTS:
             comp = self.core.registry.get(name)
             if comp is None:
                 continue
