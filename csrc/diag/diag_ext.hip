@@ -796,6 +796,223 @@ __global__ __launch_bounds__(512, 2) void gemm_mxfp8_8phase_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// GEMM v3: quadrant-phase schedule — the template's deep pipeline. All 8
+// waves compute ONE 128x128 C-quadrant per phase (Q00->Q01->Q11->Q10), so
+// each phase needs at most one NEW operand half-tile; halves are staged in
+// consumption order (A0,B0,B1,A1) one per phase and retired by a per-phase
+// s_waitcnt vmcnt(4) — in steady state NOTHING ever drains to vmcnt(0):
+//   phase p0: issue A0(t+1) -> wait retires B1(t)  (used p1)
+//   phase p1: issue B0(t+1) -> wait retires A1(t)  (used p2)
+//   phase p2: issue B1(t+1) -> wait retires A0(t+1) (early)
+//   phase p3: issue A1(t+1) -> wait retires B0(t+1) (used t+1.p0)
+// A retired half becomes cross-wave visible after that phase's trailing
+// barrier, one phase before its first reader. Per-phase ds_reads follow
+// the template's 12/4/8/4 pattern (A reloads at p0/p2, B at p0/p1/p3).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v3_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  __shared__ __hip_bfloat16 lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = wave >> 2;  // 0..1: 64-row band within the quadrant
+  const int wn = wave & 3;   // 0..3: 32-col band within the quadrant
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  // stage one half (h: 0=A rows 0-127, 1=B rows 0-127, 2=B rows 128-255,
+  // 3=A rows 128-255): 16 KiB = 16 slots, wave w issues slots {2w, 2w+1}
+  auto stage_half = [&](int buf, int kt, int h) {
+    const int img = (h == 0 || h == 3) ? 0 : 1;
+    const int row0 = (h == 0 || h == 1) ? 0 : 128;
+    const int k0 = kt * TK;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int s = wave * 2 + j;                  // slot within the half
+      const int img_slot = (row0 >> 3) + s;        // slot within the image
+      const int D = img_slot * 1024 + lane * 16;
+      const int L = swz(D);
+      const int row = L >> 7;
+      const int k = (L & 127) >> 1;
+      const __hip_bfloat16* g =
+          img == 0 ? A + (size_t)(brow + row) * K + k0 + k
+                   : Bt + (size_t)(bcol + row) * K + k0 + k;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)(
+              &lds[buf][img * TM * TK] + img_slot * 512),
+          16, 0, 0);
+    }
+  };
+
+  auto read_a = [&](int buf, int rbase, int kh) -> bf16x8 {
+    const int r = rbase + (lane & 15);
+    const int off = swz(r * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][0] + off);
+  };
+  auto read_b = [&](int buf, int cbase, int kh) -> bf16x8 {
+    const int c = cbase + (lane & 15);
+    const int off = swz(c * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][TM * TK] + off);
+  };
+
+  // acc[p][rf][cf]: quadrant p, 4 row-frags x 2 col-frags of 16x16
+  accfrag_t acc[4][4][2] = {};
+  bf16x8 afrag[4][2];  // 4 row-frags x 2 k-halves (reloaded at p0/p2)
+  bf16x8 bfrag[2][2];  // 2 col-frags x 2 k-halves (reloaded at p0/p1/p3)
+  static constexpr int QM[4] = {0, 0, 1, 1};
+  static constexpr int QN[4] = {0, 1, 1, 0};
+
+  // prologue: A0+B0 of tile 0 fully landed; B1+A1 of tile 0 in flight
+  stage_half(0, 0, 0);
+  stage_half(0, 0, 1);
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+  stage_half(0, 0, 2);
+  stage_half(0, 0, 3);
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    const bool has_next = kt + 1 < ntiles;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      static constexpr int STAGE_H[4] = {0, 1, 2, 3};  // A0,B0,B1,A1
+      if (has_next) stage_half(nxt, kt + 1, STAGE_H[p]);
+      // retire schedule (see header comment); on the last tile drain the
+      // remaining in-flight halves with counted waits instead
+      if (has_next) {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      } else if (p == 0) {
+        asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      } else if (p == 1) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      // per-phase fragment loads (A at p0/p2, B at p0/p1/p3)
+      if (p == 0 || p == 2) {
+#pragma unroll
+        for (int rf = 0; rf < 4; ++rf)
+#pragma unroll
+          for (int kh = 0; kh < 2; ++kh)
+            afrag[rf][kh] =
+                read_a(cur, QM[p] * 128 + wm * 64 + rf * 16, kh);
+      }
+      if (p != 2) {
+#pragma unroll
+        for (int cf = 0; cf < 2; ++cf)
+#pragma unroll
+          for (int kh = 0; kh < 2; ++kh)
+            bfrag[cf][kh] =
+                read_b(cur, QN[p] * 128 + wn * 32 + cf * 16, kh);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+#pragma unroll
+        for (int rf = 0; rf < 4; ++rf) {
+#pragma unroll
+          for (int cf = 0; cf < 2; ++cf) {
+            acc[p][rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[rf][kh], bfrag[cf][kh], acc[p][rf][cf], 0, 0, 0);
+          }
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+#pragma unroll
+    for (int rf = 0; rf < 4; ++rf) {
+#pragma unroll
+      for (int cf = 0; cf < 2; ++cf) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int row = brow + QM[p] * 128 + wm * 64 + rf * 16 +
+                          (lane >> 4) * 4 + reg;
+          const int col =
+              bcol + QN[p] * 128 + wn * 32 + cf * 16 + (lane & 15);
+          C[(size_t)row * N + col] = acc[p][rf][cf][reg];
+        }
+      }
+    }
+  }
+}
+
+py::dict gemm_stress_bf16_v3(int size, int iters) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_bf16_v3_kernel, dim3(blocks), dim3(512), 0, 0, d_a,
+                     d_bt, d_c, M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_bf16_v3_kernel, dim3(blocks), dim3(512), 0, 0,
+                       d_a, d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s2 = 0; s2 < sample; ++s2)
+      idx[s2] = ((size_t)s2 * 2654435761u) % ((size_t)M * N);
+    for (int s2 = 0; s2 < sample; ++s2) {
+      HIP_CHECK(hipMemcpy(&host[s2], d_c + idx[s2], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s2] / N), j = (int)(idx[s2] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s2] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-quadrant-phase";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_mxfp8(int size, int iters) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -1180,6 +1397,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2", &gemm_stress_bf16_v2, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, 256^2 8-phase structure (swizzled LDS, setprio)");
+  m.def("gemm_stress_bf16_v3", &gemm_stress_bf16_v3, py::arg("size") = 8192,
+        py::arg("iters") = 5,
+        "bf16 GEMM stress, quadrant-phase deep pipeline (no boundary drain)");
   m.def("gemm_stress_mxfp8", &gemm_stress_mxfp8, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "MX-scaled fp8 GEMM stress, 8-phase structure (K=128 MFMAs)");
