@@ -180,6 +180,7 @@ def run(
 def scan(
     expected_gpu_count: int = typer.Option(0),
     mock: bool = typer.Option(False, help="use the mock SMI backend"),
+    output: str = typer.Option("table", help="table | json"),
 ):
     """One-shot health scan, no daemon/DB (reference: pkg/scan/scan.go:33)."""
     log_setup(level="warning")
@@ -198,23 +199,42 @@ def scan(
         worst = HealthStateType.HEALTHY
         for comp in core.registry.all_components():
             if not comp.is_supported():
-                rows.append((comp.name, "-", "not supported on this host"))
+                rows.append((comp.name, "-", "not supported on this host", None))
                 continue
             if getattr(comp, "run_mode", "") == "manual":
-                rows.append((comp.name, "-", "manual run mode (diag)"))
+                rows.append((comp.name, "-", "manual run mode (diag)", None))
                 continue
             cr = comp.trigger_check()
             h = cr.health_state_type()
-            rows.append((comp.name, h, cr.summary()[:90]))
+            rows.append((comp.name, h, cr.summary()[:90], cr))
             if h == HealthStateType.UNHEALTHY:
                 worst = h
             elif h == HealthStateType.DEGRADED and worst == HealthStateType.HEALTHY:
                 worst = h
-        width = max(len(r[0]) for r in rows)
-        for name, health, reason in rows:
-            mark = {"Healthy": "✔", "Degraded": "~", "Unhealthy": "✘"}.get(health, " ")
-            typer.echo(f"{mark} {name:<{width}}  {health:<10} {reason}")
-        typer.echo(f"\noverall: {worst}")
+        if output == "json":
+            out = {
+                "overall": worst,
+                "components": [
+                    {
+                        "component": name,
+                        "health": health,
+                        "reason": reason,
+                        **(
+                            {"states": [st.to_dict() for st in cr.health_states()]}
+                            if cr is not None
+                            else {}
+                        ),
+                    }
+                    for name, health, reason, cr in rows
+                ],
+            }
+            typer.echo(json.dumps(out, indent=1))
+        else:
+            width = max(len(r[0]) for r in rows)
+            for name, health, reason, _cr in rows:
+                mark = {"Healthy": "✔", "Degraded": "~", "Unhealthy": "✘"}.get(health, " ")
+                typer.echo(f"{mark} {name:<{width}}  {health:<10} {reason}")
+            typer.echo(f"\noverall: {worst}")
         raise typer.Exit(code=0 if worst != HealthStateType.UNHEALTHY else 1)
     finally:
         core.close()

@@ -113,3 +113,13 @@ def test_logout_cli(tmp_path):
     conn = open_ro(cfg.state_path)
     assert metadata.get_value(conn, metadata.KEY_TOKEN) == ""
     conn.close()
+
+
+def test_scan_json_output():
+    res = runner.invoke(app, ["scan", "--mock", "--output", "json"])
+    assert res.exit_code == 0, res.output
+    out = json.loads(res.output)
+    assert out["overall"] == "Healthy"
+    comps = {c["component"]: c for c in out["components"]}
+    assert comps["accelerator-amd-temperature"]["health"] == "Healthy"
+    assert "states" in comps["cpu"]
