@@ -31,14 +31,23 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 
 class _RankFilteredSMI:
-    """Expose only this rank's GPU through the Instance snapshot API."""
+    """Expose only this rank's GPU through the Instance API so every
+    component built on top of it (snapshots, process lists, link metrics)
+    monitors exactly one GPU per rank — fixed per-GPU work as N grows."""
 
     def __init__(self, inst, index: int):
         self._inst = inst
         self.index = index
         uuids = inst.device_uuids()
         self.uuid = uuids[index % len(uuids)] if uuids else ""
-        self.failure_injector = inst.failure_injector
+
+    @property
+    def failure_injector(self):
+        return self._inst.failure_injector
+
+    @failure_injector.setter
+    def failure_injector(self, fi):
+        self._inst.failure_injector = fi
 
     @property
     def exists(self):
@@ -72,6 +81,14 @@ class _RankFilteredSMI:
         if fi is not None:
             snap = Instance._apply_injection(self.uuid, snap, fi)
         return {self.uuid: snap}
+
+    @property
+    def driver_version(self):
+        return self._inst.driver_version
+
+    @property
+    def rocm_version(self):
+        return self._inst.rocm_version
 
     def shutdown(self):
         pass
@@ -116,17 +133,11 @@ def main() -> int:
     import psutil
 
     cfg = Config()
-    core = build_core(
-        cfg,
-        in_memory_db=True,
-        kmsg_writable=False,
-        record_reboot=False,
-    )
-    inst = core.smi_instance
+    # filter to this rank's GPU BEFORE building the core so every component
+    # is constructed against the single-GPU view
+    inst = smi_pkg.new()
     if inst.exists and inst.device_count() > 0:
-        filtered = _RankFilteredSMI(inst, local_rank)
-        core.shared_snapshots.smi = filtered
-        core.gpud_instance.smi = filtered
+        smi_for_core = _RankFilteredSMI(inst, local_rank)
         n_gpus_seen = 1
         data_source = "mock" if args.mock else "amdsmi"
     else:
@@ -141,8 +152,16 @@ def main() -> int:
                 file=sys.stderr,
             )
             return 1
+        smi_for_core = inst
         n_gpus_seen = inst.device_count()
         data_source = "mock"
+    core = build_core(
+        cfg,
+        in_memory_db=True,
+        smi_instance=smi_for_core,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
 
     accel_components = [
         c
