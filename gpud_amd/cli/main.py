@@ -241,6 +241,84 @@ def scan(
 
 
 @app.command()
+def diagnose(
+    output: str = typer.Option("", help="write the JSON report here"),
+    rocprof: bool = typer.Option(
+        False, help="also capture a rocprofv3 kernel trace of the MFMA stress"
+    ),
+    quick: bool = typer.Option(False, help="shorter kernel runs"),
+):
+    """Run the active CDNA4 diagnostics (MFMA/HBM/LDS/fabric) and report
+    per-GPU numbers — the DCGM-diag analog."""
+    log_setup(level="warning")
+    from ..bootstrap import build_core
+
+    core = build_core(
+        Config(), in_memory_db=True, kmsg_writable=False, record_reboot=False
+    )
+    try:
+        report = {"diagnostics": {}}
+        for name in (
+            "accelerator-amd-diag-mfma",
+            "accelerator-amd-diag-bandwidth",
+            "accelerator-amd-diag-fabric",
+        ):
+            comp = core.registry.get(name)
+            if comp is None:
+                continue
+            if quick and hasattr(comp, "iters"):
+                comp.iters = 512
+            if quick and hasattr(comp, "gemm_iters"):
+                comp.gemm_iters = 1
+            cr = comp.trigger_check()
+            report["diagnostics"][name] = {
+                "health": cr.health,
+                "reason": cr.reason,
+                "error": cr.error,
+                "measurements": cr.extra_info or {},
+            }
+            mark = {"Healthy": "✔", "Unhealthy": "✘"}.get(cr.health, "~")
+            typer.echo(f"{mark} {name}: {cr.health} — {cr.reason[:100]}")
+        if rocprof:
+            import shutil as _sh
+            import subprocess as _sp
+            import tempfile as _tf
+
+            if _sh.which("rocprofv3"):
+                d = _tf.mkdtemp(prefix="gpud-rocprof-")
+                res = _sp.run(
+                    [
+                        "rocprofv3", "--kernel-trace", "--stats", "-d", d,
+                        "--", sys.executable, "-c",
+                        "from gpud_amd.diag import _diag; _diag.set_device(0);"
+                        "print(_diag.mfma_stress_bf16(iters=512, workgroups=1024))",
+                    ],
+                    capture_output=True, text=True, timeout=300,
+                    cwd=os.path.dirname(os.path.dirname(os.path.dirname(
+                        os.path.abspath(__file__)))),
+                )
+                report["rocprof"] = {
+                    "output_dir": d,
+                    "rc": res.returncode,
+                    "stdout_tail": res.stdout[-1500:],
+                }
+                typer.echo(f"rocprof trace written under {d}")
+            else:
+                report["rocprof"] = {"error": "rocprofv3 not on PATH"}
+        overall = all(
+            v["health"] == "Healthy" for v in report["diagnostics"].values()
+        )
+        report["overall"] = "Healthy" if overall else "Unhealthy"
+        if output:
+            with open(output, "w") as f:
+                json.dump(report, f, indent=1)
+            typer.echo(f"report written to {output}")
+        raise typer.Exit(code=0 if overall else 1)
+    finally:
+        core.close()
+
+
+@app.command()
 def status(
     server_url: str = typer.Option("https://localhost:15132"),
 ):
