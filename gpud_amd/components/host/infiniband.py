@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, List
 
 from ...apiv1.types import (
     Event,

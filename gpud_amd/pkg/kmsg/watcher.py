@@ -45,12 +45,17 @@ class Watcher:
             logger.warning("cannot open %s: %s", self.path, e)
             return out
         try:
+            epipe_budget = 10_000  # ring churn bound: never spin forever
             while len(out) < limit:
                 try:
                     data = os.read(fd, _READ_SIZE)
                 except OSError as e:
                     if e.errno == errno.EPIPE:
-                        continue  # ring overwrote our position; keep reading
+                        # ring overwrote our position; keep reading (bounded)
+                        epipe_budget -= 1
+                        if epipe_budget <= 0:
+                            break
+                        continue
                     if e.errno == errno.EAGAIN:
                         break  # drained
                     raise
