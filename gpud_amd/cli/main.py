@@ -124,9 +124,24 @@ def run(
 
     session = None
     if endpoint:
+        # optional login on boot (reference: cmd/gpud/run/command.go:99-127)
+        if token:
+            from ..pkg.login import do_login
+
+            err = do_login(cfg, token=token, endpoint=endpoint)
+            if err:
+                typer.echo(f"login failed (continuing): {err}", err=True)
+        from ..pkg import metadata as _md
         from ..session import Session
 
-        session = Session(core, endpoint=endpoint, token=token)
+        machine_id = ""
+        try:
+            machine_id = _md.get_value(core.db_ro, _md.KEY_MACHINE_ID)
+        except Exception:
+            pass
+        session = Session(
+            core, endpoint=endpoint, token=token, machine_id=machine_id
+        )
         session.start()
 
     # package manager reconcile loops (reference: cmd/gpud/run:425-431)
