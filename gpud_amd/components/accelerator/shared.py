@@ -69,7 +69,13 @@ class SmiComponentMixin:
     """
 
     def smi_guard(self):
-        """Returns a CheckResult to short-circuit with, or None to proceed."""
+        """Returns a CheckResult to short-circuit with, or None to proceed.
+
+        Cached once the instance is known-good: existence cannot regress at
+        runtime (GPU-lost injection filters devices(), not the session), so
+        the healthy path costs one attribute read per check."""
+        if getattr(self, "_smi_guard_ok", False):
+            return None
         from ..base import CheckResult
         from ...apiv1.types import HealthStateType
 
@@ -88,4 +94,5 @@ class SmiComponentMixin:
                 reason="amdsmi library not loaded (no AMD GPU driver)",
                 error=err if err else "",
             )
+        self._smi_guard_ok = True
         return None

@@ -48,7 +48,11 @@ class ComponentGauges:
     def set(self, name: str, doc: str, value: float, **labels: str) -> None:
         if self.registry is None:
             return
-        key = (name, *sorted(labels.items()))
+        # hot path: most gauges carry zero or one extra label (uuid)
+        if len(labels) < 2:
+            key = (name, *labels.items())
+        else:
+            key = (name, *sorted(labels.items()))
         child = self._children.get(key)
         if child is None:
             g = self.gauge(name, doc, sorted(labels.keys()))

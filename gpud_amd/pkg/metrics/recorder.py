@@ -85,12 +85,17 @@ class Recorder:
             ),
         )
 
+        self._duration_children: dict = {}
         self._thread = threading.Thread(
             target=self._run, daemon=True, name="gpud-recorder"
         )
 
     def observe_check_duration(self, component: str, seconds: float) -> None:
-        self.check_duration.labels(**{LABEL_COMPONENT: component}).observe(seconds)
+        child = self._duration_children.get(component)
+        if child is None:
+            child = self.check_duration.labels(**{LABEL_COMPONENT: component})
+            self._duration_children[component] = child
+        child.observe(seconds)
 
     def record_once(self) -> None:
         if self._proc is not None:
