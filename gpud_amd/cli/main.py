@@ -175,6 +175,30 @@ def run(
 
     _threading.Thread(target=_version_watch, daemon=True).start()
 
+    # amdsmi-appears-later restart (reference: pkg/nvidia/nvml/
+    # instance.go:121-147 refreshNVMLAndExit — a daemon booted before the
+    # GPU driver exits cleanly once the library loads, so systemd's
+    # Restart=always brings it back with the accelerator components live)
+    if core.smi_instance is not None and not core.smi_instance.exists:
+        def _smi_watch():
+            from .. import smi as _smi
+
+            while not stop["flag"]:
+                time.sleep(60.0)
+                try:
+                    probe = _smi.new()
+                except Exception:
+                    continue
+                if probe.exists:
+                    probe.shutdown()
+                    typer.echo(
+                        "amdsmi became available; exiting for restart", err=True
+                    )
+                    stop["flag"] = True
+                    return
+
+        _threading.Thread(target=_smi_watch, daemon=True).start()
+
     def _sig(_s, _f):
         stop["flag"] = True
 
