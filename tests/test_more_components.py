@@ -232,3 +232,97 @@ def test_peer_mem_component(mock_core):
     comp.has_dmabuf = lambda: True
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.HEALTHY
+
+
+def test_power_at_limit_degraded(mock_core):
+    backend = mock_core.smi_instance._b
+    backend.state[0]["power_w"] = 1390  # ≥98% of 1400
+    mock_core.shared_snapshots.refresh()
+    cr = mock_core.registry.get("accelerator-amd-power").trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "power limit" in cr.reason
+
+
+def test_gpu_memory_component_values(mock_core):
+    from gpud_amd.pkg.metrics import Scraper
+
+    mock_core.registry.get("accelerator-amd-memory").trigger_check()
+    scraped = Scraper(mock_core.metrics_registry).scrape()
+    by_name = {}
+    for m in scraped:
+        if m.name == "accelerator_amd_memory_total_bytes":
+            by_name[m.labels.get("uuid")] = m.value
+    assert by_name  # per-uuid totals present
+    assert all(v == 294_912 * 1024 * 1024 for v in by_name.values())  # 288 GB
+
+
+def test_processes_component_lists(mock_core):
+    cr = mock_core.registry.get("accelerator-amd-processes").trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "process" in cr.reason
+    # pids surfaced in extra info
+    assert any(k.endswith(".pids") for k in (cr.extra_info or {}))
+
+
+def test_error_ras_noncritical_events_stay_healthy(mock_core):
+    """Warning-class catalog events (e.g. reset succeeded) never flip the
+    state machine."""
+    from gpud_amd.apiv1.types import Event, EventType
+
+    comp = mock_core.registry.get("accelerator-amd-error-ras")
+    bucket = mock_core.event_store.bucket("accelerator-amd-error-ras")
+    bucket.insert(
+        Event(
+            time=utcnow(),
+            component=comp.name,
+            name="amdgpu_gpu_reset_succeeded",
+            type=EventType.WARNING,
+            message="GPU reset(1) succeeded",
+        )
+    )
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+
+
+def test_error_ras_reboot_clears_state(mock_core):
+    """A reboot AFTER the critical event clears the unhealthy state
+    (reference: reboot events reset error states — pkg/host/event.go)."""
+    from gpud_amd.apiv1.types import Event, EventType
+    from gpud_amd.pkg.host import EVENT_NAME_REBOOT, REBOOT_BUCKET
+
+    comp = mock_core.registry.get("accelerator-amd-error-ras")
+    bucket = mock_core.event_store.bucket("accelerator-amd-error-ras")
+    os_bucket = mock_core.event_store.bucket(REBOOT_BUCKET)
+    now = utcnow()
+    bucket.insert(
+        Event(
+            time=now - datetime.timedelta(hours=2),
+            component=comp.name,
+            name="amdgpu_ring_timeout",
+            type=EventType.CRITICAL,
+            message="timeout",
+        )
+    )
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    os_bucket.insert(
+        Event(
+            time=now - datetime.timedelta(hours=1),
+            component="os",
+            name=EVENT_NAME_REBOOT,
+            type=EventType.WARNING,
+            message="reboot",
+        )
+    )
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+
+
+def test_scraper_histogram_series(mock_core):
+    """The check-duration histogram surfaces bucket/sum/count series."""
+    from gpud_amd.pkg.metrics import Scraper
+
+    mock_core.registry.get("cpu").trigger_check()
+    names = {m.name for m in Scraper(mock_core.metrics_registry).scrape()}
+    assert "gpud_component_check_duration_seconds_bucket" in names
+    assert "gpud_component_check_duration_seconds_sum" in names
