@@ -52,6 +52,10 @@ class Config:
     # accelerator knobs
     expected_gpu_count: int = 0
     expected_xgmi_link_count: int = 0  # 7 per GPU on an 8-GPU MI355X node
+    expected_ib_ports: int = 0
+    expected_ib_rate_gbps: float = 0.0
+    # network-latency probe targets: list of (host, port)
+    latency_targets: List[Any] = field(default_factory=list)
     temperature_margin_threshold_c: float = 10.0
     zombie_degraded_threshold: int = 1000
     zombie_unhealthy_threshold: int = 2000

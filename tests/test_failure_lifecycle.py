@@ -1,0 +1,124 @@
+"""Integration story: a GPU degrades over its lifecycle and the daemon
+tracks every stage — CE creep, UE, kernel RAS event, reboot, recurrence
+escalation, manual clear. Exercises ecc + error-ras + event store + reboot
+store + kmsg replay together (the flagship detection pipeline)."""
+
+import datetime
+
+import pytest
+
+from gpud_amd.apiv1.types import Event, EventType, HealthStateType, RepairActionType, utcnow
+from gpud_amd.pkg.host import EVENT_NAME_REBOOT, REBOOT_BUCKET
+from gpud_amd.pkg.kmsg.parser import Message
+
+
+@pytest.fixture()
+def core(monkeypatch, tmp_path):
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK_GPUS", "2")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    c = build_core(
+        Config(data_dir=str(tmp_path)),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    yield c
+    c.close()
+
+
+def test_gpu_failure_lifecycle(core):
+    backend = core.smi_instance._b
+    uuid = core.smi_instance.device_uuids()[0]
+    ecc = core.registry.get("accelerator-amd-ecc")
+    ras = core.registry.get("accelerator-amd-error-ras")
+    os_bucket = core.event_store.bucket(REBOOT_BUCKET)
+    t0 = utcnow() - datetime.timedelta(hours=12)
+
+    # stage 0: healthy
+    core.shared_snapshots.refresh()
+    assert ecc.trigger_check().health == HealthStateType.HEALTHY
+    assert ras.trigger_check().health == HealthStateType.HEALTHY
+
+    # stage 1: correctable errors creep up -> still healthy, but an event
+    backend.state[0]["ecc_correctable"] = 3
+    core.shared_snapshots.refresh()
+    cr = ecc.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+    backend.state[0]["ecc_correctable"] = 9
+    core.shared_snapshots.refresh()
+    ecc.trigger_check()
+    evs = ecc.events(t0)
+    assert any(e.name == "amd_ecc_correctable_increase" for e in evs)
+
+    # stage 2: the kernel logs a RAS UE (replayed through the kmsg matcher
+    # exactly as the live follower would deliver it)
+    from gpud_amd.components.accelerator.error_ras import _match
+    from gpud_amd.pkg.kmsg.syncer import Syncer
+    from gpud_amd.pkg.kmsg.watcher import Watcher
+
+    ras_bucket = core.event_store.bucket("accelerator-amd-error-ras")
+    syncer = Syncer(Watcher(path="/nonexistent"), _match, ras_bucket)
+    syncer.replay(
+        [
+            Message(
+                message="amdgpu 0000:0a:00.0: amdgpu: uncorrectable hardware "
+                "error(ERREVENT_ATHUB_INTERRUPT) detected!",
+                time=utcnow() - datetime.timedelta(hours=10),
+            )
+        ]
+    )
+    cr = ras.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert RepairActionType.REBOOT_SYSTEM in cr.suggested_actions.repair_actions
+
+    # stage 3: UE also visible in the SMI counters
+    backend.state[0]["ecc_uncorrectable"] = 2
+    core.shared_snapshots.refresh()
+    cr = ecc.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert uuid in cr.reason
+
+    # stage 4: operator reboots -> RAS state machine clears
+    os_bucket.insert(
+        Event(
+            time=utcnow() - datetime.timedelta(hours=8),
+            component="os",
+            name=EVENT_NAME_REBOOT,
+            type=EventType.WARNING,
+            message="reboot",
+        )
+    )
+    assert ras.trigger_check().health == HealthStateType.HEALTHY
+
+    # stage 5: the UE comes BACK after a second reboot -> escalation to
+    # hardware inspection (reference xid/health_state.go:61-97 semantics)
+    os_bucket.insert(
+        Event(
+            time=utcnow() - datetime.timedelta(hours=4),
+            component="os",
+            name=EVENT_NAME_REBOOT,
+            type=EventType.WARNING,
+            message="reboot",
+        )
+    )
+    syncer.replay(
+        [
+            Message(
+                message="amdgpu 0000:0a:00.0: amdgpu: uncorrectable hardware "
+                "error(ERREVENT_ATHUB_INTERRUPT) detected!",
+                time=utcnow(),
+            )
+        ]
+    )
+    cr = ras.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert cr.suggested_actions.repair_actions == [
+        RepairActionType.HARDWARE_INSPECTION
+    ]
+
+    # stage 6: after the board swap, the operator clears the state
+    ras.set_healthy()
+    assert ras.last_check_result().health == HealthStateType.HEALTHY
