@@ -76,14 +76,26 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
     def _active_throttlers(self, uuid: str, v: Dict) -> List[Tuple[str, str]]:
         out = []
         prev = self._last_acc.get(uuid, {})
-        cur: Dict[str, int] = {}
+        cur: Dict[str, int] = {"_counter": int(v.get("acc_counter", 0))}
+        d_counter = cur["_counter"] - prev.get("_counter", 0)
         for key, desc in _THROTTLERS:
             acc = int(v.get(f"acc_{key}", 0))
             cur[key] = acc
             explicit = int(v.get(f"active_{key}", 0))
-            rising = key in prev and acc > prev[key]
+            delta = acc - prev.get(key, acc)
+            rising = key in prev and delta > 0
             if explicit or rising:
                 out.append((key, desc))
+            # violation percentage over the window between polls
+            # (PVIOL/TVIOL per the amdsmi gpu_metrics accumulator contract:
+            # delta residency / delta accumulation counter * 100)
+            if key in prev and d_counter > 0:
+                self._gauges.set(
+                    f"accelerator_amd_throttle_{key}_violation_percent",
+                    f"{key} residency percent between polls (PVIOL/TVIOL-style)",
+                    100.0 * max(delta, 0) / d_counter,
+                    uuid=uuid,
+                )
         self._last_acc[uuid] = cur
         return out
 

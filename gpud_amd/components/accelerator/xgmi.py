@@ -55,6 +55,7 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
             getattr(cfg, "expected_xgmi_link_count", 0) if cfg else 0
         )
         self._last_states: Dict[str, List[int]] = {}
+        self._last_traffic: Dict = {}
         self.get_snapshots: Callable = (
             self._shared.get if self._shared is not None else lambda: {}
         )
@@ -133,27 +134,53 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
                     int(err),
                     uuid=uuid,
                 )
-        # per-link traffic counters (separate, heavier SMI call)
+        # per-link traffic counters (separate, heavier SMI call) + derived
+        # per-second rates from deltas between this component's polls
+        import time as _time
+
         try:
+            now_mono = _time.monotonic()
             for uuid, dev in self.get_devices().items():
                 lm = dev.link_metrics()
                 for i, link in enumerate(lm.get("links", [])):
                     if int(link.get("link_type", 0)) != 2:  # XGMI only
                         continue
+                    read_kb = float(link.get("read_kb", 0))
+                    write_kb = float(link.get("write_kb", 0))
                     self._gauges.set(
                         "accelerator_amd_xgmi_read_kb_total",
                         "Accumulated xGMI read traffic (KB)",
-                        float(link.get("read_kb", 0)),
+                        read_kb,
                         uuid=uuid,
                         link=str(i),
                     )
                     self._gauges.set(
                         "accelerator_amd_xgmi_write_kb_total",
                         "Accumulated xGMI write traffic (KB)",
-                        float(link.get("write_kb", 0)),
+                        write_kb,
                         uuid=uuid,
                         link=str(i),
                     )
+                    key = (uuid, i)
+                    prev = self._last_traffic.get(key)
+                    if prev is not None:
+                        dt = now_mono - prev[2]
+                        if dt > 0:
+                            self._gauges.set(
+                                "accelerator_amd_xgmi_read_bytes_per_second",
+                                "xGMI read rate between polls",
+                                max(read_kb - prev[0], 0) * 1024.0 / dt,
+                                uuid=uuid,
+                                link=str(i),
+                            )
+                            self._gauges.set(
+                                "accelerator_amd_xgmi_write_bytes_per_second",
+                                "xGMI write rate between polls",
+                                max(write_kb - prev[1], 0) * 1024.0 / dt,
+                                uuid=uuid,
+                                link=str(i),
+                            )
+                    self._last_traffic[key] = (read_kb, write_kb, now_mono)
         except Exception:
             pass  # traffic counters are best-effort
 
