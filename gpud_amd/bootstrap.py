@@ -119,7 +119,7 @@ def build_core(
     syncer = Syncer(
         scraper,
         metrics_store,
-        sync_interval_seconds=60.0,
+        sync_interval_seconds=cfg.metrics_sync_interval_seconds,
         retention=datetime.timedelta(days=cfg.metrics_retention_days),
     )
     recorder = Recorder(registry_prom, db_rw=db_rw)

@@ -360,6 +360,7 @@ def diagnose(
 @app.command()
 def status(
     server_url: str = typer.Option("https://localhost:15132"),
+    watch: float = typer.Option(0.0, help="re-poll every N seconds"),
 ):
     """Query a running daemon's health states (reference: gpud status)."""
     from ..client import Client
@@ -368,11 +369,20 @@ def status(
     if not c.wait_healthz(timeout=5):
         typer.echo("daemon not reachable", err=True)
         raise typer.Exit(code=1)
-    states = c.get_health_states()
-    for comp, sts in sorted(states.items()):
-        for s in sts:
-            typer.echo(f"{comp:45s} {s.health:<12} {s.reason[:80]}")
-    c.close()
+    try:
+        while True:
+            states = c.get_health_states()
+            for comp, sts in sorted(states.items()):
+                for s in sts:
+                    typer.echo(f"{comp:45s} {s.health:<12} {s.reason[:80]}")
+            if watch <= 0:
+                break
+            typer.echo("-" * 80)
+            time.sleep(watch)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        c.close()
 
 
 @app.command()

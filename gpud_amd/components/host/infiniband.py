@@ -134,7 +134,6 @@ class InfinibandComponent(TickerComponent):
                 )
             return CheckResult(NAME, reason="no InfiniBand/RoCE devices")
         active, down, slow = [], [], []
-        flapped: List[str] = []
         extra = {}
         for p in ports:
             key = f"{p['device']}/{p['port']}"
@@ -169,7 +168,6 @@ class InfinibandComponent(TickerComponent):
                 and cur.get("link_downed", 0) > prev.get("link_downed", 0)
                 and self._bucket is not None
             ):
-                flapped.append(key)
                 self._bucket.insert(
                     Event(
                         time=utcnow(),

@@ -33,6 +33,7 @@ class Config:
     enable_auto_update: bool = False
     auto_update_exit_code: int = -1
     poll_interval_seconds: float = 60.0
+    metrics_sync_interval_seconds: float = 60.0
     # component selection (reference: component enable/disable list)
     enabled_components: List[str] = field(default_factory=list)
     disabled_components: List[str] = field(default_factory=list)
