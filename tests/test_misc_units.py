@@ -188,3 +188,111 @@ def test_host_boot_helpers():
     assert (utcnow() - bt).total_seconds() == pytest.approx(
         host.uptime_seconds(), rel=0.1
     )
+
+
+def test_metrics_store_component_filter(mem_db):
+    from gpud_amd.pkg.metrics.scraper import ScrapedMetric
+    from gpud_amd.pkg.metrics.store import MetricsStore
+
+    rw, ro = mem_db
+    store = MetricsStore(rw, ro)
+    store.record(
+        [
+            ScrapedMetric(unix_ms=1000, component="a", name="m1", value=1),
+            ScrapedMetric(unix_ms=1000, component="b", name="m1", value=2),
+        ]
+    )
+    only_a = store.read(components=["a"])
+    assert list(only_a.keys()) == ["a"]
+    since = store.read(
+        since=datetime.datetime.fromtimestamp(2, tz=datetime.timezone.utc)
+    )
+    assert since == {}
+
+
+def test_eventstore_purge_thread(mem_db):
+    from gpud_amd.pkg.eventstore import Store
+
+    rw, ro = mem_db
+    store = Store(rw, ro, retention=datetime.timedelta(seconds=2))
+    bucket = store.bucket("purge-thread-test")
+    old = utcnow() - datetime.timedelta(seconds=30)
+    bucket.insert(Event(time=old, name="old", type="Info", message="x"))
+    deadline = time.time() + 6
+    while time.time() < deadline:
+        if not bucket.get(utcnow() - datetime.timedelta(days=1)):
+            break
+        time.sleep(0.2)
+    assert not bucket.get(utcnow() - datetime.timedelta(days=1))
+    store.close()
+
+
+def test_plugin_component_interval_applied():
+    from gpud_amd.pkg import custom_plugins as cp
+
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "interval-check",
+            "plugin_type": "component",
+            "interval": "5m",
+            "health_state_plugin": {
+                "steps": [{"run_bash_script": {"script": "true"}}]
+            },
+        }
+    )
+    comp = cp.make_components(spec)[0]
+    assert comp.poll_interval == 300.0
+
+
+def test_detect_provider_offline():
+    from gpud_amd.pkg.machine_info import detect_provider
+
+    assert detect_provider(timeout=0.2) is None
+
+
+def test_mock_device_full_surface():
+    from gpud_amd.smi import Instance
+    from gpud_amd.smi.mock import MockBackend
+
+    inst = Instance(backend=MockBackend(num_gpus=1))
+    dev = next(iter(inst.devices().values()))
+    assert dev.vbios_info()["version"]
+    assert dev.vram_info()["vram_size_bytes"] == 294_912 * 1024 * 1024
+    lm = dev.link_metrics()
+    assert lm["num_links"] == 7
+    assert dev.temp_metric(1, 0) > 0  # hotspot current
+    assert dev.ecc_count_block(1)["uncorrectable"] == 0
+    assert dev.bad_page_info()["threshold"] == 256
+    assert dev.power_management_enabled()
+
+
+def test_registry_init_fn_failure_logged(monkeypatch, tmp_path):
+    """A component whose init raises must not break bootstrap."""
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    import gpud_amd.components.all as allmod
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    def boom(_inst):
+        raise RuntimeError("init exploded")
+
+    orig = allmod.all_init_funcs
+
+    def patched():
+        return [boom] + orig()
+
+    monkeypatch.setattr(allmod, "all_init_funcs", patched)
+    # bootstrap imports the symbol at module load; patch there too
+    import gpud_amd.bootstrap as bs
+
+    monkeypatch.setattr(bs, "all_init_funcs", patched)
+    core = build_core(
+        Config(data_dir=str(tmp_path)),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    try:
+        assert core.registry.get("cpu") is not None  # others registered fine
+    finally:
+        core.close()
