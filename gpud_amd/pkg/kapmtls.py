@@ -100,6 +100,36 @@ class Manager:
             return "no previous credentials recorded"
         return self.activate(prev)
 
+    def active_paths(self) -> Optional[Dict[str, str]]:
+        """Cert/key paths of the ACTIVE credentials, or None."""
+        v = self.active_version()
+        if not v:
+            return None
+        d = self._versioned_dir(v)
+        return {
+            "cert": os.path.join(d, "client.crt"),
+            "key": os.path.join(d, "client.key"),
+        }
+
+    def grpc_channel_credentials(self, root_ca_pem: Optional[bytes] = None):
+        """Build gRPC mTLS channel credentials from the active credentials
+        (feeds the v2 session — reference: kap-mTLS staged client certs for
+        the node-local agent)."""
+        import grpc
+
+        paths = self.active_paths()
+        if paths is None:
+            return None
+        with open(paths["cert"], "rb") as f:
+            cert = f.read()
+        with open(paths["key"], "rb") as f:
+            key = f.read()
+        return grpc.ssl_channel_credentials(
+            root_certificates=root_ca_pem,
+            private_key=key,
+            certificate_chain=cert,
+        )
+
     def status(self) -> Dict[str, str]:
         active = self.active_version()
         st = {

@@ -107,3 +107,30 @@ def test_kapmtls_session_methods(mock_core):
     assert resp["data"]["active_version"] == "5"
     resp = s.process_request({"req_id": "k4", "method": "nodeCredentials", "data": {}})
     assert "machine_id" in resp["data"]
+
+
+def test_kapmtls_grpc_credentials(tmp_path):
+    """Activated credentials build real gRPC mTLS channel credentials."""
+    import subprocess
+
+    from gpud_amd.pkg.kapmtls import Manager
+
+    d = tmp_path / "certs"
+    d.mkdir()
+    subprocess.run(
+        [
+            "openssl", "req", "-x509", "-newkey", "rsa:2048",
+            "-keyout", str(d / "k.pem"), "-out", str(d / "c.pem"),
+            "-days", "1", "-nodes", "-subj", "/CN=agent",
+        ],
+        check=True,
+        capture_output=True,
+    )
+    m = Manager(str(tmp_path / "kap"))
+    m.stage((d / "c.pem").read_bytes(), (d / "k.pem").read_bytes(), "1")
+    assert m.grpc_channel_credentials() is None  # not active yet
+    assert m.activate() is None
+    creds = m.grpc_channel_credentials()
+    import grpc
+
+    assert isinstance(creds, grpc.ChannelCredentials)
