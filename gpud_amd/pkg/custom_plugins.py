@@ -164,14 +164,21 @@ class Spec:
 
 
 def parse_duration(v: Any) -> float:
-    """'90s' / '10m' / '1h' / numeric seconds → seconds."""
+    """'90s' / '10m' / '1h' / numeric seconds → seconds (defaults to 60 on
+    unparsable input; scientific-notation numerics from YAML round-trips
+    are accepted)."""
     if isinstance(v, (int, float)):
         return float(v)
     s = str(v).strip()
-    m = re.fullmatch(r"([\d.]+)\s*(ms|s|m|h)?", s)
+    m = re.fullmatch(r"([0-9.eE+-]+)\s*(ms|s|m|h)?", s)
     if not m:
         return 60.0
-    val = float(m.group(1))
+    try:
+        val = float(m.group(1))
+    except ValueError:
+        return 60.0
+    if val < 0:
+        return 60.0
     return val * {"ms": 0.001, "s": 1, "m": 60, "h": 3600, None: 1}[m.group(2)]
 
 

@@ -1,0 +1,130 @@
+"""Property-based tests (hypothesis) for the parsing/crypto kernels of the
+daemon — the components most exposed to adversarial/arbitrary input:
+kmsg records, duration strings, JSON dot-paths, ed25519, event ordering."""
+
+import datetime
+import string
+
+from hypothesis import HealthCheck, given, settings
+from hypothesis import strategies as st
+
+from gpud_amd.apiv1.types import Event, utcnow
+from gpud_amd.pkg.kmsg.parser import parse_line
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    priority=st.integers(min_value=0, max_value=191),
+    seq=st.integers(min_value=0, max_value=2**40),
+    ts=st.integers(min_value=0, max_value=2**50),
+    msg=st.text(
+        alphabet=string.printable.replace("\n", "").replace("\r", ""),
+        max_size=200,
+    ),
+)
+def test_kmsg_parse_roundtrip(priority, seq, ts, msg):
+    line = f"{priority},{seq},{ts},-;{msg}"
+    m = parse_line(line, boot_time_epoch=0.0)
+    assert m is not None
+    assert m.priority == priority
+    assert m.sequence == seq
+    assert m.timestamp_us == ts
+    assert m.message == msg
+    assert 0 <= m.severity <= 7
+    assert m.facility == priority >> 3
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.text(max_size=80))
+def test_kmsg_parse_never_raises(garbage):
+    # arbitrary input may return None but must never raise
+    parse_line(garbage, boot_time_epoch=0.0)
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    val=st.floats(min_value=0, max_value=10_000, allow_nan=False),
+    unit=st.sampled_from(["ms", "s", "m", "h", ""]),
+)
+def test_parse_duration_units(val, unit):
+    from gpud_amd.pkg.custom_plugins import parse_duration
+
+    mult = {"ms": 0.001, "s": 1, "m": 60, "h": 3600, "": 1}[unit]
+    got = parse_duration(f"{val}{unit}")
+    assert abs(got - val * mult) < 1e-6 * max(1.0, val * mult)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(max_size=40))
+def test_parse_duration_never_raises(s):
+    from gpud_amd.pkg.custom_plugins import parse_duration
+
+    assert parse_duration(s) >= 0
+
+
+@settings(max_examples=50, deadline=None)
+@given(
+    payload=st.dictionaries(
+        st.text(alphabet=string.ascii_letters, min_size=1, max_size=8),
+        st.one_of(st.integers(), st.text(max_size=10)),
+        max_size=4,
+    ),
+    path=st.text(alphabet=string.ascii_letters + ".", max_size=20),
+)
+def test_dotpath_dig_never_raises(payload, path):
+    from gpud_amd.pkg.custom_plugins import _dig
+
+    _dig(payload, path)  # may be None, must not raise
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    message=st.binary(min_size=0, max_size=256),
+    seed=st.binary(min_size=32, max_size=32),
+)
+def test_ed25519_sign_verify_property(message, seed):
+    from gpud_amd.pkg import distsign
+
+    _, pub = distsign.generate_keypair(seed)
+    sig = distsign.sign(message, seed)
+    assert distsign.verify(message, sig, pub)
+    # a flipped message bit must not verify
+    if message:
+        tampered = bytes([message[0] ^ 1]) + message[1:]
+        assert not distsign.verify(tampered, sig, pub)
+
+
+@settings(
+    max_examples=25,
+    deadline=None,
+    suppress_health_check=[HealthCheck.function_scoped_fixture],
+)
+@given(
+    offsets=st.lists(
+        st.integers(min_value=0, max_value=86_400), min_size=1, max_size=30
+    )
+)
+def test_eventstore_returns_desc_order(mem_db, offsets):
+    from gpud_amd.pkg.eventstore import Store
+
+    rw, ro = mem_db
+    store = Store(rw, ro)
+    bucket = store.bucket("prop-order", disable_purge=True)
+    base = utcnow() - datetime.timedelta(days=2)
+    for off in offsets:
+        bucket.insert(
+            Event(
+                time=base + datetime.timedelta(seconds=off),
+                name="e",
+                type="Info",
+                message=str(off),
+            )
+        )
+    evs = bucket.get(base - datetime.timedelta(seconds=1))
+    times = [e.time for e in evs]
+    assert times == sorted(times, reverse=True)
+    assert len(evs) == len(offsets)
+    # cleanup for the next example (function-scoped via new table? same
+    # table accumulates across examples — purge everything)
+    bucket.purge(int((base + datetime.timedelta(days=5)).timestamp()))
+    store.close()
