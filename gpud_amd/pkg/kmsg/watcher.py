@@ -19,10 +19,26 @@ import threading
 from typing import Callable, List, Optional
 
 from ..log import logger
-from .parser import Message, boot_wall_time, parse_line
+from .parser import Message, boot_wall_time, parse_continuation, parse_line
 
 KMSG_PATH = "/dev/kmsg"
 _READ_SIZE = 8192
+
+
+def _parse_record(data: bytes, boot: float):
+    """One read(2) returns one record, possibly followed by indented
+    ``SUBSYSTEM=``/``DEVICE=`` continuation lines (reference:
+    pkg/kmsg/watcher.go:292 parseLine) — fold those into ``extra`` instead
+    of leaving them embedded in the message text."""
+    text = data.decode("utf-8", "replace")
+    first, _, rest = text.partition("\n")
+    m = parse_line(first, boot)
+    if m is None:
+        return None
+    for cont in rest.split("\n"):
+        if cont:
+            parse_continuation(cont, m)
+    return m
 
 
 class Watcher:
@@ -61,7 +77,7 @@ class Watcher:
                     raise
                 if not data:
                     break
-                m = parse_line(data.decode("utf-8", "replace"), boot)
+                m = _parse_record(data, boot)
                 if m is not None:
                     out.append(m)
         finally:
@@ -106,7 +122,7 @@ class Watcher:
                 return
             if not data:
                 continue
-            m = parse_line(data.decode("utf-8", "replace"), boot)
+            m = _parse_record(data, boot)
             if m is None:
                 continue
             with self._cb_lock:

@@ -378,3 +378,19 @@ def test_memory_component_kmsg_matcher():
     r = match_memory_kmsg("EDAC MC0: 2 UE on DIMM_B2")
     assert r is not None and r.name == "memory_edac_uncorrectable"
     assert match_memory_kmsg("nothing to see") is None
+
+
+def test_record_with_continuation_lines():
+    from gpud_amd.pkg.kmsg.watcher import _parse_record
+
+    raw = (
+        b"6,5,1000,-;amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!\n"
+        b" SUBSYSTEM=pci\n DEVICE=+pci:0000:0a:00.0\n"
+    )
+    m = _parse_record(raw, 0.0)
+    assert m is not None
+    assert m.message == "amdgpu 0000:0a:00.0: amdgpu: GPU reset begin!"
+    assert m.extra == {"SUBSYSTEM": "pci", "DEVICE": "+pci:0000:0a:00.0"}
+    # the clean message still matches the catalog
+    res = ras_catalog.match(m.message)
+    assert res[0].name == "amdgpu_gpu_reset_begin"
