@@ -149,3 +149,69 @@ def test_bench_fault_replay_mock():
     assert fr["injected_cycles"] > 0
     # every injected-fault cycle must be detected by some component
     assert fr["detected_cycles"] == fr["injected_cycles"]
+
+
+def test_daemon_with_plugin_specs_file(tmp_path):
+    """CLI run with --plugin-specs-file registers plugin components and
+    serves them over /v1; custom plugins are HTTP-deregisterable."""
+    specs = tmp_path / "plugins.yaml"
+    specs.write_text(
+        """
+- plugin_name: e2e-probe
+  plugin_type: component
+  interval: 1h
+  health_state_plugin:
+    steps:
+      - run_bash_script:
+          script: echo '{"ok":"yes"}'
+    parser:
+      json_paths:
+        - query: ok
+          field: ok
+          expect:
+            regex: "^yes$"
+"""
+    )
+    port = _free_port()
+    env = {
+        **os.environ,
+        "GPUD_AMDSMI_MOCK": "1",
+        "PYTHONPATH": REPO,
+    }
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run", "--in-memory-db",
+            "--address", f"127.0.0.1:{port}",
+            "--plugin-specs-file", str(specs),
+            "--log-level", "warning",
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(30)
+        assert "custom-plugin-e2e-probe" in client.get_components()
+        out = client.trigger_check(component="custom-plugin-e2e-probe")
+        st = out["states"][0]
+        assert st["health"] == "Healthy"
+        assert st["extra_info"]["ok"] == "yes"
+        assert st["component_type"] == "custom-plugin"
+        # plugins listed and deregisterable over HTTP
+        assert client.get_plugins()[0]["plugin_name"] == "e2e-probe"
+        client.deregister_component("custom-plugin-e2e-probe")
+        assert "custom-plugin-e2e-probe" not in client.get_components()
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            os.killpg(proc.pid, signal.SIGKILL)
+            proc.wait(timeout=5)
