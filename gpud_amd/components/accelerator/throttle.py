@@ -119,6 +119,12 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
                     uuid=uuid,
                 )
             active = self._active_throttlers(uuid, v)
+            # instantaneous SMU throttler bitmask from gpu_metrics — any
+            # set bit means the clock governor is actively limiting now
+            gm = snap.get("gpu_metrics") or {}
+            indep = int(gm.get("indep_throttle_status", 0) or 0)
+            if indep and not active:
+                active = [("indep", f"throttler bitmask 0x{indep:x}")]
             if active:
                 descs = [d for _k, d in active]
                 active_by_uuid[uuid] = descs

@@ -394,3 +394,22 @@ def test_record_with_continuation_lines():
     # the clean message still matches the catalog
     res = ras_catalog.match(m.message)
     assert res[0].name == "amdgpu_gpu_reset_begin"
+
+
+def test_ras_doc_in_sync():
+    """docs/RAS_CATALOG.md must be regenerated when the catalog changes."""
+    import importlib.util
+
+    spec = importlib.util.spec_from_file_location(
+        "gen_ras_doc",
+        os.path.join(os.path.dirname(__file__), "..", "scripts", "gen_ras_doc.py"),
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    doc_path = os.path.join(
+        os.path.dirname(__file__), "..", "docs", "RAS_CATALOG.md"
+    )
+    with open(doc_path) as f:
+        assert f.read() == mod.render(), (
+            "docs/RAS_CATALOG.md is stale — run scripts/gen_ras_doc.py"
+        )
