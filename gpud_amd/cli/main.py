@@ -45,6 +45,12 @@ def run(
     data_dir: str = typer.Option(DEFAULT_DATA_DIR, help="state directory"),
     address: str = typer.Option("localhost:15132", help="listen address"),
     expected_gpu_count: int = typer.Option(0),
+    expected_xgmi_link_count: int = typer.Option(
+        0, help="xGMI links per GPU that must be UP (7 on an 8-OAM node)"
+    ),
+    expected_ib_ports: int = typer.Option(0),
+    expected_ib_rate_gbps: float = typer.Option(0.0),
+    poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
     token: str = typer.Option("", help="control-plane token"),
@@ -62,6 +68,10 @@ def run(
         data_dir, address, expected_gpu_count, plugin_specs_file, endpoint
     )
     cfg.token = token
+    cfg.expected_xgmi_link_count = expected_xgmi_link_count
+    cfg.expected_ib_ports = expected_ib_ports
+    cfg.expected_ib_rate_gbps = expected_ib_rate_gbps
+    cfg.poll_interval_seconds = poll_interval_seconds
     if not in_memory_db:
         os.makedirs(cfg.data_dir, exist_ok=True)
 
