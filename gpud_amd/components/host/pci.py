@@ -16,21 +16,14 @@ from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 NAME = "pci"
 
 
-def bridges_with_acs_enabled(lspci_command: str = "") -> Optional[List[str]]:
-    cmd = lspci_command or "lspci"
-    try:
-        out = subprocess.run(
-            [cmd, "-vvv"], capture_output=True, text=True, timeout=30
-        )
-        if out.returncode != 0:
-            return None
-    except (OSError, subprocess.TimeoutExpired):
-        return None
+def parse_acs_bridges(lspci_vvv_output: str) -> List[str]:
+    """Bridges whose ACS control has SrcValid enabled, from `lspci -vvv`
+    text (split out for fixture-driven tests)."""
     enabled = []
     current_dev = ""
     is_bridge = False
     in_acs_cap = False
-    for line in out.stdout.splitlines():
+    for line in lspci_vvv_output.splitlines():
         if line and not line[0].isspace():
             current_dev = line.split(" ", 1)[0]
             is_bridge = "PCI bridge" in line
@@ -48,6 +41,19 @@ def bridges_with_acs_enabled(lspci_command: str = "") -> Optional[List[str]]:
                 enabled.append(current_dev)
             in_acs_cap = False
     return enabled
+
+
+def bridges_with_acs_enabled(lspci_command: str = "") -> Optional[List[str]]:
+    cmd = lspci_command or "lspci"
+    try:
+        out = subprocess.run(
+            [cmd, "-vvv"], capture_output=True, text=True, timeout=30
+        )
+        if out.returncode != 0:
+            return None
+    except (OSError, subprocess.TimeoutExpired):
+        return None
+    return parse_acs_bridges(out.stdout)
 
 
 class PCIComponent(TickerComponent):
