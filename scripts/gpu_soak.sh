@@ -5,7 +5,7 @@ set -x
 cd /root/repo
 mkdir -p gpurun_out
 export HSA_ENABLE_IPC_MODE_LEGACY=0 PYTHONPATH=/root/repo
-timeout 170 python - > gpurun_out/soak.log 2>&1 <<'PYEOF'
+timeout 380 python - > gpurun_out/soak.log 2>&1 <<'PYEOF'
 import json, os, signal, socket, subprocess, sys, time
 import psutil
 
@@ -13,7 +13,8 @@ s = socket.socket(); s.bind(("127.0.0.1", 0)); port = s.getsockname()[1]; s.clos
 env = {**os.environ}
 proc = subprocess.Popen(
     [sys.executable, "-m", "gpud_amd", "run", "--in-memory-db",
-     "--address", f"127.0.0.1:{port}", "--log-level", "warning"],
+     "--address", f"127.0.0.1:{port}", "--log-level", "warning",
+     "--poll-interval-seconds", "1"],
     start_new_session=True, env=env)
 try:
     sys.path.insert(0, "/root/repo")
@@ -22,7 +23,7 @@ try:
     assert c.wait_healthz(60)
     p = psutil.Process(proc.pid)
     rss0 = p.memory_info().rss
-    t_end = time.time() + 90
+    t_end = time.time() + 300
     i = 0
     
     while time.time() < t_end:
