@@ -25,19 +25,32 @@ try:
     rss0 = p.memory_info().rss
     t_end = time.time() + 300
     i = 0
-    
+    diag_done = False
+    import httpx as _hx
+    long_client = _hx.Client(verify=False, timeout=240.0)
     while time.time() < t_end:
-        # hammer the API + trigger checks + inject a fault every 10th pass
+        # hammer the API + trigger fast checks; one full diag battery
+        # (MFMA/GEMM/HBM/LDS, ~30-60 s synchronous) fires mid-soak
         states = c.get_health_states()
-        c.trigger_check(tag="amd")
+        c.trigger_check(component="accelerator-amd-temperature")
+        c.trigger_check(component="accelerator-amd-ecc")
         if i % 10 == 0:
             try:
                 c.inject_fault(ras_event_name="amdgpu_ring_timeout")
             except Exception:
                 pass  # /dev/kmsg absent on some boxes -> expected 400
+        if not diag_done and time.time() > t_end - 240:
+            r = long_client.get(
+                f"https://127.0.0.1:{port}/v1/components/trigger-check",
+                params={"componentName": "accelerator-amd-diag-mfma"},
+            )
+            assert r.status_code == 200, r.text[:200]
+            assert r.json()["states"][0]["health"] == "Healthy", r.text[:300]
+            diag_done = True
         c.prometheus_metrics()
         i += 1
         time.sleep(1.0)
+    assert diag_done
     rss1 = p.memory_info().rss
     cpu = p.cpu_percent(interval=2.0)
     states = c.get_health_states()
