@@ -14,7 +14,7 @@ export PYTHONPATH=/root/repo
 timeout 420 python -m pytest tests -m gpu -q > gpurun_out/pytest_gpu3.log 2>&1
 echo "pytest rc=$?" >> gpurun_out/pytest_gpu3.log
 
-timeout 200 python bench.py --steps 500 --warmup 50 > gpurun_out/bench_poll.json 2>/dev/null
+timeout 300 python bench.py --steps 2000 --warmup 100 > gpurun_out/bench_poll.json 2>/dev/null
 timeout 200 python bench.py --steps 200 --warmup 20 --fault-replay > gpurun_out/bench_faults.json 2>/dev/null
 
 timeout 300 python - > gpurun_out/diag_summary.json 2> gpurun_out/diag_summary.err <<'EOF'
@@ -25,8 +25,10 @@ out = {}
 out["mfma_bf16"] = _diag.mfma_stress_bf16(iters=2048, workgroups=1024)
 out["mfma_fp8"] = _diag.mfma_stress_fp8(iters=2048, workgroups=1024)
 out["mfma_mxfp8"] = _diag.mfma_stress_mxfp8(iters=2048, workgroups=1024)
-out["gemm_bf16_4096"] = _diag.gemm_stress_bf16(size=4096, iters=5)
-out["gemm_bf16_8192"] = _diag.gemm_stress_bf16(size=8192, iters=5)
+out["mfma_mxfp4"] = _diag.mfma_stress_mxfp4(iters=2048, workgroups=1024)
+out["gemm_bf16_v2_4096"] = _diag.gemm_stress_bf16_v2(size=4096, iters=5)
+out["gemm_bf16_v2_8192"] = _diag.gemm_stress_bf16_v2(size=8192, iters=5)
+out["gemm_mxfp8_8192"] = _diag.gemm_stress_mxfp8(size=8192, iters=5)
 out["hbm"] = _diag.hbm_bandwidth(buffer_gb=4.0, iters=8)
 out["lds"] = _diag.lds_bandwidth(iters=20000, workgroups=512)
 env = dict(os.environ); env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
