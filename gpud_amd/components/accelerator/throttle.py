@@ -119,12 +119,13 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
                     uuid=uuid,
                 )
             active = self._active_throttlers(uuid, v)
-            # instantaneous SMU throttler bitmask from gpu_metrics — any
-            # set bit means the clock governor is actively limiting now
-            gm = snap.get("gpu_metrics") or {}
-            indep = int(gm.get("indep_throttle_status", 0) or 0)
-            if indep and not active:
-                active = [("indep", f"throttler bitmask 0x{indep:x}")]
+            # NOTE: gpu_metrics' indep_throttle_status bitmask is NOT used
+            # as an activity signal: it carries benign always-set bits
+            # (low-utilization / gfx-clk-below-host-limit) on idle GPUs —
+            # measured: treating it as active inserted an event every poll,
+            # grew the event table unboundedly (bench p50 2.0 -> 4.0 ms
+            # over 2000 cycles) and would flag idle boards as throttling.
+            # The named residency accumulators above are the real signal.
             if active:
                 descs = [d for _k, d in active]
                 active_by_uuid[uuid] = descs
