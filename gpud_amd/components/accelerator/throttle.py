@@ -119,6 +119,15 @@ class ThrottleComponent(TickerComponent, SmiComponentMixin):
                     uuid=uuid,
                 )
             active = self._active_throttlers(uuid, v)
+            # raw SMU bitmask exported for operators (informational only —
+            # see the NOTE below for why it is not a health signal)
+            gm = snap.get("gpu_metrics") or {}
+            self._gauges.set(
+                "accelerator_amd_throttle_indep_status_bits",
+                "Raw gpu_metrics indep_throttle_status bitmask",
+                float(int(gm.get("indep_throttle_status", 0) or 0)),
+                uuid=uuid,
+            )
             # NOTE: gpu_metrics' indep_throttle_status bitmask is NOT used
             # as an activity signal: it carries benign always-set bits
             # (low-utilization / gfx-clk-below-host-limit) on idle GPUs —
