@@ -123,3 +123,33 @@ def test_scan_json_output():
     comps = {c["component"]: c for c in out["components"]}
     assert comps["accelerator-amd-temperature"]["health"] == "Healthy"
     assert "states" in comps["cpu"]
+
+
+def test_example_plugins_load_and_run():
+    """The shipped examples/plugins.yaml must stay valid and runnable."""
+    import os
+
+    path = os.path.join(
+        os.path.dirname(__file__), "..", "examples", "plugins.yaml"
+    )
+    from gpud_amd.pkg.custom_plugins import load_specs, make_components, run_init_plugins
+
+    specs = load_specs(path)
+    assert [s.plugin_name for s in specs] == [
+        "rocm-present",
+        "sysfs-gpu-count",
+        "mount-writable",
+    ]
+    assert run_init_plugins(specs) is None  # /opt/rocm exists here
+    comps = [
+        c
+        for s in specs
+        if s.plugin_type != "init"
+        for c in make_components(s)
+    ]
+    names = [c.name for c in comps]
+    assert "custom-plugin-mount-writable-tmp" in names
+    # the tmp mount-writable probe should pass anywhere
+    tmp_comp = next(c for c in comps if c.name.endswith("-tmp"))
+    cr = tmp_comp.trigger_check()
+    assert cr.health == "Healthy", cr.raw_output
