@@ -153,3 +153,8 @@ def test_example_plugins_load_and_run():
     tmp_comp = next(c for c in comps if c.name.endswith("-tmp"))
     cr = tmp_comp.trigger_check()
     assert cr.health == "Healthy", cr.raw_output
+    # the sysfs crosscheck parses its own JSON (health depends on whether
+    # this host has AMD cards in sysfs)
+    sysfs_comp = next(c for c in comps if c.name.endswith("gpu-count"))
+    cr = sysfs_comp.trigger_check()
+    assert "amd_cards" in (cr.extra_info or {}), cr.raw_output
