@@ -489,6 +489,7 @@ __device__ __forceinline__ int swz(int byte_off) {
          (((byte_off >> 8) & 1) << 6);
 }
 
+template <bool SETPRIO>
 __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
     const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -603,7 +604,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
       } else {
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       }
-      __builtin_amdgcn_s_setprio(1);
+      if (SETPRIO) __builtin_amdgcn_s_setprio(1);
       // kh outer: 8 independent MFMAs between accumulator reuses (the
       // dependent-accumulator latency exceeds the issue interval)
 #pragma unroll
@@ -618,7 +619,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
           }
         }
       }
-      __builtin_amdgcn_s_setprio(0);
+      if (SETPRIO) __builtin_amdgcn_s_setprio(0);
       if (p == 3) {
         // K-tile boundary: only tile kt+1's 8 glds are outstanding, and
         // they were issued >=2 phases ago — this drain is cheap and no
@@ -1077,7 +1078,7 @@ py::dict gemm_stress_mxfp8(int size, int iters) {
   return d;
 }
 
-py::dict gemm_stress_bf16_v2(int size, int iters) {
+py::dict gemm_stress_bf16_v2_impl(int size, int iters, bool setprio) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
   if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
@@ -1095,13 +1096,15 @@ py::dict gemm_stress_bf16_v2(int size, int iters) {
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
-  hipLaunchKernelGGL(gemm_bf16_8phase_kernel, dim3(blocks), dim3(512), 0, 0,
+  auto* kern = setprio ? gemm_bf16_8phase_kernel<true>
+                       : gemm_bf16_8phase_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0,
                      d_a, d_bt, d_c, M, N, K);  // warmup
   HIP_CHECK(hipGetLastError());
   HIP_CHECK(hipDeviceSynchronize());
   HIP_CHECK(hipEventRecord(t0));
   for (int i = 0; i < iters; ++i) {
-    hipLaunchKernelGGL(gemm_bf16_8phase_kernel, dim3(blocks), dim3(512), 0, 0,
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0,
                        d_a, d_bt, d_c, M, N, K);
   }
   HIP_CHECK(hipEventRecord(t1));
@@ -1139,6 +1142,17 @@ py::dict gemm_stress_bf16_v2(int size, int iters) {
   d["verify_failures"] = (long)bad;
   d["verified"] = (bad == 0);
   return d;
+}
+
+py::dict gemm_stress_bf16_v2(int size, int iters) {
+  return gemm_stress_bf16_v2_impl(size, iters, true);
+}
+
+// A/B seam: identical kernel with the s_setprio(1) MFMA-burst hint compiled
+// out, for same-box interleaved comparisons (run-to-run and box-to-box DVFS
+// variance exceeds the effect size, so cross-call comparisons are invalid).
+py::dict gemm_stress_bf16_v2_nosp(int size, int iters) {
+  return gemm_stress_bf16_v2_impl(size, iters, false);
 }
 
 py::dict gemm_stress_bf16(int size, int iters) {
@@ -1397,6 +1411,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2", &gemm_stress_bf16_v2, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, 256^2 8-phase structure (swizzled LDS, setprio)");
+  m.def("gemm_stress_bf16_v2_nosp", &gemm_stress_bf16_v2_nosp,
+        py::arg("size") = 8192, py::arg("iters") = 5,
+        "v2 GEMM with s_setprio disabled (A/B seam)");
   m.def("gemm_stress_bf16_v3", &gemm_stress_bf16_v3, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, quadrant-phase deep pipeline (no boundary drain)");
