@@ -541,6 +541,136 @@ py::dict energy_count(int index) {
 }
 
 // ---------------------------------------------------------------------------
+// partitioning + CPER RAS records
+// ---------------------------------------------------------------------------
+
+// Compute/memory partition mode (SPX/DPX/.../NPS1...) plus the accelerator
+// partition profile. Read-only surface for the partition component — gpud is
+// a monitor, never a partition setter. Reference analog: none (NVML MIG is
+// not monitored by gpud); this is MI355X-specific coverage (SURVEY.md
+// ROADMAP: partition-aware enumeration).
+py::dict partition_info(int index) {
+  auto h = handle_at(index);
+  py::dict d;
+  char buf[64] = {0};
+  {
+    py::gil_scoped_release nogil;
+    if (amdsmi_get_gpu_compute_partition(h, buf, sizeof(buf)) ==
+        AMDSMI_STATUS_SUCCESS) {
+      // re-acquire handled after block
+    } else {
+      buf[0] = '\0';
+    }
+  }
+  if (buf[0]) d["compute_partition"] = std::string(buf);
+  char mbuf[64] = {0};
+  {
+    py::gil_scoped_release nogil;
+    if (amdsmi_get_gpu_memory_partition(h, mbuf, sizeof(mbuf)) !=
+        AMDSMI_STATUS_SUCCESS)
+      mbuf[0] = '\0';
+  }
+  if (mbuf[0]) d["memory_partition"] = std::string(mbuf);
+  amdsmi_accelerator_partition_profile_t prof;
+  std::memset(&prof, 0, sizeof(prof));
+  uint32_t part_ids[AMDSMI_MAX_ACCELERATOR_PARTITIONS] = {0};
+  amdsmi_status_t prc;
+  {
+    py::gil_scoped_release nogil;
+    prc = amdsmi_get_gpu_accelerator_partition_profile(h, &prof, part_ids);
+  }
+  if (prc == AMDSMI_STATUS_SUCCESS) {
+    static const char* kTypes[] = {"INVALID", "SPX", "DPX", "TPX", "QPX",
+                                   "CPX"};
+    int t = static_cast<int>(prof.profile_type);
+    d["accelerator_profile_type"] =
+        (t >= 0 && t <= 5) ? std::string(kTypes[t]) : std::to_string(t);
+    d["num_partitions"] = prof.num_partitions;
+    d["partition_id"] = part_ids[0];
+  }
+  return d;
+}
+
+// CPER (Common Platform Error Record) entries cached by the amdgpu driver —
+// structured RAS with severity + notify-type GUID, richer than dmesg text.
+// Cursor-based drain: the caller passes the cursor from the previous call
+// (0 at daemon start) and we loop while the library reports MORE_DATA.
+// Returns {"entries": [...], "cursor": next_cursor}.
+py::dict cper_entries(int index, uint32_t severity_mask, uint64_t cursor,
+                      int max_rounds) {
+  auto h = handle_at(index);
+  py::list entries;
+  std::vector<char> data(1 << 20);  // 1 MiB CPER payload buffer per round
+  std::vector<amdsmi_cper_hdr_t*> hdrs(256);
+  bool supported = true;
+  for (int round = 0; round < max_rounds; ++round) {
+    uint64_t buf_size = data.size();
+    uint64_t entry_count = hdrs.size();
+    amdsmi_status_t rc;
+    {
+      py::gil_scoped_release nogil;
+      rc = amdsmi_get_gpu_cper_entries(h, severity_mask, data.data(),
+                                       &buf_size, hdrs.data(), &entry_count,
+                                       &cursor);
+    }
+    if (rc != AMDSMI_STATUS_SUCCESS && rc != AMDSMI_STATUS_MORE_DATA) {
+      if (round == 0) supported = false;
+      break;
+    }
+    for (uint64_t i = 0; i < entry_count; ++i) {
+      const amdsmi_cper_hdr_t* hd = hdrs[i];
+      if (hd == nullptr) continue;
+      py::dict e;
+      e["severity"] = static_cast<int>(hd->error_severity);
+      static const char* kSev[] = {"non_fatal_uncorrected", "fatal",
+                                   "non_fatal_corrected"};
+      int sv = static_cast<int>(hd->error_severity);
+      e["severity_name"] =
+          (sv >= 0 && sv <= 2) ? std::string(kSev[sv]) : std::to_string(sv);
+      e["section_count"] = hd->sec_cnt;
+      e["record_length"] = hd->record_length;
+      e["record_id"] = std::string(hd->record_id, strnlen(hd->record_id, 8));
+      if (hd->cper_valid_bits.valid_bits.timestamp) {
+        char ts[40];
+        const amdsmi_cper_timestamp_t& t = hd->timestamp;
+        std::snprintf(ts, sizeof(ts), "%02u%02u-%02u-%02uT%02u:%02u:%02u",
+                      t.century, t.year, t.month, t.day, t.hours, t.minutes,
+                      t.seconds);
+        e["timestamp"] = std::string(ts);
+      }
+      // notify-type GUID: first 8 bytes little-endian match the
+      // amdsmi_cper_notify_type_t discriminants
+      uint64_t ng = 0;
+      std::memcpy(&ng, hd->notify_type.b, 8);
+      const char* nt = nullptr;
+      switch (ng) {
+        case 0x450eBDD72DCE8BB1ull: nt = "CMC"; break;
+        case 0x4a55D8434E292F96ull: nt = "CPE"; break;
+        case 0x4cc5919CE8F56FFEull: nt = "MCE"; break;
+        case 0x4dfc1A16CF93C01Full: nt = "PCIE"; break;
+        case 0x454a9308CC5263E8ull: nt = "INIT"; break;
+        case 0x42c9B7E65BAD89FFull: nt = "NMI"; break;
+        case 0x409aAB403D61A466ull: nt = "BOOT"; break;
+        case 0x4c27C6B3667DD791ull: nt = "DMAR"; break;
+        case 0x11E4BBE89A78788Aull: nt = "SEA"; break;
+        case 0x4E87B0AE5C284C81ull: nt = "SEI"; break;
+        case 0x4214520409A9D5ACull: nt = "PEI"; break;
+        case 0x49A341DF69293BC9ull: nt = "CXL"; break;
+        default: break;
+      }
+      if (nt != nullptr) e["notify_type"] = std::string(nt);
+      entries.append(e);
+    }
+    if (rc != AMDSMI_STATUS_MORE_DATA) break;
+  }
+  py::dict out;
+  out["supported"] = supported;
+  out["entries"] = entries;
+  out["cursor"] = cursor;
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // the poll hot path: everything a telemetry sweep needs, in one native call
 // ---------------------------------------------------------------------------
 
@@ -1009,6 +1139,12 @@ PYBIND11_MODULE(_amdsmi, m) {
   m.def("xgmi_info", &xgmi_info, py::arg("index"));
   m.def("link_metrics", &link_metrics, py::arg("index"));
   m.def("energy_count", &energy_count, py::arg("index"));
+  m.def("partition_info", &partition_info, py::arg("index"),
+        "Compute/memory partition mode and accelerator partition profile");
+  m.def("cper_entries", &cper_entries, py::arg("index"),
+        py::arg("severity_mask") = 0xffffffffu, py::arg("cursor") = 0,
+        py::arg("max_rounds") = 8,
+        "Drain CPER RAS records cached by the driver (cursor-based)");
   m.def("metrics_snapshot", &metrics_snapshot, py::arg("index"),
         "Full telemetry snapshot for one GPU in a single native call");
   m.def("snapshot_timings", &snapshot_timings, py::arg("index"),

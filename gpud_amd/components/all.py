@@ -14,11 +14,13 @@ from .base import Component, GPUdInstance
 from .accelerator import (
     bad_pages,
     clock_speed,
+    cper,
     ecc,
     error_ras,
     gpm,
     gpu_counts,
     gpu_memory,
+    partition,
     peer_mem,
     power,
     power_management,
@@ -65,6 +67,8 @@ def all_init_funcs() -> List[InitFunc]:
         power_management.new,
         processes.new,
         bad_pages.new,
+        partition.new,
+        cper.new,
         temperature.new,
         utilization.new,
         error_ras.new,

@@ -53,6 +53,10 @@ class Config:
     # accelerator knobs
     expected_gpu_count: int = 0
     expected_xgmi_link_count: int = 0  # 7 per GPU on an 8-GPU MI355X node
+    # partition-mode policy: empty = informational only; set to e.g.
+    # "SPX"/"NPS1" to make a mismatch Unhealthy (accelerator-amd-partition)
+    expected_compute_partition: str = ""
+    expected_memory_partition: str = ""
     expected_ib_ports: int = 0
     expected_ib_rate_gbps: float = 0.0
     # network-latency probe targets: list of (host, port)

@@ -110,6 +110,16 @@ class Device:
     def vbios_info(self) -> Dict[str, Any]:
         return self._b.vbios_info(self.index)
 
+    def partition_info(self) -> Dict[str, Any]:
+        return self._b.partition_info(self.index)
+
+    def cper_entries(
+        self, severity_mask: int = 0xFFFFFFFF, cursor: int = 0
+    ) -> Dict[str, Any]:
+        return self._b.cper_entries(
+            self.index, severity_mask=severity_mask, cursor=cursor
+        )
+
 
 def rocm_version() -> str:
     try:

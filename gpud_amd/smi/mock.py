@@ -67,6 +67,10 @@ class MockBackend:
             "xgmi_states": [1] * MI355X_XGMI_LINKS,
             "xgmi_error_status": 0,
             "throttle": {},
+            "compute_partition": "SPX",
+            "memory_partition": "NPS1",
+            # list of CPER header dicts; tests append to inject RAS records
+            "cper": [],
             "processes": [
                 {
                     "name": "python3",
@@ -287,6 +291,27 @@ class MockBackend:
                 }
             )
         return {"num_links": len(links), "links": links}
+
+    def partition_info(self, i: int) -> Dict[str, Any]:
+        s = self._check(i)
+        return {
+            "compute_partition": s.get("compute_partition", "SPX"),
+            "memory_partition": s.get("memory_partition", "NPS1"),
+            "accelerator_profile_type": s.get("compute_partition", "SPX"),
+            "num_partitions": 1,
+            "partition_id": 0,
+        }
+
+    def cper_entries(
+        self, i: int, severity_mask: int = 0xFFFFFFFF, cursor: int = 0, **_kw
+    ) -> Dict[str, Any]:
+        s = self._check(i)
+        entries = [dict(e) for e in s.get("cper", [])[int(cursor):]]
+        return {
+            "supported": True,
+            "entries": entries,
+            "cursor": int(cursor) + len(entries),
+        }
 
     def energy_count(self, i: int) -> Dict[str, Any]:
         self._check(i)
