@@ -226,6 +226,8 @@ def run(
 
 
 @app.command()
+@app.command("check", hidden=True)  # reference alias: scan == check == s
+@app.command("s", hidden=True)
 def scan(
     expected_gpu_count: int = typer.Option(0),
     mock: bool = typer.Option(False, help="use the mock SMI backend"),
@@ -476,6 +478,41 @@ def list_plugins(specs_file: str = typer.Argument(...)):
         )
 
 
+@app.command("custom-plugins")
+@app.command("plugins", hidden=True)  # reference aliases: cs/plugin/plugins
+def custom_plugins_cmd(
+    specs_file: str = typer.Argument(...),
+    run: bool = typer.Option(False, "--run", "-r", help="run the plugins"),
+    fail_fast: bool = typer.Option(
+        True, "--fail-fast/--no-fail-fast", "-f", help="exit on first unhealthy"
+    ),
+):
+    """Check (and optionally run) a custom-plugin specs file
+    (reference: gpud custom-plugins — cmd/gpud/command/command.go:805)."""
+    from ..pkg import custom_plugins
+
+    try:
+        specs = custom_plugins.load_specs(specs_file)
+    except Exception as e:  # noqa: BLE001 — surface any parse error as exit 1
+        typer.echo(f"invalid plugin specs: {e}", err=True)
+        raise typer.Exit(code=1)
+    typer.echo(f"valid plugin specs: {len(specs)} plugin(s)")
+    if not run:
+        for spec in specs:
+            typer.echo(f"  {spec.plugin_name} ({spec.plugin_type}, {spec.run_mode})")
+        raise typer.Exit(code=0)
+    failed = 0
+    for spec in specs:
+        for comp in custom_plugins.make_components(spec):
+            cr = comp.trigger_check()
+            typer.echo(f"{comp.name}: {cr.health} ({cr.reason})")
+            if cr.health != "Healthy":
+                failed += 1
+                if fail_fast:
+                    raise typer.Exit(code=1)
+    raise typer.Exit(code=1 if failed else 0)
+
+
 @app.command("run-plugin-group")
 def run_plugin_group(
     specs_file: str = typer.Argument(...),
@@ -533,6 +570,20 @@ def notify(
     except httpx.HTTPError as e:
         typer.echo(f"notify failed: {e}", err=True)
         raise typer.Exit(code=1)
+
+
+@app.command("update-check")
+def update_check(data_dir: str = typer.Option(DEFAULT_DATA_DIR)):
+    """Check whether the version file requests a new version
+    (reference: gpud update check — cmd/gpud/command/command.go:453)."""
+    from ..pkg.update import check_version_file
+
+    cfg = Config(data_dir=data_dir)
+    pending = check_version_file(cfg)
+    if pending:
+        typer.echo(f"update available: {pending}")
+        raise typer.Exit(code=0)
+    typer.echo("up to date")
 
 
 @app.command()

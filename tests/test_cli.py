@@ -158,3 +158,73 @@ def test_example_plugins_load_and_run():
     sysfs_comp = next(c for c in comps if c.name.endswith("gpu-count"))
     cr = sysfs_comp.trigger_check()
     assert "amd_cards" in (cr.extra_info or {}), cr.raw_output
+
+
+def test_custom_plugins_validate_and_run(tmp_path):
+    """Reference: gpud custom-plugins (validate; -r to run; fail-fast)."""
+    from typer.testing import CliRunner
+
+    from gpud_amd.cli.main import app
+
+    spec = tmp_path / "plugins.yaml"
+    spec.write_text(
+        """
+- plugin_name: ok-plugin
+  plugin_type: component
+  run_mode: auto
+  health_state_plugin:
+    steps:
+      - name: run
+        run_bash_script:
+          content_type: plaintext
+          script: "echo fine"
+- plugin_name: bad-plugin
+  plugin_type: component
+  run_mode: auto
+  health_state_plugin:
+    steps:
+      - name: run
+        run_bash_script:
+          content_type: plaintext
+          script: "exit 3"
+"""
+    )
+    runner = CliRunner()
+    r = runner.invoke(app, ["custom-plugins", str(spec)])
+    assert r.exit_code == 0
+    assert "valid plugin specs: 2" in r.output
+    # run with fail-fast (default): the bad plugin stops the run with code 1
+    r = runner.invoke(app, ["custom-plugins", str(spec), "-r"])
+    assert r.exit_code == 1
+    assert "ok-plugin: Healthy" in r.output
+    # invalid file ⇒ exit 1
+    bad = tmp_path / "broken.yaml"
+    bad.write_text("{not valid yaml: [")
+    r = runner.invoke(app, ["custom-plugins", str(bad)])
+    assert r.exit_code == 1
+
+
+def test_update_check_pending(tmp_path):
+    from typer.testing import CliRunner
+
+    from gpud_amd.cli.main import app
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.update import write_target_version
+
+    cfg = Config(data_dir=str(tmp_path))
+    runner = CliRunner()
+    r = runner.invoke(app, ["update-check", "--data-dir", str(tmp_path)])
+    assert r.exit_code == 0 and "up to date" in r.output
+    write_target_version(cfg, "v99.0.0")
+    r = runner.invoke(app, ["update-check", "--data-dir", str(tmp_path)])
+    assert "update available: v99.0.0" in r.output
+
+
+def test_scan_alias_check(monkeypatch):
+    from typer.testing import CliRunner
+
+    from gpud_amd.cli.main import app
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    r = CliRunner().invoke(app, ["check", "--mock", "--output", "json"])
+    assert r.exit_code == 0
