@@ -57,12 +57,50 @@ def run(
     in_memory_db: bool = typer.Option(False, help="ephemeral state (testing)"),
     no_tls: bool = typer.Option(False, help="serve plain HTTP (testing)"),
     log_level: str = typer.Option("info"),
+    # SMI-level failure injection for exercising the daemon on healthy
+    # hardware (reference: gpud run --gpu-uuids-with-* flags,
+    # cmd/gpud/run/command.go:272-335); comma-separated GPU uuids
+    gpu_uuids_with_gpu_lost: str = typer.Option("", help="inject: GPU lost"),
+    gpu_uuids_with_requires_reset: str = typer.Option(""),
+    gpu_uuids_with_bad_page_pending: str = typer.Option(
+        "", help="inject: bad pages pending (row-remapping-pending analog)"
+    ),
+    gpu_uuids_with_bad_page_threshold: str = typer.Option(
+        "", help="inject: bad-page count at threshold (row-remapping-failed analog)"
+    ),
+    gpu_uuids_with_throttle: str = typer.Option(
+        "", help="inject: power throttling (hw-slowdown analog)"
+    ),
+    gpu_uuids_with_thermal_throttle: str = typer.Option(
+        "", help="inject: thermal throttling (hw-slowdown-thermal analog)"
+    ),
+    gpu_uuids_with_xgmi_unhealthy: str = typer.Option(
+        "", help="inject: xGMI link down (fabric-state analog)"
+    ),
+    gpu_uuids_with_ecc_uncorrectable: str = typer.Option(
+        "", help="inject: uncorrectable ECC"
+    ),
 ):
     """Run the daemon (reference: cmd/gpud/run/command.go:42)."""
     log_setup(level=log_level)
     from ..bootstrap import build_core
     from ..pkg import custom_plugins
+    from ..pkg.fault_injector import SMIFailureInjector
     from ..server import Server
+
+    def _uuid_set(v: str):
+        return {u.strip() for u in v.split(",") if u.strip()}
+
+    injector = SMIFailureInjector(
+        gpu_lost_uuids=_uuid_set(gpu_uuids_with_gpu_lost),
+        requires_reset_uuids=_uuid_set(gpu_uuids_with_requires_reset),
+        bad_page_pending_uuids=_uuid_set(gpu_uuids_with_bad_page_pending),
+        bad_page_threshold_uuids=_uuid_set(gpu_uuids_with_bad_page_threshold),
+        throttle_uuids=_uuid_set(gpu_uuids_with_throttle),
+        thermal_throttle_uuids=_uuid_set(gpu_uuids_with_thermal_throttle),
+        xgmi_unhealthy_uuids=_uuid_set(gpu_uuids_with_xgmi_unhealthy),
+        ecc_uncorrectable_uuids=_uuid_set(gpu_uuids_with_ecc_uncorrectable),
+    )
 
     cfg = _load_config(
         data_dir, address, expected_gpu_count, plugin_specs_file, endpoint
@@ -83,7 +121,9 @@ def run(
             typer.echo(f"init plugin failure: {err}", err=True)
             raise typer.Exit(code=1)
 
-    core = build_core(cfg, in_memory_db=in_memory_db)
+    core = build_core(
+        cfg, in_memory_db=in_memory_db, smi_failure_injector=injector
+    )
 
     # token FIFO (reference: pkg/server/server.go:640-710 — a named pipe at
     # <dataDir>/gpud.fifo other local processes write a fresh CP token into)

@@ -215,3 +215,61 @@ def test_daemon_with_plugin_specs_file(tmp_path):
         except subprocess.TimeoutExpired:
             os.killpg(proc.pid, signal.SIGKILL)
             proc.wait(timeout=5)
+
+
+def test_daemon_boot_with_smi_injection_flags():
+    """`gpud run --gpu-uuids-with-*` exercises unhealthy paths on healthy
+    hardware (reference: cmd/gpud/run/command.go:272-335)."""
+    port = _free_port()
+    env = {
+        **os.environ,
+        "GPUD_AMDSMI_MOCK": "1",
+        "GPUD_AMDSMI_MOCK_GPUS": "2",
+        "PYTHONPATH": REPO,
+    }
+    # mock uuids are deterministic
+    from gpud_amd.smi.mock import MockBackend
+
+    uuid0 = MockBackend(2).device_uuid(0)
+    proc = subprocess.Popen(
+        [
+            sys.executable,
+            "-m",
+            "gpud_amd",
+            "run",
+            "--in-memory-db",
+            "--address",
+            f"127.0.0.1:{port}",
+            "--log-level",
+            "warning",
+            "--gpu-uuids-with-ecc-uncorrectable",
+            uuid0,
+            "--gpu-uuids-with-bad-page-pending",
+            uuid0,
+        ],
+        cwd=REPO,
+        env=env,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(30), "daemon did not become healthy"
+        states = client.get_health_states(
+            components=["accelerator-amd-ecc", "accelerator-amd-bad-pages"]
+        )
+        assert states["accelerator-amd-ecc"][0].health == "Unhealthy"
+        assert states["accelerator-amd-bad-pages"][0].health in (
+            "Unhealthy",
+            "Degraded",
+        )
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        proc.wait(timeout=15)
