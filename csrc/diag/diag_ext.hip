@@ -489,7 +489,7 @@ __device__ __forceinline__ int swz(int byte_off) {
          (((byte_off >> 8) & 1) << 6);
 }
 
-template <bool SETPRIO, int BARRIER_MASK = 0xF>
+template <bool SETPRIO, int BARRIER_MASK = 0xF, int PANEL = 0>
 __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
     const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -502,8 +502,36 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_8phase_kernel(
   const int wave_m = wave >> 2;       // 0..1 -> output rows [wave_m*128, +128)
   const int wave_n = wave & 3;        // 0..3 -> output cols [wave_n*64, +64)
   const int tiles_n = N / TN;
-  const int brow = (blockIdx.x / tiles_n) * TM;
-  const int bcol = (blockIdx.x % tiles_n) * TN;
+  int brow, bcol;
+  if (PANEL == -1 && (M / TM) % 16 == 0 && tiles_n % 16 == 0) {
+    // two-level XCD-aware supertiling: consecutive ids round-robin the 8
+    // XCDs (private L2 each), so give each XCD a compact 8-row x 4-col
+    // sub-block (12 tile-bands in its L2) while the 8 sub-blocks tile a
+    // 16x16 global supertile (32 bands at the shared level vs 40 for a
+    // row-major walk).
+    const int sub = blockIdx.x & 7;           // = XCD under round-robin
+    const int k = (blockIdx.x & 255) >> 3;    // 0..31 within sub-block
+    const int st = blockIdx.x >> 8;           // supertile index
+    const int st_cols = tiles_n / 16;
+    const int ST_r = (st / st_cols) * 16, ST_c = (st % st_cols) * 16;
+    brow = (ST_r + (sub & 1) * 8 + (k & 7)) * TM;
+    bcol = (ST_c + (sub >> 1) * 4 + (k >> 3)) * TN;
+  } else if (PANEL > 0 && tiles_n % PANEL == 0) {
+    // L2-locality supertiling: consecutive blockIdx values walk a
+    // PANEL-column x tiles_m panel column-major, so the ~256 concurrently
+    // resident workgroups cover a near-square tile set (16 A-bands +
+    // 16 B-bands of L2/HBM traffic instead of 8+32 for row-major walk
+    // at 8192). The XCD round-robin dispatch takes consecutive ids.
+    const int tiles_m = M / TM;
+    const int per_panel = PANEL * tiles_m;
+    const int panel = blockIdx.x / per_panel;
+    const int rem = blockIdx.x % per_panel;
+    brow = (rem / PANEL) * TM;
+    bcol = (panel * PANEL + rem % PANEL) * TN;
+  } else {
+    brow = (blockIdx.x / tiles_n) * TM;
+    bcol = (blockIdx.x % tiles_n) * TN;
+  }
   const int ntiles = K / TK;
 
   // staging: 64 slots of 1 KiB cover [A image | B image]; wave w owns
@@ -1088,7 +1116,7 @@ py::dict gemm_stress_mxfp8_impl(int size, int iters, int barrier_mask) {
 }
 
 py::dict gemm_stress_bf16_v2_impl(int size, int iters, bool setprio,
-                                  int barrier_mask = 0x8) {
+                                  int barrier_mask = 0x8, int panel = 0) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
   if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
@@ -1115,6 +1143,10 @@ py::dict gemm_stress_bf16_v2_impl(int size, int iters, bool setprio,
     kern = gemm_bf16_8phase_kernel<true, 0xA>;
   else if (barrier_mask == 0xF)
     kern = gemm_bf16_8phase_kernel<true, 0xF>;
+  if (panel == -1) kern = gemm_bf16_8phase_kernel<true, 0x8, -1>;
+  else if (panel == 8) kern = gemm_bf16_8phase_kernel<true, 0x8, 8>;
+  else if (panel == 16) kern = gemm_bf16_8phase_kernel<true, 0x8, 16>;
+  else if (panel == 32) kern = gemm_bf16_8phase_kernel<true, 0x8, 32>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0,
                      d_a, d_bt, d_c, M, N, K);  // warmup
   HIP_CHECK(hipGetLastError());
@@ -1174,6 +1206,10 @@ py::dict gemm_stress_bf16_v2_nosp(int size, int iters) {
 
 py::dict gemm_stress_bf16_v2_bmask(int size, int iters, int barrier_mask) {
   return gemm_stress_bf16_v2_impl(size, iters, true, barrier_mask);
+}
+
+py::dict gemm_stress_bf16_v2_panel(int size, int iters, int panel) {
+  return gemm_stress_bf16_v2_impl(size, iters, true, 0x8, panel);
 }
 
 py::dict gemm_stress_mxfp8(int size, int iters) {
@@ -1447,6 +1483,9 @@ PYBIND11_MODULE(_diag, m) {
         py::arg("size") = 8192, py::arg("iters") = 5,
         py::arg("barrier_mask") = 0xF,
         "A/B variant of v2 with a phase-barrier mask (0xF/0xA/0x8)");
+  m.def("gemm_stress_bf16_v2_panel", &gemm_stress_bf16_v2_panel,
+        py::arg("size") = 8192, py::arg("iters") = 5, py::arg("panel") = 16,
+        "A/B variant of v2 with L2 panel supertiling (8/16/32, 0=off)");
   m.def("gemm_stress_bf16_v3", &gemm_stress_bf16_v3, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, quadrant-phase deep pipeline (no boundary drain)");
