@@ -102,10 +102,21 @@ def get_machine_info(smi_instance: Any = None) -> MachineInfo:
                         board_id=dev.oam_id,
                     )
                 )
+            partition = ""
+            try:
+                devs = smi_instance.devices()
+                if devs:
+                    pi = next(iter(devs.values())).partition_info()
+                    comp = pi.get("compute_partition", "")
+                    mem = pi.get("memory_partition", "")
+                    if comp or mem:
+                        partition = f"{comp or '?'}/{mem or '?'}"
+            except Exception:
+                pass
             info.gpu_info = MachineGPUInfo(
                 product=smi_instance.product_name,
                 manufacturer="AMD",
-                architecture="gfx950",
+                architecture=("gfx950" + (f" ({partition})" if partition else "")),
                 driver_version=smi_instance.driver_version,
                 rocm_version=smi_instance.rocm_version,
                 memory=f"{total_vram_mb} MB" if total_vram_mb else "",
