@@ -50,6 +50,12 @@ def run(
     ),
     expected_ib_ports: int = typer.Option(0),
     expected_ib_rate_gbps: float = typer.Option(0.0),
+    expected_compute_partition: str = typer.Option(
+        "", help="alert if compute partition mode differs (e.g. SPX)"
+    ),
+    expected_memory_partition: str = typer.Option(
+        "", help="alert if memory partition mode differs (e.g. NPS1)"
+    ),
     poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
@@ -107,6 +113,8 @@ def run(
     )
     cfg.token = token
     cfg.expected_xgmi_link_count = expected_xgmi_link_count
+    cfg.expected_compute_partition = expected_compute_partition
+    cfg.expected_memory_partition = expected_memory_partition
     cfg.expected_ib_ports = expected_ib_ports
     cfg.expected_ib_rate_gbps = expected_ib_rate_gbps
     cfg.poll_interval_seconds = poll_interval_seconds

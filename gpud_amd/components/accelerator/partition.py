@@ -34,13 +34,9 @@ class PartitionComponent(TickerComponent, SmiComponentMixin):
         self._smi = inst.smi
         self._shared = inst.shared_snapshots
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
-        cfg = inst.config
-        self._expected_compute = (
-            getattr(cfg, "expected_compute_partition", "") or ""
-        ).upper()
-        self._expected_memory = (
-            getattr(cfg, "expected_memory_partition", "") or ""
-        ).upper()
+        # read through to the live Config each check so control-plane
+        # updateConfig takes effect without a restart
+        self._cfg = inst.config
         self.get_partition_info: Callable[[], Dict[str, Dict]] = (
             self._read_partitions
         )
@@ -71,6 +67,14 @@ class PartitionComponent(TickerComponent, SmiComponentMixin):
         if self._shared is not None:
             return self._shared.get_aux("partition_info", fetch)
         return fetch()
+
+    @property
+    def _expected_compute(self) -> str:
+        return (getattr(self._cfg, "expected_compute_partition", "") or "").upper()
+
+    @property
+    def _expected_memory(self) -> str:
+        return (getattr(self._cfg, "expected_memory_partition", "") or "").upper()
 
     def check(self) -> CheckResult:
         guard = self.smi_guard()

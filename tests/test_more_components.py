@@ -448,3 +448,37 @@ def test_cper_restart_dedup_via_event_store(mock_core):
     uuid0 = mock_core.smi_instance.device_uuids()[0]
     assert comp2._counts[uuid0][2] == 0  # corrected count not re-incremented
     assert cr.health == HealthStateType.HEALTHY
+
+
+def test_partition_policy_hot_settable_via_updateconfig(mock_core):
+    """Control-plane updateConfig changes the partition policy without a
+    component restart (live Config read-through)."""
+    from gpud_amd.session import Session
+
+    comp = mock_core.registry.get("accelerator-amd-partition")
+    assert comp.trigger_check().health == HealthStateType.HEALTHY
+    s = Session(
+        mock_core,
+        endpoint="unused",
+        open_reader=lambda: iter(()),
+        send_response=lambda f: None,
+    )
+    resp = s.process_request(
+        {
+            "req_id": "u",
+            "method": "updateConfig",
+            "data": {"expected_compute_partition": "CPX"},
+        }
+    )
+    assert "expected_compute_partition" in resp["data"]["applied"]
+    mock_core.shared_snapshots.refresh()
+    assert comp.trigger_check().health == HealthStateType.UNHEALTHY
+    s.process_request(
+        {
+            "req_id": "u2",
+            "method": "updateConfig",
+            "data": {"expected_compute_partition": ""},
+        }
+    )
+    mock_core.shared_snapshots.refresh()
+    assert comp.trigger_check().health == HealthStateType.HEALTHY
