@@ -319,3 +319,36 @@ def test_fault_injection_e2e_live_gpu():
         except subprocess.TimeoutExpired:
             os.killpg(proc.pid, signal.SIGKILL)
             proc.wait(timeout=5)
+
+
+def test_partition_and_cper_live(smi_instance):
+    """Partition mode reads on real hardware; CPER degrades gracefully when
+    the driver does not cache records (container driver stacks)."""
+    dev = next(iter(smi_instance.devices().values()))
+    pi = dev.partition_info()
+    # SPX is the only mode these single-OAM boxes run; tolerate any string
+    # but require the compute mode key when the API succeeds
+    if pi:
+        assert isinstance(pi.get("compute_partition", ""), str)
+    res = dev.cper_entries()
+    assert set(res) >= {"supported", "entries", "cursor"}
+    if res["supported"]:
+        for e in res["entries"]:
+            assert "severity" in e and "record_id" in e
+
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    core = build_core(
+        Config(data_dir="/tmp/gpud-partition-test"),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    try:
+        cr = core.registry.get("accelerator-amd-partition").trigger_check()
+        assert cr.health == "Healthy", cr.reason
+        cr = core.registry.get("accelerator-amd-cper").trigger_check()
+        assert cr.health == "Healthy", cr.reason
+    finally:
+        core.close()
