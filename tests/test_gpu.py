@@ -321,10 +321,17 @@ def test_fault_injection_e2e_live_gpu():
             proc.wait(timeout=5)
 
 
-def test_partition_and_cper_live(smi_instance):
+def test_partition_and_cper_live():
     """Partition mode reads on real hardware; CPER degrades gracefully when
     the driver does not cache records (container driver stacks)."""
-    dev = next(iter(smi_instance.devices().values()))
+    # fresh instance: earlier live tests core.close() the global amdsmi,
+    # which invalidates the module-scoped fixture's handles
+    from gpud_amd import smi
+
+    inst = smi.new()
+    if not inst.exists:
+        pytest.skip(f"no AMD GPU: {inst.init_error()}")
+    dev = next(iter(inst.devices().values()))
     pi = dev.partition_info()
     # SPX is the only mode these single-OAM boxes run; tolerate any string
     # but require the compute mode key when the API succeeds
