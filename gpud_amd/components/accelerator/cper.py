@@ -110,6 +110,12 @@ class CPERComponent(TickerComponent, SmiComponentMixin):
         guard = self.smi_guard()
         if guard is not None:
             return guard
+        # once the driver reports the interface unsupported, stop paying a
+        # per-GPU probe every cycle (support cannot appear at runtime)
+        if not self._supported:
+            return CheckResult(
+                NAME, reason="CPER interface not supported by this driver"
+            )
         now = self.get_now()
         fresh_by_uuid: Dict[str, List[Dict]] = {}
         for uuid, dev in self._smi.devices().items():
