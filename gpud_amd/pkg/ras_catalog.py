@@ -365,6 +365,93 @@ CATALOG: List[Detail] = [
         EventType.WARNING,
         _APP,
     ),
+    # ---- additional amdgpu / platform signatures -------------------------
+    _d(
+        "amdgpu_vram_lost",
+        r"amdgpu.*VRAM is lost due to GPU reset",
+        "VRAM contents were lost across a GPU reset — every context running "
+        "on the device lost its data; jobs must be restarted",
+        EventType.CRITICAL,
+        _APP,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_aca_error",
+        r"amdgpu.*(?:ACA|aca).*(?:error|bank)|Accelerator Check Architecture",
+        "ACA (Accelerator Check Architecture) error bank report — the "
+        "MI3xx-generation RAS telemetry block logged a hardware error",
+        EventType.WARNING,
+        _HW,
+    ),
+    _d(
+        "amdgpu_kiq_timeout",
+        r"amdgpu.*(?:KIQ|kiq).*(?:timeout|failed)",
+        "KIQ (kernel interface queue) register access timeout — the command "
+        "processor stopped servicing privileged requests; usually precedes "
+        "a ring timeout and GPU reset",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_psp_cmd_failed",
+        r"amdgpu.*(?:PSP|psp).*(?:command|cmd|load).*fail",
+        "PSP (platform security processor) command failure — firmware "
+        "loading or security setup failed",
+        EventType.CRITICAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_mes_error",
+        r"amdgpu.*MES.*(?:failed|hang|timeout)",
+        "MES (micro-engine scheduler) failure",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_fence_fallback",
+        r"amdgpu.*[Ff]ence fallback timer expired",
+        "Fence interrupt was lost and the fallback timer fired — benign "
+        "when rare; frequent occurrences indicate interrupt delivery "
+        "problems",
+        EventType.WARNING,
+        _IGNORE,
+    ),
+    _d(
+        "pcie_aer_corrected",
+        r"AER:.*[Cc]orrected error (?:received|message)",
+        "PCIe AER corrected error — recovered by hardware; track the rate",
+        EventType.INFO,
+        _IGNORE,
+    ),
+    _d(
+        "pcie_bandwidth_limited",
+        r"available PCIe bandwidth, limited by",
+        "Device trained at lower PCIe speed/width than the platform "
+        "supports — check slot seating and BIOS lane configuration",
+        EventType.WARNING,
+        _HW,
+    ),
+    _d(
+        "host_hung_task",
+        r"INFO: task .* blocked for more than \d+ seconds",
+        "Kernel hung-task watchdog: a task sat in uninterruptible sleep — "
+        "often storage or driver stalls; correlates with the os component's "
+        "D-state tracker",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "host_soft_lockup",
+        r"BUG: soft lockup - CPU#\d+ stuck",
+        "CPU soft lockup — a kernel thread monopolized a CPU; node health "
+        "is suspect",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
 ]
 
 _CATALOG_BY_NAME: Dict[str, Detail] = {d.name: d for d in CATALOG}
@@ -389,6 +476,15 @@ def match(line: str) -> Optional[Tuple[Detail, Dict[str, str]]]:
 # ---------------------------------------------------------------------------
 
 INJECTABLE: Dict[str, str] = {
+    "amdgpu_vram_lost": (
+        "amdgpu 0000:0a:00.0: amdgpu: VRAM is lost due to GPU reset!"
+    ),
+    "amdgpu_kiq_timeout": (
+        "amdgpu 0000:0a:00.0: amdgpu: KIQ reg write timeout (0x1f2c)"
+    ),
+    "host_soft_lockup": (
+        "BUG: soft lockup - CPU#12 stuck for 23s! [kworker/12:1:12345]"
+    ),
     "amdgpu_ring_timeout": (
         "[drm:amdgpu_job_timedout [amdgpu]] *ERROR* ring gfx_0.0.0 timeout, "
         "signaled seq=1234, emitted seq=1236"

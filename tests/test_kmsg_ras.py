@@ -413,3 +413,47 @@ def test_ras_doc_in_sync():
         assert f.read() == mod.render(), (
             "docs/RAS_CATALOG.md is stale — run scripts/gen_ras_doc.py"
         )
+
+
+def test_catalog_new_signatures_r1b():
+    """The round-1 catalog extension: ACA, KIQ, PSP, MES, VRAM-lost, PCIe
+    AER/bandwidth, hung-task and soft-lockup signatures."""
+    from gpud_amd.pkg.ras_catalog import match
+
+    cases = {
+        "amdgpu 0000:0a:00.0: amdgpu: VRAM is lost due to GPU reset!":
+            ("amdgpu_vram_lost", True),
+        "amdgpu 0000:0a:00.0: amdgpu: ACA error bank 3 logged":
+            ("amdgpu_aca_error", False),
+        "amdgpu 0000:0a:00.0: amdgpu: KIQ reg write timeout (0x1f2c)":
+            ("amdgpu_kiq_timeout", True),
+        "amdgpu 0000:0a:00.0: amdgpu: PSP load sos command failed":
+            ("amdgpu_psp_cmd_failed", True),
+        "amdgpu: MES failed to respond to msg=SET_HW_RES":
+            ("amdgpu_mes_error", True),
+        "pcieport 0000:00:01.1: AER: Corrected error received: 0000:0a:00.0":
+            ("pcie_aer_corrected", False),
+        "32.000 Gb/s available PCIe bandwidth, limited by 2.5 GT/s PCIe x16 link":
+            ("pcie_bandwidth_limited", False),
+        "INFO: task python:12345 blocked for more than 122 seconds.":
+            ("host_hung_task", False),
+        "BUG: soft lockup - CPU#12 stuck for 23s! [kworker/12:1:12345]":
+            ("host_soft_lockup", True),
+    }
+    for line, (name, critical) in cases.items():
+        m = match(line)
+        assert m is not None, line
+        assert m[0].name == name, (line, m[0].name)
+        assert m[0].critical == critical, name
+    # ordering: a KIQ timeout must not be swallowed by the generic ring
+    # timeout signature, and a PSP failure not by firmware-load
+    assert match("[drm:amdgpu_job_timedout [amdgpu]] *ERROR* ring gfx_0.0.0 "
+                 "timeout")[0].name in ("amdgpu_ring_timeout", "amdgpu_job_timeout")
+
+
+def test_every_injectable_matches_its_entry():
+    from gpud_amd.pkg.ras_catalog import INJECTABLE, match
+
+    for name, line in INJECTABLE.items():
+        m = match(line)
+        assert m is not None and m[0].name == name, name
