@@ -489,6 +489,37 @@ def set_healthy(
     c.close()
 
 
+@app.command()
+def bundle(
+    output: str = typer.Option("", help="output tar.gz path"),
+    mock: bool = typer.Option(False, help="use the mock SMI backend"),
+):
+    """Collect a support bundle (the nvidia-bug-report.sh analog):
+    amdsmi state, kernel messages, component health, host context."""
+    log_setup(level="warning")
+    if mock:
+        os.environ["GPUD_AMDSMI_MOCK"] = "1"
+    import time as _time
+
+    from ..bootstrap import build_core
+    from ..pkg.bundle import collect_bundle
+
+    out = output or f"gpud-bundle-{_time.strftime('%Y%m%d-%H%M%S')}.tar.gz"
+    core = build_core(
+        Config(), in_memory_db=True, kmsg_writable=False, record_reboot=False
+    )
+    try:
+        for comp in core.registry.all_components():
+            try:
+                comp.trigger_check()
+            except Exception:  # noqa: BLE001 — bundle what we can
+                pass
+        path = collect_bundle(out, core=core)
+        typer.echo(f"bundle written: {path} ({os.path.getsize(path)} bytes)")
+    finally:
+        core.close()
+
+
 @app.command("machine-info")
 def machine_info(mock: bool = typer.Option(False)):
     if mock:

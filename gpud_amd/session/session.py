@@ -16,6 +16,7 @@ from __future__ import annotations
 import base64
 import datetime
 import json
+import os
 import random
 import threading
 import time
@@ -319,6 +320,25 @@ class Session:
         diag components). ``{"async": true}`` starts them in the
         background — results land in each component's cached health state
         (queryable via the ``states`` method or /v1/states)."""
+        if payload.get("bundle"):
+            # support-bundle mode (reference diagnostic.go collects
+            # nvidia-bug-report.sh output and PUTs it to a presigned URL)
+            import tempfile
+
+            from ..pkg.bundle import collect_bundle, upload_bundle
+
+            path = payload.get("path") or os.path.join(
+                tempfile.gettempdir(), "gpud-bundle.tar.gz"
+            )
+            collect_bundle(path, core=self.core)
+            out = {"bundle": path, "size": os.path.getsize(path)}
+            url = payload.get("upload_url", "")
+            if url:
+                err = upload_bundle(path, url)
+                out["uploaded"] = err is None
+                if err:
+                    out["upload_error"] = err
+            return out
         if payload.get("async"):
             def _run():
                 for name in self.DIAG_COMPONENTS:

@@ -420,3 +420,56 @@ esac
     sts = package_statuses(cfg)
     assert sts[0].phase == "Installed"
     assert sts[0].current_version == "1.2.3"
+
+
+def test_support_bundle_collect_and_session(tmp_path, monkeypatch):
+    """Bundle collector (nvidia-bug-report analog) + diagnostic bundle mode."""
+    import tarfile
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK_GPUS", "2")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.bundle import collect_bundle
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.session import Session
+
+    core = build_core(
+        Config(data_dir=str(tmp_path)),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    try:
+        core.registry.get("cpu").trigger_check()
+        out = str(tmp_path / "bundle.tar.gz")
+        path = collect_bundle(out, core=core)
+        with tarfile.open(path) as tf:
+            names = tf.getnames()
+            assert "gpud-bundle/bundle-info.json" in names
+            assert "gpud-bundle/amdsmi.json" in names
+            assert "gpud-bundle/states.json" in names
+            import json as _json
+
+            smi_doc = _json.load(tf.extractfile("gpud-bundle/amdsmi.json"))
+            assert smi_doc["device_count"] == 2
+            assert smi_doc["devices"]
+            states = _json.load(tf.extractfile("gpud-bundle/states.json"))
+            assert "cpu" in states
+
+        s = Session(
+            core,
+            endpoint="unused",
+            open_reader=lambda: iter(()),
+            send_response=lambda f: None,
+        )
+        resp = s.process_request(
+            {
+                "req_id": "b",
+                "method": "diagnostic",
+                "data": {"bundle": True, "path": str(tmp_path / "b2.tar.gz")},
+            }
+        )
+        assert resp["data"]["size"] > 0
+        assert tarfile.is_tarfile(resp["data"]["bundle"])
+    finally:
+        core.close()
