@@ -212,3 +212,36 @@ def test_admin_endpoints(daemon):
     assert any("gpud" in k or "MainThread" in k for k in r.json())
     r = httpx.get(server.base_url + "/admin/packages", verify=False)
     assert r.status_code == 200
+
+
+def test_malformed_requests_never_500(daemon):
+    """Adversarial/garbage inputs degrade to 4xx, never 5xx (the control
+    plane retries 5xx; a junk query param must not look like an outage)."""
+    import httpx
+
+    core, server, client = daemon
+    base = server.base_url
+    h = httpx.Client(verify=False, timeout=10)
+    cases = [
+        ("GET", "/v1/states", {"components": "no-such-component"}),
+        ("GET", "/v1/states", {"components": "../../etc/passwd"}),
+        ("GET", "/v1/events", {"startTime": "not-a-time"}),
+        ("GET", "/v1/events", {"startTime": "99999999999999999999"}),
+        ("GET", "/v1/metrics", {"since": "-1h%00"}),
+        ("GET", "/v1/components/trigger-check", {"componentName": "\x00weird"}),
+        ("GET", "/v1/components/trigger-tag", {"tagName": "'; DROP TABLE x;--"}),
+        ("GET", "/v1/info", {"components": ","}),
+    ]
+    for method, path, params in cases:
+        r = h.request(method, base + path, params=params)
+        assert r.status_code < 500, (path, params, r.status_code, r.text[:200])
+    # malformed bodies on POST endpoints
+    for path, body in [
+        ("/v1/health-states/set-healthy", b"{broken json"),
+        ("/inject-fault", b"\xff\xfe binary"),
+    ]:
+        r = h.post(
+            base + path, content=body, headers={"Content-Type": "application/json"}
+        )
+        assert r.status_code < 500, (path, r.status_code, r.text[:200])
+    h.close()
