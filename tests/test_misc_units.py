@@ -344,3 +344,71 @@ def test_ed25519_rfc8032_vectors():
         assert sig == bytes.fromhex(sig_hex), "signature mismatch"
         assert distsign.verify(msg, sig, pub)
         assert not distsign.verify(msg + b"x", sig, pub)
+
+
+def test_login_against_fake_control_plane(tmp_path):
+    """Reference pattern: httptest server per test (pkg/login login_test)."""
+    import threading
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    from gpud_amd.pkg import metadata
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.login import do_login
+    from gpud_amd.pkg.sqlite_util import open_ro
+
+    seen = {}
+
+    class H(BaseHTTPRequestHandler):
+        def do_POST(self):
+            import json as _json
+
+            body = _json.loads(self.rfile.read(int(self.headers["Content-Length"])))
+            seen.update(body)
+            if body.get("token") == "bad":
+                self.send_response(401)
+                self.end_headers()
+                self.wfile.write(b"denied")
+                return
+            resp = _json.dumps(
+                {
+                    "machineID": "cp-assigned-id",
+                    "token": "rotated-token",
+                    "machineProof": "proof123",
+                }
+            ).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(resp)))
+            self.end_headers()
+            self.wfile.write(resp)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), H)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    endpoint = f"http://127.0.0.1:{srv.server_port}"
+    try:
+        cfg = Config(data_dir=str(tmp_path))
+        import os
+
+        os.makedirs(cfg.data_dir, exist_ok=True)
+        # rejected token surfaces as an error string, persists nothing
+        err = do_login(cfg, token="bad", endpoint=endpoint, gpu_count=8)
+        assert err is not None and "401" in err
+        # accepted login persists CP-assigned identity + rotated token
+        err = do_login(cfg, token="good", endpoint=endpoint, node_group="g1")
+        assert err is None
+        assert seen["token"] == "good" and seen["nodeGroup"] == "g1"
+        conn = open_ro(cfg.state_path)
+        assert metadata.get_value(conn, metadata.KEY_MACHINE_ID) == "cp-assigned-id"
+        assert metadata.get_value(conn, metadata.KEY_TOKEN) == "rotated-token"
+        assert metadata.get_value(conn, metadata.KEY_MACHINE_PROOF) == "proof123"
+        assert metadata.get_value(conn, metadata.KEY_ENDPOINT) == endpoint
+        conn.close()
+        # unreachable endpoint degrades to an error, not an exception
+        err = do_login(cfg, token="x", endpoint="http://127.0.0.1:1", timeout=2)
+        assert err is not None and "failed" in err
+    finally:
+        srv.shutdown()
