@@ -112,3 +112,55 @@ def test_v1_http_transport_roundtrip(fake_cp, monkeypatch, tmp_path):
         s.stop()
     finally:
         core.close()
+
+
+def test_reconnect_backoff_doubles_and_caps(monkeypatch, tmp_path):
+    """Reference: session_reconnect.go:190-234 — exponential backoff with
+    jitter, capped, reset by a healthy stream. Tested with injected
+    sleep/jitter fns (the reference pattern: timeAfterFunc/jitterFunc)."""
+    import os
+
+    os.environ["GPUD_AMDSMI_MOCK"] = "1"
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.session.session import (
+        RECONNECT_BASE,
+        RECONNECT_MAX,
+        Session,
+    )
+
+    core = build_core(in_memory_db=True, kmsg_writable=False, record_reboot=False)
+    sleeps = []
+    attempts = {"n": 0}
+
+    def reader():
+        attempts["n"] += 1
+        if attempts["n"] == 4:
+            # 4th attempt: one healthy frame, then the stream dies again
+            yield {"req_id": "x", "method": "ping", "data": {}}
+        if attempts["n"] >= 6:
+            raise SystemExit  # stop the loop from inside
+        raise ConnectionError("stream down")
+
+    sent = []
+    s = Session(
+        core,
+        endpoint="http://unused",
+        open_reader=reader,
+        send_response=sent.append,
+        sleep_fn=sleeps.append,
+        jitter_fn=lambda: 1.0,  # deterministic
+    )
+    try:
+        s._serve_loop()
+    except SystemExit:
+        pass
+    finally:
+        core.close()
+    # three failed attempts: base, 2x, 4x; healthy frame resets to base
+    assert sleeps[0] == RECONNECT_BASE
+    assert sleeps[1] == RECONNECT_BASE * 2
+    assert sleeps[2] == RECONNECT_BASE * 4
+    assert sleeps[3] == RECONNECT_BASE  # reset after the healthy frame
+    assert all(x <= RECONNECT_MAX for x in sleeps)
+    assert sent and sent[0]["data"].get("pong") is True
+    assert s.reconnects >= 4
