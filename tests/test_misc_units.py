@@ -412,3 +412,40 @@ def test_login_against_fake_control_plane(tmp_path):
         assert err is not None and "failed" in err
     finally:
         srv.shutdown()
+
+
+def test_run_bash_timeout_kills_process_group():
+    """A timed-out script's whole process GROUP dies (reference pkg/process:
+    group kill), including children it spawned — no orphan `sleep`s."""
+    import subprocess
+    import time
+
+    from gpud_amd.pkg.process_runner import run_bash
+
+    marker = f"gpud-test-orphan-{time.time_ns()}"
+    res = run_bash(
+        f"(sleep 300 && echo {marker}) & echo started; wait",
+        timeout_seconds=1.0,
+    )
+    assert res.timed_out
+    assert "started" in res.output
+    time.sleep(0.2)
+    out = subprocess.run(
+        ["ps", "axo", "args"], capture_output=True, text=True
+    ).stdout
+    assert marker not in out, "child of timed-out script still running"
+
+
+def test_run_bash_output_capped():
+    from gpud_amd.pkg.process_runner import run_bash
+
+    res = run_bash("yes x | head -c 100000", max_output_bytes=1024)
+    assert res.exit_code == 0
+    assert len(res.output) <= 1024 + 64  # cap plus truncation marker slack
+
+
+def test_run_bash_env_passthrough():
+    from gpud_amd.pkg.process_runner import run_bash
+
+    res = run_bash("echo $GPUD_TEST_VAR", env={"GPUD_TEST_VAR": "hello42"})
+    assert "hello42" in res.output
