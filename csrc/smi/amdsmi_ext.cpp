@@ -573,6 +573,33 @@ py::dict energy_count(int index) {
   return d;
 }
 
+// PCIe link health: width/speed plus the replay / L0-recovery / NAK
+// counters — the PCIe analog of the reference's NVLink replay/recovery/CRC
+// error counters (components/accelerator/nvidia/nvlink/nvlink.go:86-93).
+py::dict pcie_info(int index) {
+  auto h = handle_at(index);
+  amdsmi_pcie_info_t info;
+  std::memset(&info, 0, sizeof(info));
+  {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> call_lk(g_call_mu);
+    check(amdsmi_get_pcie_info(h, &info), "amdsmi_get_pcie_info");
+  }
+  py::dict d;
+  d["max_width"] = info.pcie_static.max_pcie_width;
+  d["max_speed_gts"] = info.pcie_static.max_pcie_speed;
+  d["interface_version"] = info.pcie_static.pcie_interface_version;
+  d["width"] = info.pcie_metric.pcie_width;
+  d["speed_mts"] = info.pcie_metric.pcie_speed;
+  d["bandwidth_mbps"] = info.pcie_metric.pcie_bandwidth;
+  d["replay_count"] = info.pcie_metric.pcie_replay_count;
+  d["l0_to_recovery_count"] = info.pcie_metric.pcie_l0_to_recovery_count;
+  d["replay_rollover_count"] = info.pcie_metric.pcie_replay_roll_over_count;
+  d["nak_sent_count"] = info.pcie_metric.pcie_nak_sent_count;
+  d["nak_received_count"] = info.pcie_metric.pcie_nak_received_count;
+  return d;
+}
+
 // ---------------------------------------------------------------------------
 // partitioning + CPER RAS records
 // ---------------------------------------------------------------------------
@@ -1179,6 +1206,8 @@ PYBIND11_MODULE(_amdsmi, m) {
   m.def("xgmi_info", &xgmi_info, py::arg("index"));
   m.def("link_metrics", &link_metrics, py::arg("index"));
   m.def("energy_count", &energy_count, py::arg("index"));
+  m.def("pcie_info", &pcie_info, py::arg("index"),
+        "PCIe link width/speed + replay/recovery/NAK counters");
   m.def("partition_info", &partition_info, py::arg("index"),
         "Compute/memory partition mode and accelerator partition profile");
   m.def("cper_entries", &cper_entries, py::arg("index"),

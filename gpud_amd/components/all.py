@@ -21,6 +21,7 @@ from .accelerator import (
     gpu_counts,
     gpu_memory,
     partition,
+    pcie,
     peer_mem,
     power,
     power_management,
@@ -68,6 +69,7 @@ def all_init_funcs() -> List[InitFunc]:
         processes.new,
         bad_pages.new,
         partition.new,
+        pcie.new,
         cper.new,
         temperature.new,
         utilization.new,

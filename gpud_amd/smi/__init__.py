@@ -113,6 +113,9 @@ class Device:
     def partition_info(self) -> Dict[str, Any]:
         return self._b.partition_info(self.index)
 
+    def pcie_info(self) -> Dict[str, Any]:
+        return self._b.pcie_info(self.index)
+
     def cper_entries(
         self, severity_mask: int = 0xFFFFFFFF, cursor: int = 0
     ) -> Dict[str, Any]:

@@ -292,6 +292,22 @@ class MockBackend:
             )
         return {"num_links": len(links), "links": links}
 
+    def pcie_info(self, i: int) -> Dict[str, Any]:
+        s = self._check(i)
+        return {
+            "max_width": 16,
+            "max_speed_gts": 32,
+            "interface_version": 5,
+            "width": s.get("pcie_width", 16),
+            "speed_mts": s.get("pcie_speed_mts", 32000),
+            "bandwidth_mbps": 512000,
+            "replay_count": s.get("pcie_replay_count", 0),
+            "l0_to_recovery_count": s.get("pcie_l0_to_recovery_count", 0),
+            "replay_rollover_count": 0,
+            "nak_sent_count": s.get("pcie_nak_sent_count", 0),
+            "nak_received_count": s.get("pcie_nak_received_count", 0),
+        }
+
     def partition_info(self, i: int) -> Dict[str, Any]:
         s = self._check(i)
         return {

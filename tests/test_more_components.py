@@ -482,3 +482,48 @@ def test_partition_policy_hot_settable_via_updateconfig(mock_core):
     )
     mock_core.shared_snapshots.refresh()
     assert comp.trigger_check().health == HealthStateType.HEALTHY
+
+
+def test_pcie_healthy_then_downtrained(mock_core):
+    backend = mock_core.smi_instance._b
+    comp = mock_core.registry.get("accelerator-amd-pcie")
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "PCIe link healthy" in cr.reason
+    backend.state[0]["pcie_width"] = 8  # down-trained from x16
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "down-trained" in cr.reason and "x8" in cr.reason
+    del backend.state[0]["pcie_width"]
+
+
+def test_pcie_error_counter_deltas(mock_core):
+    from gpud_amd.components.accelerator.pcie import EVENT_NAME
+
+    backend = mock_core.smi_instance._b
+    comp = mock_core.registry.get("accelerator-amd-pcie")
+    comp.trigger_check()  # prime counters
+    backend.state[1]["pcie_replay_count"] = 42
+    backend.state[1]["pcie_nak_sent_count"] = 3
+    cr = comp.trigger_check()
+    evs = comp.events(utcnow() - datetime.timedelta(minutes=5))
+    assert any(EVENT_NAME == e.name and "replay +42" in e.message for e in evs)
+    # sustained rate over the window degrades
+    bucket = mock_core.event_store.bucket("accelerator-amd-pcie")
+    from gpud_amd.apiv1.types import Event, EventType
+
+    now = utcnow()
+    for i in range(8):
+        bucket.insert(
+            Event(
+                time=now - datetime.timedelta(minutes=i),
+                component=comp.name,
+                name=EVENT_NAME,
+                type=EventType.WARNING,
+                message=f"synthetic {i}",
+            )
+        )
+    comp.get_now = lambda: now
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "sustained PCIe" in cr.reason
