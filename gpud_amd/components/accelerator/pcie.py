@@ -92,8 +92,12 @@ class PCIeComponent(TickerComponent, SmiComponentMixin):
             n += 1
             width, speed = int(pi.get("width", 0)), int(pi.get("speed_mts", 0))
             max_w = int(pi.get("max_width", 0))
-            # max speed is reported in GT/s, current in MT/s
-            max_s_mts = int(pi.get("max_speed_gts", 0)) * 1000
+            # the header documents max speed in GT/s but the live driver
+            # returns MT/s (measured: 32000 on a Gen5 x16 MI355X, where the
+            # doc'd unit would read 32) — normalize either convention
+            max_s_mts = int(pi.get("max_speed_gts", 0))
+            if 0 < max_s_mts < 1000:
+                max_s_mts *= 1000
             self._gauges.set(
                 "accelerator_amd_pcie_link_width", "Current PCIe width",
                 width, uuid=uuid,
