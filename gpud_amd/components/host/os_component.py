@@ -19,7 +19,6 @@ from ...apiv1.types import (
     HealthStateType,
     RepairActionType,
     SuggestedActions,
-    utcnow,
 )
 from ...pkg import host as pkghost
 from ...pkg.pstore import Scanner
