@@ -56,6 +56,28 @@ def run(
     expected_memory_partition: str = typer.Option(
         "", help="alert if memory partition mode differs (e.g. NPS1)"
     ),
+    events_retention_days: float = typer.Option(
+        3.0, help="days to retain component events (reference: "
+        "events-retention-period)"
+    ),
+    compact_period_hours: float = typer.Option(24.0),
+    components: str = typer.Option(
+        "", help="comma-separated allow-list of components to enable"
+    ),
+    disabled_components: str = typer.Option(
+        "", help="comma-separated components to disable"
+    ),
+    kernel_modules_to_check: str = typer.Option(
+        "", help="comma-separated kernel modules that must be loaded"
+    ),
+    temperature_margin_celsius: float = typer.Option(
+        10.0, help="degraded when a temp is within this margin of its limit "
+        "(reference: threshold-celsius-slowdown-margin)"
+    ),
+    ras_reboot_threshold: int = typer.Option(
+        2, help="reboots tolerated before RAS errors escalate to hardware "
+        "inspection (reference: xid-reboot-threshold)"
+    ),
     poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
@@ -115,6 +137,20 @@ def run(
     cfg.expected_xgmi_link_count = expected_xgmi_link_count
     cfg.expected_compute_partition = expected_compute_partition
     cfg.expected_memory_partition = expected_memory_partition
+    cfg.events_retention_days = events_retention_days
+    cfg.compact_period_hours = compact_period_hours
+    if components:
+        cfg.enabled_components = [c.strip() for c in components.split(",") if c.strip()]
+    if disabled_components:
+        cfg.disabled_components = [
+            c.strip() for c in disabled_components.split(",") if c.strip()
+        ]
+    if kernel_modules_to_check:
+        cfg.kernel_modules_to_check = [
+            m.strip() for m in kernel_modules_to_check.split(",") if m.strip()
+        ]
+    cfg.temperature_margin_threshold_c = temperature_margin_celsius
+    cfg.ras_reboot_threshold = ras_reboot_threshold
     cfg.expected_ib_ports = expected_ib_ports
     cfg.expected_ib_rate_gbps = expected_ib_rate_gbps
     cfg.poll_interval_seconds = poll_interval_seconds

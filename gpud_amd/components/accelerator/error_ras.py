@@ -62,7 +62,11 @@ class ErrorRASComponent(TickerComponent):
         self._reboot_store = inst.reboot_event_store
         self._kmsg = inst.kmsg_reader
         self._syncer: Optional[Syncer] = None
-        self.reboot_threshold = DEFAULT_REBOOT_THRESHOLD
+        cfg = inst.config
+        self.reboot_threshold = int(
+            getattr(cfg, "ras_reboot_threshold", DEFAULT_REBOOT_THRESHOLD)
+            or DEFAULT_REBOOT_THRESHOLD
+        )
         self.get_now: Callable = utcnow
 
     @property

@@ -62,6 +62,9 @@ class Config:
     # network-latency probe targets: list of (host, port)
     latency_targets: List[Any] = field(default_factory=list)
     temperature_margin_threshold_c: float = 10.0
+    # error-ras escalation: reboots tolerated before HARDWARE_INSPECTION
+    # (reference: xid-reboot-threshold flag, cmd/gpud/run/command.go)
+    ras_reboot_threshold: int = 2
     zombie_degraded_threshold: int = 1000
     zombie_unhealthy_threshold: int = 2000
     # control plane

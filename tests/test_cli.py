@@ -228,3 +228,56 @@ def test_scan_alias_check(monkeypatch):
     monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
     r = CliRunner().invoke(app, ["check", "--mock", "--output", "json"])
     assert r.exit_code == 0
+
+
+def test_run_flags_map_to_config(monkeypatch):
+    """The reference's key run flags have CLI analogs that reach Config
+    (cmd/gpud/run/command.go flag surface)."""
+    import subprocess
+    import sys
+
+    import gpud_amd.cli.main as cli_main
+
+    captured = {}
+
+    def fake_build_core(cfg, **kw):
+        captured["cfg"] = cfg
+        raise SystemExit(0)  # stop before the server boots
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setattr("gpud_amd.bootstrap.build_core", fake_build_core)
+    from typer.testing import CliRunner
+
+    r = CliRunner().invoke(
+        cli_main.app,
+        [
+            "run", "--in-memory-db",
+            "--events-retention-days", "7",
+            "--components", "cpu,memory",
+            "--kernel-modules-to-check", "amdgpu,amdkfd",
+            "--temperature-margin-celsius", "5",
+            "--ras-reboot-threshold", "4",
+            "--expected-compute-partition", "spx",
+        ],
+    )
+    cfg = captured["cfg"]
+    assert cfg.events_retention_days == 7
+    assert cfg.enabled_components == ["cpu", "memory"]
+    assert cfg.kernel_modules_to_check == ["amdgpu", "amdkfd"]
+    assert cfg.temperature_margin_threshold_c == 5
+    assert cfg.ras_reboot_threshold == 4
+    assert cfg.expected_compute_partition == "spx"
+
+
+def test_ras_reboot_threshold_reaches_component(monkeypatch, tmp_path):
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    cfg = Config(data_dir=str(tmp_path), ras_reboot_threshold=5)
+    core = build_core(cfg, in_memory_db=True, kmsg_writable=False, record_reboot=False)
+    try:
+        comp = core.registry.get("accelerator-amd-error-ras")
+        assert comp.reboot_threshold == 5
+    finally:
+        core.close()
