@@ -527,3 +527,22 @@ def test_pcie_error_counter_deltas(mock_core):
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.DEGRADED
     assert "sustained PCIe" in cr.reason
+
+
+def test_partition_ttl_cache(mock_core):
+    """Partition reads are TTL-cached (poll-cycle cost control) but a
+    zero TTL reads through — the seam tests use."""
+    backend = mock_core.smi_instance._b
+    comp = mock_core.registry.get("accelerator-amd-partition")
+    comp.trigger_check()
+    backend.state[0]["compute_partition"] = "CPX"
+    # within the TTL the cached SPX answer stands
+    cr = comp.trigger_check()
+    assert "CPX" not in cr.reason
+    # expiring the cache picks up the live mode
+    comp.cache_ttl_seconds = 0.0
+    comp._cache = {}
+    cr = comp.trigger_check()
+    assert "CPX" in cr.reason
+    backend.state[0]["compute_partition"] = "SPX"
+    comp._cache = {}
