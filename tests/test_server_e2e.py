@@ -245,3 +245,16 @@ def test_malformed_requests_never_500(daemon):
         )
         assert r.status_code < 500, (path, r.status_code, r.text[:200])
     h.close()
+
+
+def test_v1_metrics_includes_new_component_gauges(daemon):
+    """The newest components' gauges survive the scrape→store→/v1/metrics
+    path (the gpud_component label filter must not drop them)."""
+    core, server, client = daemon
+    core.registry.get("accelerator-amd-pcie").trigger_check()
+    core.registry.get("accelerator-amd-partition").trigger_check()
+    core.metrics_syncer.sync_once()
+    metrics = client.get_metrics()
+    names = {m.name for comp in metrics.values() for m in comp}
+    assert any("pcie_link_width" in n for n in names), sorted(names)[:20]
+    assert any("partition_count" in n for n in names)
