@@ -258,3 +258,26 @@ def test_v1_metrics_includes_new_component_gauges(daemon):
     names = {m.name for comp in metrics.values() for m in comp}
     assert any("pcie_link_width" in n for n in names), sorted(names)[:20]
     assert any("partition_count" in n for n in names)
+
+
+def test_client_error_paths(daemon):
+    """Client surfaces server 4xx as exceptions with useful messages, and
+    wait_healthz times out cleanly on a dead port."""
+    import httpx
+
+    from gpud_amd.client import Client
+
+    core, server, client = daemon
+    with pytest.raises(httpx.HTTPStatusError):
+        client.deregister_component("no-such-component")
+    with pytest.raises(httpx.HTTPStatusError):
+        client.trigger_check(component="also-missing")
+    dead = Client("https://127.0.0.1:9")  # discard port, nothing listens
+    assert dead.wait_healthz(timeout=1.5) is False
+    dead.close()
+
+
+def test_trigger_tag_unknown_tag_empty(daemon):
+    core, server, client = daemon
+    out = client.trigger_check(tag="no-such-tag")
+    assert out.get("components", []) == [] or out.get("states", []) == []
