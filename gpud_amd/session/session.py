@@ -320,9 +320,72 @@ class Session:
         diag components). ``{"async": true}`` starts them in the
         background — results land in each component's cached health state
         (queryable via the ``states`` method or /v1/states)."""
+        if payload.get("report_id"):
+            # managed-diagnostic protocol (reference docs/INTEGRATION.md +
+            # pkg/session/diagnostic.go): the request carries report_id, a
+            # FIXED type and timeout_seconds — never user commands. We
+            # accept the amd_bug_report analog, run the bundle collection
+            # asynchronously under the timeout, upload to the presigned
+            # URL, and notify the control plane's failure endpoint on any
+            # failure (collection error, timeout, no artifact, upload
+            # failure after it produced nothing).
+            report_id = str(payload["report_id"])
+            dtype = payload.get("type", "amd_bug_report")
+            if dtype not in ("amd_bug_report", "nvidia_bug_report"):
+                return {"error": f"unsupported diagnostic type {dtype!r}"}
+            timeout_s = float(payload.get("timeout_seconds", 600))
+            upload_url = payload.get("upload_url", "")
+
+            def _notify_failure() -> None:
+                if not self.endpoint:
+                    return
+                try:
+                    self._client.post(
+                        f"{self.endpoint}/api/v1/diagnostics/{report_id}/failure",
+                        headers=self._headers(),
+                        json={"report_id": report_id},
+                        timeout=30,
+                    )
+                except Exception:
+                    logger.exception("diagnostic failure notification failed")
+
+            def _run_report():
+                import tempfile
+
+                from ..pkg.bundle import collect_bundle, upload_bundle
+
+                path = os.path.join(
+                    tempfile.gettempdir(), f"gpud-diag-{report_id}.tar.gz"
+                )
+                done = threading.Event()
+                err_box: list = []
+
+                def _collect():
+                    try:
+                        collect_bundle(path, core=self.core)
+                    except Exception as e:  # noqa: BLE001
+                        err_box.append(str(e))
+                    finally:
+                        done.set()
+
+                t = threading.Thread(target=_collect, daemon=True)
+                t.start()
+                if not done.wait(timeout_s) or err_box or not os.path.exists(path):
+                    logger.warning("diagnostic %s failed/timed out", report_id)
+                    _notify_failure()
+                    return
+                if upload_url:
+                    err = upload_bundle(path, upload_url)
+                    if err:
+                        logger.warning("diagnostic %s upload: %s", report_id, err)
+                        _notify_failure()
+
+            threading.Thread(
+                target=_run_report, daemon=True, name=f"gpud-diag-{report_id}"
+            ).start()
+            return {"status": "accepted", "report_id": report_id, "type": dtype}
         if payload.get("bundle"):
-            # support-bundle mode (reference diagnostic.go collects
-            # nvidia-bug-report.sh output and PUTs it to a presigned URL)
+            # direct support-bundle mode (synchronous; operator-driven)
             import tempfile
 
             from ..pkg.bundle import collect_bundle, upload_bundle

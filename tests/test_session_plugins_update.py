@@ -473,3 +473,86 @@ def test_support_bundle_collect_and_session(tmp_path, monkeypatch):
         assert tarfile.is_tarfile(resp["data"]["bundle"])
     finally:
         core.close()
+
+
+def test_managed_diagnostic_protocol(tmp_path, monkeypatch):
+    """Reference docs/INTEGRATION.md: session diagnostic carries report_id +
+    fixed type + timeout; runs async; uploads to the presigned URL; POSTs
+    the failure endpoint when the upload fails."""
+    import threading
+    import time
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.session import Session
+
+    hits = []
+
+    class H(BaseHTTPRequestHandler):
+        def do_PUT(self):
+            self.rfile.read(int(self.headers.get("Content-Length", 0)))
+            hits.append(("PUT", self.path))
+            code = 500 if "fail-upload" in self.path else 200
+            self.send_response(code)
+            self.end_headers()
+
+        def do_POST(self):
+            self.rfile.read(int(self.headers.get("Content-Length", 0)))
+            hits.append(("POST", self.path))
+            self.send_response(200)
+            self.end_headers()
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), H)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    base = f"http://127.0.0.1:{srv.server_port}"
+
+    core = build_core(
+        Config(data_dir=str(tmp_path)),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    s = Session(
+        core,
+        endpoint=base,
+        open_reader=lambda: iter(()),
+        send_response=lambda f: None,
+    )
+    try:
+        # unsupported type is refused
+        resp = s.process_request(
+            {"req_id": "d0", "method": "diagnostic",
+             "data": {"report_id": "r0", "type": "run_my_script"}}
+        )
+        assert "unsupported" in resp["data"]["error"]
+        # happy path: accepted, bundle uploaded
+        resp = s.process_request(
+            {"req_id": "d1", "method": "diagnostic",
+             "data": {"report_id": "r1", "upload_url": f"{base}/up/ok"}}
+        )
+        assert resp["data"]["status"] == "accepted"
+        for _ in range(100):
+            if ("PUT", "/up/ok") in hits:
+                break
+            time.sleep(0.1)
+        assert ("PUT", "/up/ok") in hits
+        assert not any(m == "POST" for m, _ in hits)
+        # failing upload triggers the failure endpoint
+        s.process_request(
+            {"req_id": "d2", "method": "diagnostic",
+             "data": {"report_id": "r2", "upload_url": f"{base}/up/fail-upload"}}
+        )
+        for _ in range(100):
+            if ("POST", "/api/v1/diagnostics/r2/failure") in hits:
+                break
+            time.sleep(0.1)
+        assert ("POST", "/api/v1/diagnostics/r2/failure") in hits
+    finally:
+        s.stop()
+        core.close()
+        srv.shutdown()
