@@ -87,6 +87,7 @@ class Spec:
     timeout_seconds: float = 60.0
     interval_seconds: float = 600.0
     component_list: List[str] = dc_field(default_factory=list)
+    component_list_file: str = ""
 
     @property
     def component_name(self) -> str:
@@ -113,6 +114,7 @@ class Spec:
             timeout_seconds=parse_duration(timeout),
             interval_seconds=parse_duration(interval),
             component_list=list(d.get("component_list") or []),
+            component_list_file=str(d.get("component_list_file") or ""),
         )
 
     def to_dict(self) -> Dict[str, Any]:
@@ -160,6 +162,10 @@ class Spec:
             return f"invalid plugin_type {self.plugin_type!r}"
         if not self.steps:
             return "at least one step is required"
+        if self.plugin_type == PLUGIN_TYPE_COMPONENT_LIST and not (
+            self.component_list or self.component_list_file
+        ):
+            return "component_list or component_list_file is required"
         return None
 
 
@@ -212,17 +218,27 @@ def _dig(obj: Any, dotpath: str) -> Any:
 class PluginComponent(TickerComponent):
     """One custom plugin as a registry component."""
 
-    def __init__(self, spec: Spec, param: str = ""):
+    def __init__(
+        self,
+        spec: Spec,
+        param: str = "",
+        entry_name: str = "",
+        run_mode_override: str = "",
+    ):
         super().__init__()
         self.spec = spec
         self.param = param
+        self.entry_name = entry_name
         self.poll_interval = spec.interval_seconds
-        self.run_mode = spec.run_mode
+        # component-specific run_mode wins over the parent plugin's
+        # (docs/PLUGIN.md "Parameter Inheritance and Priority")
+        self.run_mode = run_mode_override or spec.run_mode
 
     @property
     def name(self) -> str:
-        if self.param:
-            return f"{self.spec.component_name}-{self.param}"
+        sub = self.entry_name or self.param
+        if sub:
+            return f"{self.spec.component_name}-{sub}"
         return self.spec.component_name
 
     def tags(self) -> List[str]:
@@ -235,8 +251,13 @@ class PluginComponent(TickerComponent):
         outputs = []
         for step in self.spec.steps:
             script = step.decoded()
-            if self.param:
-                script = script.replace("${NAME}", self.param)
+            # ${NAME} = component entry name, ${PAR} = its parameter
+            # (docs/PLUGIN.md "Parameter Substitution"); name-only entries
+            # historically doubled as the parameter, so ${NAME} falls back
+            # to the param when no entry name was parsed
+            if self.entry_name or self.param:
+                script = script.replace("${NAME}", self.entry_name or self.param)
+                script = script.replace("${PAR}", self.param)
             res = run_bash(script, timeout_seconds=self.spec.timeout_seconds)
             outputs.append(res.output)
             if res.timed_out or res.exit_code != 0:
@@ -312,9 +333,62 @@ class PluginComponent(TickerComponent):
         return health, "plugin succeeded", sa, extra
 
 
+def parse_component_list_entry(entry: str):
+    """One component-list entry -> (name, run_mode_override, param).
+
+    Reference formats (docs/PLUGIN.md "Component List Format"):
+    ``name#run_mode:param`` | ``name#run_mode`` | ``name:param`` | ``name``.
+    run_mode has the highest priority for that component; timeout/interval
+    always inherit from the parent plugin."""
+    run_mode = ""
+    param = ""
+    name = entry
+    if "#" in entry:
+        head, rest = entry.split("#", 1)
+        mode = rest.split(":", 1)[0]
+        # '#' only separates a run_mode; otherwise it is part of the name
+        # (the doc's indent-escape for names starting with '#')
+        if mode in ("auto", "manual", "once"):
+            name = head
+            if ":" in rest:
+                run_mode, param = rest.split(":", 1)
+            else:
+                run_mode = rest
+        elif ":" in entry:
+            name, param = entry.split(":", 1)
+    elif ":" in entry:
+        name, param = entry.split(":", 1)
+    return name.strip(), run_mode.strip(), param
+
+
+def load_component_list_file(path: str) -> List[str]:
+    """Plain-text component list: one entry per line; empty lines and lines
+    whose FIRST character is ``#`` are ignored (a name starting with # can
+    be escaped by indenting with a space — docs/PLUGIN.md:174)."""
+    entries = []
+    with open(path) as f:
+        for line in f:
+            line = line.rstrip("\n")
+            if not line.strip() or line.startswith("#"):
+                continue
+            entries.append(line.strip())
+    return entries
+
+
 def make_components(spec: Spec) -> List[Component]:
     if spec.plugin_type == PLUGIN_TYPE_COMPONENT_LIST:
-        return [PluginComponent(spec, param=p) for p in spec.component_list]
+        entries = list(spec.component_list)
+        if spec.component_list_file:
+            entries.extend(load_component_list_file(spec.component_list_file))
+        out = []
+        for entry in entries:
+            name, run_mode, param = parse_component_list_entry(entry)
+            out.append(
+                PluginComponent(
+                    spec, param=param, entry_name=name, run_mode_override=run_mode
+                )
+            )
+        return out
     return [PluginComponent(spec)]
 
 

@@ -556,3 +556,62 @@ def test_managed_diagnostic_protocol(tmp_path, monkeypatch):
         s.stop()
         core.close()
         srv.shutdown()
+
+
+def test_component_list_entry_formats(tmp_path):
+    """The four entry formats + file-based lists + ${NAME}/${PAR}
+    substitution (reference docs/PLUGIN.md Component List Format)."""
+    from gpud_amd.pkg import custom_plugins as cp
+
+    assert cp.parse_component_list_entry("db#manual:/var/db") == (
+        "db", "manual", "/var/db",
+    )
+    assert cp.parse_component_list_entry("db#manual") == ("db", "manual", "")
+    assert cp.parse_component_list_entry("db:/var/db") == ("db", "", "/var/db")
+    assert cp.parse_component_list_entry("db") == ("db", "", "")
+
+    listfile = tmp_path / "list.txt"
+    listfile.write_text(
+        "# a comment\n"
+        "\n"
+        "root:/\n"
+        "slow#manual:/var\n"
+        " #weird\n"  # indented => a real (odd) name, per the doc's escape
+    )
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "fmt",
+            "plugin_type": "component_list",
+            "component_list_file": str(listfile),
+            "health_state_plugin": {
+                "steps": [
+                    {"run_bash_script": {
+                        "content_type": "plaintext",
+                        "script": "echo name=${NAME} par=${PAR}",
+                    }}
+                ]
+            },
+        }
+    )
+    assert spec.validate() is None
+    comps = cp.make_components(spec)
+    by_name = {c.name: c for c in comps}
+    assert set(by_name) == {
+        "custom-plugin-fmt-root",
+        "custom-plugin-fmt-slow",
+        "custom-plugin-fmt-#weird",
+    }
+    assert by_name["custom-plugin-fmt-slow"].run_mode == "manual"
+    cr = by_name["custom-plugin-fmt-root"].trigger_check()
+    assert "name=root par=/" in cr.raw_output
+    # a list-type spec with neither list nor file is invalid
+    bad = cp.Spec.from_dict(
+        {
+            "plugin_name": "empty",
+            "plugin_type": "component_list",
+            "health_state_plugin": {
+                "steps": [{"run_bash_script": {"script": "true"}}]
+            },
+        }
+    )
+    assert bad.validate() is not None
