@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import base64
 import json
+import os
 import re
 from dataclasses import dataclass
 from dataclasses import field as dc_field
@@ -84,6 +85,9 @@ class Spec:
     tags: List[str] = dc_field(default_factory=list)
     steps: List[RunBashScript] = dc_field(default_factory=list)
     json_paths: List[JSONPathRule] = dc_field(default_factory=list)
+    # optional plugin-output log with ${PLUGIN}/${TRIGGER} substitution
+    # (reference: docs/PLUGIN.md "Log Path Variable Substitution")
+    log_path: str = ""
     timeout_seconds: float = 60.0
     interval_seconds: float = 600.0
     component_list: List[str] = dc_field(default_factory=list)
@@ -111,6 +115,7 @@ class Spec:
             tags=list(d.get("tags") or []),
             steps=steps,
             json_paths=rules,
+            log_path=str(parser.get("log_path") or ""),
             timeout_seconds=parse_duration(timeout),
             interval_seconds=parse_duration(interval),
             component_list=list(d.get("component_list") or []),
@@ -274,6 +279,7 @@ class PluginComponent(TickerComponent):
                     run_mode=self.spec.run_mode,
                 )
         raw = "\n".join(outputs)
+        self._log_output(raw)
         health, reason, actions, extra = self._parse(raw)
         return CheckResult(
             self.name,
@@ -285,6 +291,31 @@ class PluginComponent(TickerComponent):
             component_type=ComponentType.CUSTOM_PLUGIN,
             run_mode=self.spec.run_mode,
         )
+
+    def _log_output(self, raw: str) -> None:
+        """Append one RFC3339-stamped line to the parser's log_path with
+        ${PLUGIN}/${TRIGGER} substituted; skipped when a referenced
+        variable is empty (docs/PLUGIN.md Log Path Variable Substitution)."""
+        path = self.spec.log_path
+        if not path:
+            return
+        plugin, trigger = self.spec.plugin_name, (self.run_mode or "")
+        if "${PLUGIN}" in path and not plugin:
+            return
+        if "${TRIGGER}" in path and not trigger:
+            return
+        path = path.replace("${PLUGIN}", plugin).replace("${TRIGGER}", trigger)
+        try:
+            os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+            from ..apiv1.types import rfc3339, utcnow
+
+            with open(path, "a") as f:
+                f.write(
+                    f"[{rfc3339(utcnow())}] plugin={plugin} "
+                    f"trigger={trigger} output={raw.strip()[-2048:]}\n"
+                )
+        except OSError:
+            pass  # logging is best-effort; never fail the check over it
 
     def _parse(self, raw: str):
         if not self.spec.json_paths:

@@ -615,3 +615,32 @@ def test_component_list_entry_formats(tmp_path):
         }
     )
     assert bad.validate() is not None
+
+
+def test_plugin_log_path_substitution(tmp_path):
+    """log_path writes RFC3339-stamped output lines with ${PLUGIN}/${TRIGGER}
+    substituted (reference docs/PLUGIN.md)."""
+    from gpud_amd.pkg import custom_plugins as cp
+
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "logger",
+            "plugin_type": "component",
+            "run_mode": "auto",
+            "health_state_plugin": {
+                "steps": [{"run_bash_script": {"script": "echo hello-log"}}],
+                "parser": {
+                    "log_path": str(tmp_path / "${PLUGIN}" / "${TRIGGER}.log")
+                },
+            },
+        }
+    )
+    comp = cp.make_components(spec)[0]
+    comp.trigger_check()
+    comp.trigger_check()
+    log = (tmp_path / "logger" / "auto.log").read_text()
+    lines = [l for l in log.splitlines() if l]
+    assert len(lines) == 2
+    assert "plugin=logger trigger=auto" in lines[0]
+    assert "hello-log" in lines[0]
+    assert lines[0].startswith("[20")  # RFC3339 timestamp
