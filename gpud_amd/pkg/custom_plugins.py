@@ -320,9 +320,12 @@ class PluginComponent(TickerComponent):
     def _parse(self, raw: str):
         if not self.spec.json_paths:
             return HealthStateType.HEALTHY, "plugin succeeded", None, {}
-        # parse the LAST JSON-looking line of output (reference convention)
+        # the reference extracts the FIRST valid JSON object found in the
+        # output, even when embedded in other text (docs/PLUGIN.md parser
+        # notes); line-leading objects are tried first, then an embedded
+        # brace-scan, then the legacy last-line pass
         parsed = None
-        for line in reversed(raw.strip().splitlines()):
+        for line in raw.strip().splitlines():
             line = line.strip()
             if line.startswith("{"):
                 try:
@@ -330,6 +333,17 @@ class PluginComponent(TickerComponent):
                     break
                 except json.JSONDecodeError:
                     continue
+        if parsed is None:
+            decoder = json.JSONDecoder()
+            idx = raw.find("{")
+            while idx != -1 and parsed is None:
+                try:
+                    candidate, _end = decoder.raw_decode(raw[idx:])
+                    if isinstance(candidate, dict):
+                        parsed = candidate
+                except json.JSONDecodeError:
+                    pass
+                idx = raw.find("{", idx + 1)
         if parsed is None:
             return (
                 HealthStateType.UNHEALTHY,

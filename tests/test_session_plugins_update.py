@@ -644,3 +644,33 @@ def test_plugin_log_path_substitution(tmp_path):
     assert "plugin=logger trigger=auto" in lines[0]
     assert "hello-log" in lines[0]
     assert lines[0].startswith("[20")  # RFC3339 timestamp
+
+
+def test_plugin_parser_extracts_embedded_json():
+    """Parser finds the first valid JSON object even when embedded in other
+    text (reference docs/PLUGIN.md parser note)."""
+    from gpud_amd.pkg import custom_plugins as cp
+
+    spec = cp.Spec.from_dict(
+        {
+            "plugin_name": "embed",
+            "plugin_type": "component",
+            "health_state_plugin": {
+                "steps": [
+                    {"run_bash_script": {
+                        "script": 'echo \'prefix text {"result": "success"} suffix\''
+                    }}
+                ],
+                "parser": {
+                    "json_paths": [
+                        {"query": "result", "field": "result",
+                         "expect": {"regex": "^success$"}}
+                    ]
+                },
+            },
+        }
+    )
+    comp = cp.make_components(spec)[0]
+    cr = comp.trigger_check()
+    assert cr.health == "Healthy", cr.reason
+    assert cr.extra_info["result"] == "success"
