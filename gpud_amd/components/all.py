@@ -12,6 +12,7 @@ from typing import Callable, List
 from .base import Component, GPUdInstance
 
 from .accelerator import (
+    bad_envs,
     bad_pages,
     clock_speed,
     cper,
@@ -68,6 +69,7 @@ def all_init_funcs() -> List[InitFunc]:
         power_management.new,
         processes.new,
         bad_pages.new,
+        bad_envs.new,
         partition.new,
         pcie.new,
         cper.new,

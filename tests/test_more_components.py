@@ -546,3 +546,44 @@ def test_partition_ttl_cache(mock_core):
     assert "CPX" in cr.reason
     backend.state[0]["compute_partition"] = "SPX"
     comp._cache = {}
+
+
+def test_bad_envs_detection(mock_core):
+    """Dangerous global GPU env vars degrade; per-process device hiding in
+    the daemon scope is tolerated (launchers set it legitimately)."""
+    from gpud_amd.components.accelerator.bad_envs import scan_bad_envs
+
+    comp = mock_core.registry.get("accelerator-amd-bad-envs")
+    comp.get_scopes = lambda: {"/etc/environment": {}, "pid1": {}, "daemon": {}}
+    assert comp.trigger_check().health == HealthStateType.HEALTHY
+
+    comp.get_scopes = lambda: {
+        "/etc/environment": {"AMD_SERIALIZE_KERNEL": "3"},
+        "pid1": {},
+        "daemon": {},
+    }
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "AMD_SERIALIZE_KERNEL" in cr.reason
+
+    # value-sensitive rules
+    assert scan_bad_envs({"pid1": {"HSA_ENABLE_SDMA": "1"}}) == []
+    assert scan_bad_envs({"pid1": {"HSA_ENABLE_SDMA": "0"}}) != []
+    assert scan_bad_envs({"pid1": {"NCCL_P2P_DISABLE": "0"}}) == []
+    assert scan_bad_envs({"pid1": {"NCCL_P2P_DISABLE": "1"}}) != []
+    # device hiding: global scope bad, daemon scope tolerated
+    assert scan_bad_envs({"pid1": {"HIP_VISIBLE_DEVICES": "0"}}) != []
+    assert scan_bad_envs({"daemon": {"HIP_VISIBLE_DEVICES": "0"}}) == []
+
+
+def test_bad_envs_etc_environment_parse(tmp_path):
+    from gpud_amd.components.accelerator.bad_envs import read_etc_environment
+
+    f = tmp_path / "environment"
+    f.write_text(
+        '# comment\nPATH="/usr/bin"\nAMD_SERIALIZE_KERNEL=3\nBROKENLINE\n'
+    )
+    env = read_etc_environment(str(f))
+    assert env["AMD_SERIALIZE_KERNEL"] == "3"
+    assert env["PATH"] == "/usr/bin"
+    assert "BROKENLINE" not in env
