@@ -27,6 +27,7 @@ Degraded with the offending scope/variable named.
 from __future__ import annotations
 
 import os
+import time
 from typing import Callable, Dict, List, Tuple
 
 from ...apiv1.types import HealthStateType
@@ -119,17 +120,24 @@ class BadEnvsComponent(TickerComponent):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
+        # the global scopes change only via operator action; re-read them
+        # on a 60 s TTL and keep the daemon's own env live
+        self._global_cache: Dict[str, Dict[str, str]] = {}
+        self._cached_at = 0.0
+        self.cache_ttl_seconds = 60.0
         self.get_scopes: Callable[[], Dict[str, Dict[str, str]]] = (
             self._default_scopes
         )
 
-    @staticmethod
-    def _default_scopes() -> Dict[str, Dict[str, str]]:
-        return {
-            "/etc/environment": read_etc_environment(),
-            "pid1": read_pid1_environ(),
-            "daemon": dict(os.environ),
-        }
+    def _default_scopes(self) -> Dict[str, Dict[str, str]]:
+        now = time.monotonic()
+        if not self._global_cache or now - self._cached_at > self.cache_ttl_seconds:
+            self._global_cache = {
+                "/etc/environment": read_etc_environment(),
+                "pid1": read_pid1_environ(),
+            }
+            self._cached_at = now
+        return {**self._global_cache, "daemon": dict(os.environ)}
 
     @property
     def name(self) -> str:
