@@ -587,3 +587,17 @@ def test_bad_envs_etc_environment_parse(tmp_path):
     assert env["AMD_SERIALIZE_KERNEL"] == "3"
     assert env["PATH"] == "/usr/bin"
     assert "BROKENLINE" not in env
+
+
+def test_bad_envs_ttl_and_live_daemon_scope(mock_core, tmp_path, monkeypatch):
+    """Global scopes are TTL-cached; the daemon's own env is always live."""
+    comp = mock_core.registry.get("accelerator-amd-bad-envs")
+    comp.get_scopes = comp._default_scopes  # real path
+    monkeypatch.delenv("AMD_SERIALIZE_KERNEL", raising=False)
+    base = comp.trigger_check()
+    monkeypatch.setenv("AMD_SERIALIZE_KERNEL", "3")
+    cr = comp.trigger_check()  # daemon scope is read live, no TTL wait
+    assert cr.health == HealthStateType.DEGRADED
+    assert "daemon" in cr.reason
+    monkeypatch.delenv("AMD_SERIALIZE_KERNEL")
+    assert comp.trigger_check().health == base.health
