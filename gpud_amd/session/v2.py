@@ -1,18 +1,20 @@
 """Session v2: gRPC bidirectional stream to the control plane.
 
 Reference: pkg/session/v2/session.proto:12-53 — ``SessionService.Connect
-(stream AgentPacket) returns (stream ManagerPacket)``, a protobuf mirror of
-the v1 method set, selected via ``--session-protocol`` (reference:
-cmd/gpud/run/command.go:156). This environment has grpcio but no protobuf
-codegen toolchain, so packets are JSON-encoded bytes over a real gRPC
-stream-stream method (generic handlers, identity serializers); the method
-set maps 1:1 onto the v1 dispatcher like the reference's v2 adapter
-(pkg/session/session_v2_adapter.go:72).
+(stream AgentPacket) returns (stream ManagerPacket)``. Two wire modes:
 
-Packet shapes:
+- ``wire="proto"`` (default): the reference's actual protobuf wire format,
+  hand-encoded by ``protowire.py`` (no protoc in this environment) over
+  the reference's method path ``/gpud.session.v2.SessionService/Connect``.
+  The agent sends ``Hello`` then ``Result{request_id, payload_json}``
+  packets; decoded ManagerPacket oneofs map 1:1 onto the v1 dispatcher
+  exactly like the reference's v2 adapter
+  (pkg/session/session_v2_adapter.go:72).
+- ``wire="json"``: the plain JSON-bytes framing kept for debugging and
+  transport tests; packet shapes:
   manager -> agent: {"req_id": str, "method": str, "data": {...}}
   agent -> manager: {"req_id": str, "method": str, "data": {...}}
-The first agent packet is a hello: {"hello": {"machine_id", "token"}}.
+  with an initial {"hello": {"machine_id", "token"}} agent packet.
 """
 
 from __future__ import annotations
@@ -28,7 +30,9 @@ import grpc
 from ..pkg.log import logger
 from .session import Session
 
-SERVICE_METHOD = "/gpud.v2.SessionService/Connect"
+SERVICE_METHOD_JSON = "/gpud.v2.SessionService/Connect"
+# the reference's generated service path (session.proto package gpud.session.v2)
+SERVICE_METHOD_PROTO = "/gpud.session.v2.SessionService/Connect"
 
 
 def _ser(obj: dict) -> bytes:
@@ -49,19 +53,29 @@ class V2Session:
         machine_id: str = "",
         token: str = "",
         credentials: Optional[grpc.ChannelCredentials] = None,
+        wire: str = "proto",
     ):
         self.dispatcher = dispatcher
         self.endpoint = endpoint
         self.machine_id = machine_id
         self.token = token
         self.credentials = credentials
+        self.wire = wire
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._outbox: "queue.Queue[Optional[dict]]" = queue.Queue()
         self.reconnects = 0
 
     def _request_iter(self) -> Iterator[bytes]:
-        yield _ser({"hello": {"machine_id": self.machine_id, "token": self.token}})
+        if self.wire == "proto":
+            from .. import __version__
+            from . import protowire as pw
+
+            yield pw.hello_bytes(agent_version=__version__)
+        else:
+            yield _ser(
+                {"hello": {"machine_id": self.machine_id, "token": self.token}}
+            )
         while not self._stop.is_set():
             try:
                 item = self._outbox.get(timeout=0.5)
@@ -69,7 +83,7 @@ class V2Session:
                 continue
             if item is None:
                 return
-            yield _ser(item)
+            yield item if isinstance(item, bytes) else _ser(item)
 
     def _run_once(self) -> None:
         if self.credentials is not None:
@@ -78,13 +92,16 @@ class V2Session:
             channel = grpc.insecure_channel(self.endpoint)
         try:
             callable_ = channel.stream_stream(
-                SERVICE_METHOD,
+                SERVICE_METHOD_PROTO if self.wire == "proto" else SERVICE_METHOD_JSON,
                 request_serializer=lambda b: b,
                 response_deserializer=lambda b: b,
             )
             for raw in callable_(self._request_iter()):
                 if self._stop.is_set():
                     return
+                if self.wire == "proto":
+                    self._handle_proto_packet(raw)
+                    continue
                 try:
                     frame = _deser(raw)
                 except (ValueError, UnicodeDecodeError):
@@ -94,6 +111,32 @@ class V2Session:
                     self._outbox.put(resp)
         finally:
             channel.close()
+
+    def _handle_proto_packet(self, raw: bytes) -> None:
+        from . import protowire as pw
+
+        try:
+            frame = pw.manager_packet_to_frame(raw)
+        except ValueError:
+            logger.warning("v2: undecodable ManagerPacket (%d bytes)", len(raw))
+            return
+        if frame is None:
+            return
+        if frame.get("_control") == "hello_ack":
+            logger.info(
+                "v2 session established: manager %s revision %s",
+                frame.get("manager_instance_id", "?"),
+                frame.get("protocol_revision", "?"),
+            )
+            return
+        if frame.get("_control") == "drain_notice":
+            delay = frame.get("reconnect_after_millis", 0) / 1000.0
+            logger.info("v2 drain notice: reconnect after %.1fs", delay)
+            time.sleep(min(delay, 60.0))
+            raise grpc.RpcError()  # leave the stream; the loop reconnects
+        resp = self.dispatcher.process_request(frame)
+        if resp is not None:
+            self._outbox.put(pw.result_bytes(resp.get("req_id", ""), resp.get("data")))
 
     def start(self) -> None:
         self._thread = threading.Thread(
@@ -129,9 +172,14 @@ class V2Session:
 
 class FakeManagerService:
     """In-process gRPC 'control plane' for contract tests: sends queued
-    requests to connected agents and records their responses."""
+    requests to connected agents and records their responses.
 
-    def __init__(self):
+    ``wire="proto"`` speaks the reference protobuf framing: queue
+    ManagerPacket dicts on ``to_send``; ``responses`` receives decoded
+    Result dicts {"req_id", "data"}; ``hello`` is the decoded Hello."""
+
+    def __init__(self, wire: str = "json"):
+        self.wire = wire
         self.to_send: "queue.Queue[dict]" = queue.Queue()
         self.responses: "queue.Queue[dict]" = queue.Queue()
         self.hello: Optional[dict] = None
@@ -140,6 +188,24 @@ class FakeManagerService:
     def handler(self, request_iterator, context):
         def reader():
             for raw in request_iterator:
+                if self.wire == "proto":
+                    from . import protowire as pw
+
+                    pkt = pw.decode_message("AgentPacket", raw)
+                    if "hello" in pkt:
+                        self.hello = pkt["hello"]
+                        self._hello_evt.set()
+                    elif "result" in pkt:
+                        res = pkt["result"]
+                        self.responses.put(
+                            {
+                                "req_id": res.get("request_id", ""),
+                                "data": json.loads(
+                                    res.get("payload_json", b"null") or b"null"
+                                ),
+                            }
+                        )
+                    continue
                 frame = _deser(raw)
                 if "hello" in frame:
                     self.hello = frame["hello"]
@@ -156,16 +222,21 @@ class FakeManagerService:
                 continue
             if req is None:
                 return
-            yield _ser(req)
+            if self.wire == "proto":
+                from . import protowire as pw
+
+                yield pw.encode_message("ManagerPacket", req)
+            else:
+                yield _ser(req)
 
     def wait_hello(self, timeout: float = 10.0) -> Optional[dict]:
         self._hello_evt.wait(timeout)
         return self.hello
 
 
-def serve_fake_manager(port: int = 0):
+def serve_fake_manager(port: int = 0, wire: str = "json"):
     """Returns (server, service, bound_port)."""
-    service = FakeManagerService()
+    service = FakeManagerService(wire=wire)
 
     method_handlers = {
         "Connect": grpc.stream_stream_rpc_method_handler(
@@ -175,7 +246,9 @@ def serve_fake_manager(port: int = 0):
         )
     }
     generic = grpc.method_handlers_generic_handler(
-        "gpud.v2.SessionService", method_handlers
+        "gpud.session.v2.SessionService" if wire == "proto"
+        else "gpud.v2.SessionService",
+        method_handlers,
     )
     server = grpc.server(
         __import__("concurrent.futures", fromlist=["ThreadPoolExecutor"])

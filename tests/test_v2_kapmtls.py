@@ -41,6 +41,7 @@ def test_v2_grpc_roundtrip(mock_core):
             endpoint=f"127.0.0.1:{port}",
             machine_id="machine-7",
             token="tok-7",
+            wire="json",
         )
         agent.start()
         hello = service.wait_hello(10)
@@ -134,3 +135,90 @@ def test_kapmtls_grpc_credentials(tmp_path):
     import grpc
 
     assert isinstance(creds, grpc.ChannelCredentials)
+
+
+def test_v2_protobuf_wire_contract(mock_core):
+    """Full protobuf-framed contract: the agent speaks the reference's
+    session.proto wire format over /gpud.session.v2.SessionService/Connect —
+    Hello, hello_ack, request oneofs, Result{request_id, payload_json}."""
+    from gpud_amd import __version__
+    from gpud_amd.session import Session
+    from gpud_amd.session.v2 import V2Session, serve_fake_manager
+
+    server, service, port = serve_fake_manager(wire="proto")
+    try:
+        dispatcher = Session(
+            mock_core,
+            endpoint="unused",
+            open_reader=lambda: iter(()),
+            send_response=lambda f: None,
+        )
+        agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}")
+        assert agent.wire == "proto"  # the default is the reference framing
+        agent.start()
+        hello = service.wait_hello(10)
+        assert hello["agent_version"] == __version__
+        assert hello["min_protocol_revision"] == 1
+
+        service.to_send.put(
+            {"hello_ack": {"protocol_revision": 1, "manager_instance_id": "m1"}}
+        )
+        service.to_send.put({"request_id": "p1", "get_health_states": {}})
+        resp = service.responses.get(timeout=10)
+        assert resp["req_id"] == "p1"
+        comps = [x["component"] for x in resp["data"]["states"]]
+        assert "cpu" in comps
+
+        service.to_send.put(
+            {
+                "request_id": "p2",
+                "update_config": {"values": {"poll_interval_seconds": "30"}},
+            }
+        )
+        resp = service.responses.get(timeout=10)
+        assert resp["req_id"] == "p2"
+        assert "poll_interval_seconds" in resp["data"]["applied"]
+
+        service.to_send.put(
+            {
+                "request_id": "p3",
+                "trigger_component": {"component_name": "cpu"},
+            }
+        )
+        resp = service.responses.get(timeout=10)
+        assert resp["req_id"] == "p3"
+        assert resp["data"]["states"][0]["health"] in ("Healthy", "Degraded")
+
+        service.to_send.put(
+            {
+                "request_id": "p4",
+                "set_plugin_specs": {
+                    "specs_present": True,
+                    "specs": [
+                        {
+                            "plugin_name": "wired",
+                            "plugin_type": "component",
+                            "run_mode": "manual",
+                            "health_state_plugin": {
+                                "steps": [
+                                    {
+                                        "name": "s",
+                                        "run_bash_script": {
+                                            "content_type": "plaintext",
+                                            "script": "echo from-proto",
+                                        },
+                                    }
+                                ]
+                            },
+                            "timeout_nanos": 30_000_000_000,
+                        }
+                    ],
+                },
+            }
+        )
+        resp = service.responses.get(timeout=10)
+        assert resp["req_id"] == "p4"
+        assert "custom-plugin-wired" in resp["data"]["registered"]
+        agent.stop()
+    finally:
+        server.stop(grace=None)
