@@ -78,6 +78,10 @@ def run(
         2, help="reboots tolerated before RAS errors escalate to hardware "
         "inspection (reference: xid-reboot-threshold)"
     ),
+    session_protocol: str = typer.Option(
+        "v1", help="control-plane session protocol: v1 (HTTP dual-stream) "
+        "or v2 (gRPC, reference protobuf wire format)"
+    ),
     poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
@@ -233,9 +237,32 @@ def run(
             machine_id = _md.get_value(core.db_ro, _md.KEY_MACHINE_ID)
         except Exception:
             pass
-        session = Session(
-            core, endpoint=endpoint, token=token, machine_id=machine_id
-        )
+        if session_protocol == "v2":
+            # gRPC bidi stream in the reference's protobuf wire format
+            # (reference: --session-protocol, cmd/gpud/run/command.go:156)
+            from ..session.v2 import V2Session
+
+            dispatcher = Session(
+                core,
+                endpoint=endpoint,
+                token=token,
+                machine_id=machine_id,
+                open_reader=lambda: iter(()),
+                send_response=lambda f: None,
+            )
+            grpc_endpoint = (
+                endpoint.replace("https://", "").replace("http://", "")
+            )
+            session = V2Session(
+                dispatcher,
+                endpoint=grpc_endpoint,
+                machine_id=machine_id,
+                token=token,
+            )
+        else:
+            session = Session(
+                core, endpoint=endpoint, token=token, machine_id=machine_id
+            )
         session.start()
 
     # package manager reconcile loops (reference: cmd/gpud/run:425-431)

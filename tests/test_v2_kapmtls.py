@@ -222,3 +222,49 @@ def test_v2_protobuf_wire_contract(mock_core):
         agent.stop()
     finally:
         server.stop(grace=None)
+
+
+def test_daemon_boot_with_v2_session(tmp_path):
+    """`gpud run --session-protocol v2` connects the protobuf-framed gRPC
+    session to a live fake manager (the reference's --session-protocol)."""
+    import os
+    import signal
+    import socket
+    import subprocess
+    import sys
+
+    from gpud_amd.session.v2 import serve_fake_manager
+
+    server, service, grpc_port = serve_fake_manager(wire="proto")
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    http_port = s.getsockname()[1]; s.close()
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = {**os.environ, "GPUD_AMDSMI_MOCK": "1", "PYTHONPATH": repo}
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run",
+            "--in-memory-db", "--address", f"127.0.0.1:{http_port}",
+            "--log-level", "warning",
+            "--endpoint", f"127.0.0.1:{grpc_port}",
+            "--session-protocol", "v2",
+        ],
+        cwd=repo, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        hello = service.wait_hello(30)
+        assert hello is not None, "agent never connected over v2"
+        assert hello["min_protocol_revision"] == 1
+        service.to_send.put({"request_id": "b1", "get_health_states": {}})
+        resp = service.responses.get(timeout=15)
+        assert resp["req_id"] == "b1"
+        assert any(x["component"] == "cpu" for x in resp["data"]["states"])
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        proc.wait(timeout=15)
+        server.stop(grace=None)
