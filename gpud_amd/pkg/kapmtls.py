@@ -34,8 +34,19 @@ class Manager:
 
     # -- operations ----------------------------------------------------------
 
-    def stage(self, cert_pem: bytes, key_pem: bytes, version: str = "") -> str:
-        """Write new credentials into a staged versioned directory."""
+    def stage(
+        self,
+        cert_pem: bytes,
+        key_pem: bytes,
+        version: str = "",
+        gateway_ca_pem: bytes = b"",
+        gateway_endpoint: str = "",
+        server_name: str = "",
+    ) -> str:
+        """Write new credentials into a staged versioned directory. The
+        optional gateway metadata (CA bundle, endpoint, TLS server name —
+        reference: UpdateKAPMTLSCredentialsRequest fields) travels with
+        the version so rollback restores the matching gateway config."""
         version = version or str(int(time.time() * 1000))
         d = self._versioned_dir(version)
         os.makedirs(d, exist_ok=True)
@@ -45,9 +56,30 @@ class Manager:
         with open(key_path, "wb") as f:
             f.write(key_pem)
         os.chmod(key_path, 0o600)
+        if gateway_ca_pem:
+            with open(os.path.join(d, "gateway-ca.crt"), "wb") as f:
+                f.write(gateway_ca_pem)
+        if gateway_endpoint or server_name:
+            import json as _json
+
+            with open(os.path.join(d, "gateway.json"), "w") as f:
+                _json.dump(
+                    {"endpoint": gateway_endpoint, "server_name": server_name},
+                    f,
+                )
         with open(os.path.join(self.base_dir, STAGED_DIR), "w") as f:
             f.write(version)
         return version
+
+    def gateway_info(self) -> Dict[str, str]:
+        """Gateway metadata of the ACTIVE version ({} when absent)."""
+        try:
+            import json as _json
+
+            with open(os.path.join(self.active_link, "gateway.json")) as f:
+                return _json.load(f)
+        except (OSError, ValueError):
+            return {}
 
     def staged_version(self) -> str:
         try:

@@ -552,7 +552,9 @@ def manager_packet_to_frame(data: bytes) -> Optional[Dict[str, Any]]:
             {
                 "cert": _b64.b64encode(p.get("certificate_pem", b"")).decode(),
                 "key": _b64.b64encode(p.get("private_key_pem", b"")).decode(),
+                "gateway_ca": _b64.b64encode(p.get("gateway_ca_pem", b"")).decode(),
                 "gateway_endpoint": p.get("gateway_endpoint", ""),
+                "server_name": p.get("server_name", ""),
             },
         )
     if "activate_kap_mtls" in pkt:

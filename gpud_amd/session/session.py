@@ -506,7 +506,14 @@ class Session:
         key = base64.b64decode(payload.get("key", ""))
         if not cert or not key:
             return {"error": "cert and key (base64) required"}
-        version = self._kapmtls().stage(cert, key, payload.get("version", ""))
+        version = self._kapmtls().stage(
+            cert,
+            key,
+            payload.get("version", ""),
+            gateway_ca_pem=base64.b64decode(payload.get("gateway_ca", "")),
+            gateway_endpoint=payload.get("gateway_endpoint", ""),
+            server_name=payload.get("server_name", ""),
+        )
         return {"staged_version": version}
 
     def _m_activateKAPMTLS(self, payload: dict) -> dict:

@@ -268,3 +268,20 @@ def test_daemon_boot_with_v2_session(tmp_path):
             pass
         proc.wait(timeout=15)
         server.stop(grace=None)
+
+
+def test_kapmtls_gateway_metadata_travels_with_version(tmp_path):
+    from gpud_amd.pkg.kapmtls import Manager
+
+    m = Manager(str(tmp_path / "kap"))
+    m.stage(
+        b"C1", b"K1", version="1",
+        gateway_ca_pem=b"CA1", gateway_endpoint="gw1:443", server_name="gw1",
+    )
+    assert m.activate() is None
+    assert m.gateway_info() == {"endpoint": "gw1:443", "server_name": "gw1"}
+    m.stage(b"C2", b"K2", version="2", gateway_endpoint="gw2:443")
+    assert m.activate() is None
+    assert m.gateway_info()["endpoint"] == "gw2:443"
+    assert m.rollback() is None
+    assert m.gateway_info()["endpoint"] == "gw1:443"
