@@ -523,15 +523,75 @@ class Session:
             return {"error": err}
         return {"active_version": mgr.active_version()}
 
+    # nodeCredentials: the control plane places identity files on the node
+    # (reference: pkg/session/node_credentials.go). The destination travels
+    # with the file, so the allow-list below is the entire safety boundary:
+    # without it this method is an arbitrary root file write. These trees
+    # hold node identity material and nothing that grants code execution.
+    NODE_CREDENTIAL_ALLOWED_PREFIXES = ["/var/lib/gpud/", "/etc/kubernetes/"]
+    DEFAULT_NODE_CREDENTIAL_MODE = 0o600
+
+    def _allowed_credential_prefixes(self) -> List[str]:
+        return list(self.NODE_CREDENTIAL_ALLOWED_PREFIXES)
+
+    def _validate_credential_file(self, f: dict) -> Optional[str]:
+        path = f.get("path", "")
+        if not path:
+            return "node credential path is required"
+        if not os.path.isabs(path):
+            return f"node credential path {path!r} must be absolute"
+        # compared after normalization so '..' cannot escape an allowed tree
+        cleaned = os.path.normpath(path)
+        prefixes = self._allowed_credential_prefixes()
+        if not any(cleaned.startswith(p) for p in prefixes):
+            return (
+                f"node credential path {path!r} is outside "
+                + ", ".join(prefixes)
+            )
+        if not f.get("contents"):
+            return f"node credential {path!r} has no contents"
+        return None
+
+    def _write_credential_file(self, f: dict) -> Optional[str]:
+        """Atomic publish: staged in the destination directory with the
+        final mode, then renamed — a reader sees old or new contents, never
+        a partial write, and the material is never briefly world-readable."""
+        path = os.path.normpath(f["path"])
+        contents = base64.b64decode(f["contents"])
+        mode = int(f.get("mode") or 0) or self.DEFAULT_NODE_CREDENTIAL_MODE
+        try:
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            staged = path + ".gpud-staged"
+            fd = os.open(staged, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, mode)
+            try:
+                os.write(fd, contents)
+            finally:
+                os.close(fd)
+            os.chmod(staged, mode)
+            os.replace(staged, path)
+        except OSError as e:
+            return f"writing node credential {path!r}: {e}"
+        return None
+
     def _m_nodeCredentials(self, payload: dict) -> dict:
-        if self.core.db_ro is None:
-            return {"error": "no metadata store"}
-        return {
-            "machine_id": metadata.get_value(
-                self.core.db_ro, metadata.KEY_MACHINE_ID
-            ),
-            "endpoint": metadata.get_value(self.core.db_ro, metadata.KEY_ENDPOINT),
-            "has_token": bool(
-                metadata.get_value(self.core.db_ro, metadata.KEY_TOKEN)
-            ),
-        }
+        kubelet = payload.get("kubelet") or {}
+        files = [
+            f
+            for f in (kubelet.get("config"), kubelet.get("client_certificate"))
+            if f
+        ]
+        if not files:
+            return {"error": "node credentials contain no files"}
+        # validate everything before writing anything: one bad path leaves
+        # the node exactly as it was, not half-updated
+        for f in files:
+            err = self._validate_credential_file(f)
+            if err:
+                return {"error": err}
+        written = []
+        for f in files:
+            err = self._write_credential_file(f)
+            if err:
+                return {"error": err, "written": written}
+            written.append(os.path.normpath(f["path"]))
+        return {"written": written}

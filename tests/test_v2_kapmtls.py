@@ -107,7 +107,7 @@ def test_kapmtls_session_methods(mock_core):
     resp = s.process_request({"req_id": "k3", "method": "kapMTLSStatus", "data": {}})
     assert resp["data"]["active_version"] == "5"
     resp = s.process_request({"req_id": "k4", "method": "nodeCredentials", "data": {}})
-    assert "machine_id" in resp["data"]
+    assert "no files" in resp["data"]["error"]  # reference semantics
 
 
 def test_kapmtls_grpc_credentials(tmp_path):
@@ -285,3 +285,53 @@ def test_kapmtls_gateway_metadata_travels_with_version(tmp_path):
     assert m.gateway_info()["endpoint"] == "gw2:443"
     assert m.rollback() is None
     assert m.gateway_info()["endpoint"] == "gw1:443"
+
+
+def test_node_credentials_placement(mock_core, tmp_path):
+    """Reference pkg/session/node_credentials.go: allow-listed paths,
+    validate-all-before-write-any, atomic publish with owner-only mode."""
+    import base64 as b64
+    import os
+
+    from gpud_amd.session import Session
+
+    s = Session(
+        mock_core,
+        endpoint="unused",
+        open_reader=lambda: iter(()),
+        send_response=lambda f: None,
+    )
+    allowed = str(tmp_path / "allowed") + "/"
+    s.NODE_CREDENTIAL_ALLOWED_PREFIXES = [allowed]
+
+    def req(files):
+        return s.process_request(
+            {"req_id": "n", "method": "nodeCredentials",
+             "data": {"kubelet": files}}
+        )["data"]
+
+    enc = lambda b: b64.b64encode(b).decode()  # noqa: E731
+    # outside the allow-list: refused, nothing written
+    out = req({"config": {"path": "/etc/passwd", "contents": enc(b"x")}})
+    assert "outside" in out["error"]
+    # '..' escape is normalized before the prefix check
+    sneaky = allowed + "../escape"
+    out = req({"config": {"path": sneaky, "contents": enc(b"x")}})
+    assert "outside" in out["error"]
+    # empty contents refused (a truncated credential must not look written)
+    out = req({"config": {"path": allowed + "kubeconfig", "contents": ""}})
+    assert "no contents" in out["error"]
+    # one bad file in a pair: NOTHING is written
+    good = {"path": allowed + "kubeconfig", "contents": enc(b"KC")}
+    bad = {"path": "/etc/shadow", "contents": enc(b"x")}
+    out = req({"config": good, "client_certificate": bad})
+    assert "outside" in out["error"]
+    assert not os.path.exists(good["path"])
+    # happy path: both written, default mode 0600
+    cert = {"path": allowed + "pki/kubelet.pem", "contents": enc(b"CERT"),
+            "mode": 0o640}
+    out = req({"config": good, "client_certificate": cert})
+    assert out["written"] == [good["path"], cert["path"]]
+    assert open(good["path"], "rb").read() == b"KC"
+    assert oct(os.stat(good["path"]).st_mode & 0o777) == "0o600"
+    assert oct(os.stat(cert["path"]).st_mode & 0o777) == "0o640"
