@@ -484,9 +484,28 @@ def diagnose(
 def status(
     server_url: str = typer.Option("https://localhost:15132"),
     watch: float = typer.Option(0.0, help="re-poll every N seconds"),
+    data_dir: str = typer.Option(
+        DEFAULT_DATA_DIR, help="state dir (for the session-login history line)"
+    ),
 ):
     """Query a running daemon's health states (reference: gpud status)."""
     from ..client import Client
+
+    # control-plane session history (reference: pkg/session/states)
+    try:
+        from ..pkg import session_states as _ss
+        from ..pkg.sqlite_util import open_ro as _open_ro
+
+        conn = _open_ro(Config(data_dir=data_dir).state_path)
+        last = _ss.read_last(conn)
+        conn.close()
+        if last is not None:
+            mark = "ok" if last.success else "FAILED"
+            typer.echo(
+                f"control-plane session: last login {mark} ({last.message})"
+            )
+    except Exception:  # noqa: BLE001 — no state file yet is fine
+        pass
 
     c = Client(server_url)
     if not c.wait_healthz(timeout=5):

@@ -19,6 +19,22 @@ from .log import logger
 from .sqlite_util import open_rw
 
 
+def _record_state(cfg: Config, success: bool, message: str) -> None:
+    """Track the outcome in the session_states table (reference:
+    pkg/session/states — last 10 outcomes, queryable by gpud status)."""
+    try:
+        from . import session_states
+
+        conn = open_rw(cfg.state_path)
+        try:
+            session_states.create_table(conn)
+            session_states.insert(conn, success, message)
+        finally:
+            conn.close()
+    except Exception:  # noqa: BLE001 — history is best-effort
+        logger.debug("session state recording failed", exc_info=True)
+
+
 def do_login(
     cfg: Config,
     token: str,
@@ -43,9 +59,12 @@ def do_login(
             verify=False,
         )
     except httpx.HTTPError as e:
+        _record_state(cfg, False, f"login request failed: {e}")
         return f"login request failed: {e}"
     if r.status_code != 200:
-        return f"login rejected: HTTP {r.status_code} {r.text[:200]}"
+        msg = f"login rejected: HTTP {r.status_code} {r.text[:200]}"
+        _record_state(cfg, False, msg)
+        return msg
     try:
         body = r.json()
     except json.JSONDecodeError:
@@ -65,5 +84,6 @@ def do_login(
         metadata.set_value(conn, metadata.KEY_LOGIN_SUCCESS, str(int(time.time())))
     finally:
         conn.close()
+    _record_state(cfg, True, "login ok")
     logger.info("login succeeded for machine %s", machine_id)
     return None

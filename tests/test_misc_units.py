@@ -449,3 +449,42 @@ def test_run_bash_env_passthrough():
 
     res = run_bash("echo $GPUD_TEST_VAR", env={"GPUD_TEST_VAR": "hello42"})
     assert "hello42" in res.output
+
+
+def test_session_states_table(tmp_path):
+    """Reference pkg/session/states: last-10 retention, read_last,
+    has_any_failures; login outcomes are recorded."""
+    from gpud_amd.pkg import session_states as ss
+    from gpud_amd.pkg.sqlite_util import open_memory_pair
+
+    rw, ro = open_memory_pair()
+    ss.create_table(rw)
+    assert ss.read_last(ro) is None
+    assert ss.has_any_failures(ro) is False
+    for i in range(15):
+        ss.insert(rw, success=(i != 7), message=f"m{i}", timestamp=1000 + i)
+    rows = ss.read_all(ro)
+    assert len(rows) == 10  # trimmed to the most recent 10
+    assert rows[0].timestamp == 1014 and rows[0].success
+    assert ss.read_last(ro).message == "m14"
+    assert ss.has_any_failures(ro) is True  # i=7 failure survived the trim? 
+    # (1007 is within the last 10 of 1005..1014)
+
+
+def test_login_records_session_state(tmp_path):
+    from gpud_amd.pkg import session_states as ss
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.login import do_login
+    from gpud_amd.pkg.sqlite_util import open_ro
+
+    import os
+
+    cfg = Config(data_dir=str(tmp_path))
+    os.makedirs(cfg.data_dir, exist_ok=True)
+    err = do_login(cfg, token="x", endpoint="http://127.0.0.1:1", timeout=1.5)
+    assert err is not None
+    conn = open_ro(cfg.state_path)
+    last = ss.read_last(conn)
+    assert last is not None and last.success is False
+    assert "failed" in last.message
+    conn.close()
