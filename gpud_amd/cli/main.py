@@ -79,8 +79,9 @@ def run(
         "inspection (reference: xid-reboot-threshold)"
     ),
     session_protocol: str = typer.Option(
-        "v1", help="control-plane session protocol: v1 (HTTP dual-stream) "
-        "or v2 (gRPC, reference protobuf wire format)"
+        "auto", help="control-plane session protocol: v1 (HTTP dual-stream), "
+        "v2 (gRPC, reference protobuf wire format), or auto (v2 with "
+        "fallback to v1 when the manager reports it unsupported)"
     ),
     poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
@@ -237,7 +238,7 @@ def run(
             machine_id = _md.get_value(core.db_ro, _md.KEY_MACHINE_ID)
         except Exception:
             pass
-        if session_protocol == "v2":
+        if session_protocol in ("v2", "auto"):
             # gRPC bidi stream in the reference's protobuf wire format
             # (reference: --session-protocol, cmd/gpud/run/command.go:156)
             from ..session.v2 import V2Session
@@ -253,11 +254,24 @@ def run(
             grpc_endpoint = (
                 endpoint.replace("https://", "").replace("http://", "")
             )
+            fallback = None
+            if session_protocol == "auto":
+                # reference ProtocolAuto: permanent fallback to the legacy
+                # session on UNIMPLEMENTED (session_keepalive.go:15)
+                def fallback(
+                    _core=core, _ep=endpoint, _tok=token, _mid=machine_id
+                ):
+                    legacy = Session(
+                        _core, endpoint=_ep, token=_tok, machine_id=_mid
+                    )
+                    legacy.start()
+
             session = V2Session(
                 dispatcher,
                 endpoint=grpc_endpoint,
                 machine_id=machine_id,
                 token=token,
+                on_unsupported=fallback,
             )
         else:
             session = Session(

@@ -54,6 +54,7 @@ class V2Session:
         token: str = "",
         credentials: Optional[grpc.ChannelCredentials] = None,
         wire: str = "proto",
+        on_unsupported=None,
     ):
         self.dispatcher = dispatcher
         self.endpoint = endpoint
@@ -61,6 +62,10 @@ class V2Session:
         self.token = token
         self.credentials = credentials
         self.wire = wire
+        # "auto" protocol support (reference: session_keepalive.go:15 —
+        # ProtocolAuto tries v2 and falls back to the legacy session when
+        # the manager reports it unsupported): called once on UNIMPLEMENTED
+        self.on_unsupported = on_unsupported
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._outbox: "queue.Queue[Optional[dict]]" = queue.Queue()
@@ -151,6 +156,19 @@ class V2Session:
                 self._run_once()
                 backoff = 1.0
             except grpc.RpcError as e:
+                code = e.code() if hasattr(e, "code") else None
+                if (
+                    code == grpc.StatusCode.UNIMPLEMENTED
+                    and self.on_unsupported is not None
+                ):
+                    logger.info(
+                        "v2 session unsupported by the manager; "
+                        "falling back to the legacy session"
+                    )
+                    cb, self.on_unsupported = self.on_unsupported, None
+                    self._stop.set()
+                    cb()
+                    return
                 logger.warning("v2 session stream error: %s", e)
             except Exception:
                 logger.exception("v2 session failure")

@@ -335,3 +335,38 @@ def test_node_credentials_placement(mock_core, tmp_path):
     assert open(good["path"], "rb").read() == b"KC"
     assert oct(os.stat(good["path"]).st_mode & 0o777) == "0o600"
     assert oct(os.stat(cert["path"]).st_mode & 0o777) == "0o640"
+
+
+def test_v2_auto_fallback_on_unimplemented(mock_core):
+    """A gRPC server without the Connect method (UNIMPLEMENTED) triggers
+    the one-shot legacy fallback (reference ProtocolAuto)."""
+    import threading
+    from concurrent.futures import ThreadPoolExecutor
+
+    import grpc
+
+    from gpud_amd.session import Session
+    from gpud_amd.session.v2 import V2Session
+
+    # a real gRPC server with NO handlers -> UNIMPLEMENTED for Connect
+    server = grpc.server(ThreadPoolExecutor(max_workers=2))
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    fell_back = threading.Event()
+    try:
+        dispatcher = Session(
+            mock_core,
+            endpoint="unused",
+            open_reader=lambda: iter(()),
+            send_response=lambda f: None,
+        )
+        agent = V2Session(
+            dispatcher,
+            endpoint=f"127.0.0.1:{port}",
+            on_unsupported=fell_back.set,
+        )
+        agent.start()
+        assert fell_back.wait(15), "fallback was not triggered"
+        agent.stop()
+    finally:
+        server.stop(grace=None)
