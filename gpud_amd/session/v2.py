@@ -151,15 +151,22 @@ class V2Session:
 
     def _loop(self) -> None:
         backoff = 1.0
+        consecutive_failures = 0
         while not self._stop.is_set():
             try:
                 self._run_once()
                 backoff = 1.0
+                consecutive_failures = 0
             except grpc.RpcError as e:
                 code = e.code() if hasattr(e, "code") else None
-                if (
+                consecutive_failures += 1
+                # UNIMPLEMENTED = the manager definitively lacks v2; repeated
+                # UNAVAILABLE covers a v1-only HTTP endpoint that cannot even
+                # speak gRPC (the reference's infra serves both on one port,
+                # ours may not)
+                if self.on_unsupported is not None and (
                     code == grpc.StatusCode.UNIMPLEMENTED
-                    and self.on_unsupported is not None
+                    or consecutive_failures >= 3
                 ):
                     logger.info(
                         "v2 session unsupported by the manager; "
