@@ -155,3 +155,15 @@ def detect_provider(timeout: float = 1.0) -> Optional[str]:
         except Exception:
             continue
     return None
+
+
+def detect_provider_by_asn(public_ip: str) -> Optional[str]:
+    """ASN-based provider labeling for nodes without IMDS (reference:
+    machine_info.go:339 — AS org of the public IP, normalized). Needs
+    egress; returns None air-gapped."""
+    from . import asn
+
+    res = asn.get_as_lookup(public_ip, sleep=lambda _s: None)
+    if res is None or not res.asn_name:
+        return None
+    return asn.normalize_asn_name(res.asn_name)

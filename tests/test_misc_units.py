@@ -488,3 +488,42 @@ def test_login_records_session_state(tmp_path):
     assert last is not None and last.success is False
     assert "failed" in last.message
     conn.close()
+
+
+def test_asn_lookup_and_normalization(monkeypatch):
+    """Reference pkg/asn: retry, fallback, and AS-org normalization."""
+    from gpud_amd.pkg import asn
+
+    calls = {"n": 0}
+
+    def flaky_primary(ip):
+        calls["n"] += 1
+        if calls["n"] < 3:
+            raise ConnectionError("down")
+        return asn.ASLookup(asn="16509", asn_name="AMAZON-02, Inc.", ip=ip)
+
+    monkeypatch.setattr(asn, "lookup_primary", flaky_primary)
+    monkeypatch.setattr(asn, "lookup_fallback", None)
+    res = asn.get_as_lookup("1.2.3.4", sleep=lambda s: None)
+    assert res is not None and res.asn == "16509"
+    assert asn.normalize_asn_name(res.asn_name) == "amazon"
+
+    # primary dead, fallback answers
+    monkeypatch.setattr(
+        asn, "lookup_primary",
+        lambda ip: (_ for _ in ()).throw(ConnectionError("x")),
+    )
+    monkeypatch.setattr(
+        asn, "lookup_fallback",
+        lambda ip: asn.ASLookup(asn_name="Crusoe Energy Systems LLC"),
+    )
+    res = asn.get_as_lookup("1.2.3.4", sleep=lambda s: None)
+    assert asn.normalize_asn_name(res.asn_name) == "crusoe"
+
+    # everything dead -> None (air-gapped default)
+    monkeypatch.setattr(asn, "lookup_fallback", None)
+    assert asn.get_as_lookup("1.2.3.4", sleep=lambda s: None) is None
+
+    assert asn.normalize_asn_name("GOOGLE-CLOUD-PLATFORM") == "google"
+    assert asn.normalize_asn_name("Hetzner Online GmbH") == "hetzner"
+    assert asn.normalize_asn_name("Some University") == "some university"
