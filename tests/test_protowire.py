@@ -175,3 +175,75 @@ def test_get_events_timestamp_conversion():
     frame = pw.manager_packet_to_frame(b)
     assert frame["method"] == "events"
     assert frame["data"]["startTime"].startswith("2025-09-04T")
+
+
+def test_protowire_fuzz_roundtrip():
+    """Property fuzz: random dicts for every message round-trip through
+    encode/decode losslessly (hypothesis-driven)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    scalar = {
+        "uint32": st.integers(min_value=0, max_value=2**32 - 1),
+        "int64": st.integers(min_value=-(2**63), max_value=2**63 - 1),
+        "bool": st.booleans(),
+        "string": st.text(max_size=20),
+        "bytes": st.binary(max_size=20),
+    }
+
+    def msg_strategy(name, depth=0):
+        desc = pw.MESSAGES[name]
+        fields = {}
+        for _no, (fname, kind) in desc.items():
+            if kind in scalar:
+                fields[fname] = scalar[kind]
+            elif kind == "rep_string":
+                fields[fname] = st.lists(st.text(max_size=8), max_size=3)
+            elif kind.startswith("msg:") and depth < 2:
+                fields[fname] = msg_strategy(kind[4:], depth + 1)
+            elif kind.startswith("rep_msg:") and depth < 2:
+                fields[fname] = st.lists(
+                    msg_strategy(kind[8:], depth + 1), max_size=2
+                )
+            elif kind == "map_str_str":
+                fields[fname] = st.dictionaries(
+                    st.text(min_size=1, max_size=6), st.text(max_size=6),
+                    max_size=3,
+                )
+        return st.fixed_dictionaries({}, optional=fields)
+
+    def strip_empty(name, d):
+        """proto3 has no presence for empty scalars/maps/lists: drop values
+        that encode to nothing. Empty MESSAGE fields keep presence (a zero-
+        length LEN field is still emitted)."""
+        desc = {fn: kind for _no, (fn, kind) in pw.MESSAGES[name].items()}
+        out = {}
+        for k, v in d.items():
+            kind = desc[k]
+            if kind.startswith("msg:"):
+                out[k] = strip_empty(kind[4:], v)  # presence kept
+            elif kind.startswith("rep_msg:"):
+                v2 = [strip_empty(kind[8:], x) for x in v]
+                if v2:
+                    out[k] = v2
+            elif kind.startswith("map_") or kind == "rep_string":
+                if v:
+                    out[k] = v
+            else:
+                if v not in ("", 0, False, b""):
+                    out[k] = v
+        return out
+
+    for name in ("Hello", "Result", "UpdateConfigRequest", "PluginSpec",
+                 "ManagerPacket", "DiagnosticRequest"):
+        def make_check(_name):
+            @settings(max_examples=40, deadline=None)
+            @given(sample=msg_strategy(_name))
+            def check(sample):
+                b = pw.encode_message(_name, sample)
+                back = pw.decode_message(_name, b)
+                assert back == strip_empty(_name, sample), (_name, sample, back)
+
+            return check
+
+        make_check(name)()
