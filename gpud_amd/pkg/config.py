@@ -71,7 +71,7 @@ class Config:
     endpoint: str = ""
     token: str = ""
     machine_id: str = ""
-    session_protocol: str = "v1"
+    session_protocol: str = "auto"  # v1 | v2 | auto (reference protocol.go)
 
     @property
     def state_path(self) -> str:
