@@ -212,6 +212,28 @@ class FakeManagerService:
 
     def handler(self, request_iterator, context):
         def reader():
+            try:
+                self._read_stream(request_iterator)
+            except grpc.RpcError:
+                return  # agent hung up; normal at teardown
+
+        t = threading.Thread(target=reader, daemon=True)
+        t.start()
+        while context.is_active():
+            try:
+                req = self.to_send.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            if req is None:
+                return
+            if self.wire == "proto":
+                from . import protowire as pw
+
+                yield pw.encode_message("ManagerPacket", req)
+            else:
+                yield _ser(req)
+
+    def _read_stream(self, request_iterator):
             for raw in request_iterator:
                 if self.wire == "proto":
                     from . import protowire as pw
@@ -237,22 +259,6 @@ class FakeManagerService:
                     self._hello_evt.set()
                 else:
                     self.responses.put(frame)
-
-        t = threading.Thread(target=reader, daemon=True)
-        t.start()
-        while context.is_active():
-            try:
-                req = self.to_send.get(timeout=0.2)
-            except queue.Empty:
-                continue
-            if req is None:
-                return
-            if self.wire == "proto":
-                from . import protowire as pw
-
-                yield pw.encode_message("ManagerPacket", req)
-            else:
-                yield _ser(req)
 
     def wait_hello(self, timeout: float = 10.0) -> Optional[dict]:
         self._hello_evt.wait(timeout)
