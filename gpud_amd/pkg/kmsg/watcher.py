@@ -109,7 +109,8 @@ class Watcher:
     def _follow_loop(self) -> None:
         boot = boot_wall_time()
         fd = self._follow_fd
-        assert fd is not None
+        if fd is None:  # start() only spawns the loop after opening the fd
+            return
         while not self._stop.is_set():
             try:
                 data = os.read(fd, _READ_SIZE)
