@@ -479,3 +479,94 @@ def test_kmsg_writer_chunks_oversized(tmp_path):
     assert "".join(chunks) == big
     # build_line is the unit the micro-bench measures
     assert build_line(2, "m").startswith(b"<2>")
+
+
+def test_every_catalog_entry_matches_a_representative_line():
+    """Completeness: all 44 signatures match a realistic kernel line AND
+    first-match-wins ordering attributes each line to ITS entry (a broad
+    pattern earlier in the catalog must not shadow a specific one)."""
+    from gpud_amd.pkg.ras_catalog import CATALOG, INJECTABLE, match
+
+    representatives = dict(INJECTABLE)
+    representatives.update({
+        "amdgpu_gpu_reset_succeeded":
+            "amdgpu 0000:0a:00.0: amdgpu: GPU reset(5) succeeded!",
+        "amdgpu_mode2_reset":
+            "amdgpu 0000:0a:00.0: amdgpu: GPU mode2 reset",
+        "amdgpu_job_timeout":
+            "[drm:amdgpu_job_timedout [amdgpu]] *ERROR* Process information: "
+            "process python pid 4242",
+        "amdgpu_soft_recovery":
+            "amdgpu 0000:0a:00.0: amdgpu: ring gfx_0.0.0 soft recovery "
+            "succeeded",
+        "amdgpu_vm_fault":
+            "amdgpu 0000:0a:00.0: amdgpu: VM_L2_PROTECTION_FAULT_STATUS:"
+            "0x00000B33",
+        "amdgpu_ras_poison_consumption":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS poison consumption handler "
+            "invoked",
+        "amdgpu_ras_poison_creation":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS poison creation interrupt",
+        "amdgpu_ras_bad_page_threshold":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS records:256 exceed threshold:"
+            "256",
+        "amdgpu_ras_eeprom":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS EEPROM checksum mismatch",
+        "amdgpu_ras_event":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS event of type ue detected",
+        "amdgpu_thermal_shutdown":
+            "amdgpu 0000:0a:00.0: amdgpu: emergency thermal shutdown",
+        "kfd_evict_failed":
+            "kfd kfd: amdgpu: Failed to evict process queues",
+        "kfd_hws_hang":
+            "kfd kfd: amdgpu: CP hang detected, resetting",
+        "kfd_queue_preemption_failed":
+            "kfd kfd: amdgpu: queue preemption failed for queue 3",
+        "pcie_aer_fatal":
+            "pcieport 0000:00:01.1: AER: Uncorrected (Fatal) error received: "
+            "0000:0a:00.0",
+        "amdgpu_init_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: Fatal error during GPU init",
+        "amdgpu_ib_test_failed":
+            "[drm:amdgpu_ib_ring_tests [amdgpu]] *ERROR* IB test failed on "
+            "gfx_0.0.0 (-110).",
+        "amdgpu_firmware_load_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: failed to load firmware "
+            "amdgpu/gc_9_5_0_mec.bin",
+        "memory_edac_uncorrectable":
+            "EDAC MC0: 1 UE memory read error on CPU_SrcID#0",
+        "memory_edac_correctable":
+            "EDAC MC0: 1 CE memory scrubbing error on CPU_SrcID#0",
+        "host_mce":
+            "mce: [Hardware Error]: Machine check events logged",
+        "memory_oom_kill":
+            "Out of memory: Killed process 4242 (python) total-vm:1kB",
+        "memory_oom_cgroup":
+            "Memory cgroup out of memory: Killed process 4242 (python)",
+        "amdgpu_aca_error":
+            "amdgpu 0000:0a:00.0: amdgpu: ACA error bank 2 logged",
+        "amdgpu_psp_cmd_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: PSP load ras command failed",
+        "amdgpu_mes_error":
+            "amdgpu: MES failed to respond to msg=SET_HW_RES",
+        "amdgpu_fence_fallback":
+            "amdgpu 0000:0a:00.0: amdgpu: Fence fallback timer expired on "
+            "ring sdma0",
+        "pcie_aer_corrected":
+            "pcieport 0000:00:01.1: AER: Corrected error received: "
+            "0000:0a:00.0",
+        "pcie_bandwidth_limited":
+            "32.000 Gb/s available PCIe bandwidth, limited by 2.5 GT/s PCIe "
+            "x16 link at 0000:00:01.1",
+        "host_hung_task":
+            "INFO: task python:4242 blocked for more than 122 seconds.",
+    })
+    missing = [d.name for d in CATALOG if d.name not in representatives]
+    assert not missing, f"entries without representative lines: {missing}"
+    for d in CATALOG:
+        line = representatives[d.name]
+        res = match(line)
+        assert res is not None, f"{d.name}: no match for {line!r}"
+        assert res[0].name == d.name, (
+            f"{d.name}: line attributed to {res[0].name} — ordering shadow"
+        )
