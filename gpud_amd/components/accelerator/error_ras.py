@@ -67,6 +67,11 @@ class ErrorRASComponent(TickerComponent):
             getattr(cfg, "ras_reboot_threshold", DEFAULT_REBOOT_THRESHOLD)
             or DEFAULT_REBOOT_THRESHOLD
         )
+        # per-event-name overrides (reference: per-Xid thresholds,
+        # xid/threshold.go + cmd/gpud/run/command.go xid-thresholds)
+        self.event_thresholds = dict(
+            getattr(cfg, "ras_event_thresholds", None) or {}
+        )
         self.get_now: Callable = utcnow
 
     @property
@@ -167,7 +172,10 @@ class ErrorRASComponent(TickerComponent):
         )
         reboots_since = sum(1 for r in reboots if r.time > first_occurrence)
         actions = list(active_detail.repair_actions)
-        if reboots_since >= self.reboot_threshold:
+        threshold = int(
+            self.event_thresholds.get(active.name, self.reboot_threshold)
+        )
+        if reboots_since >= threshold:
             actions = [RepairActionType.HARDWARE_INSPECTION]
             desc = (
                 f"{active_detail.description} (recurred after {reboots_since} "

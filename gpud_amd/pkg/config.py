@@ -65,6 +65,9 @@ class Config:
     # error-ras escalation: reboots tolerated before HARDWARE_INSPECTION
     # (reference: xid-reboot-threshold flag, cmd/gpud/run/command.go)
     ras_reboot_threshold: int = 2
+    # per-event-name overrides (reference: xid-thresholds per-code map),
+    # e.g. {"amdgpu_ring_timeout": 1} escalates that class after 1 reboot
+    ras_event_thresholds: Dict[str, int] = field(default_factory=dict)
     zombie_degraded_threshold: int = 1000
     zombie_unhealthy_threshold: int = 2000
     # control plane

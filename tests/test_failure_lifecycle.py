@@ -122,3 +122,53 @@ def test_gpu_failure_lifecycle(core):
     # stage 6: after the board swap, the operator clears the state
     ras.set_healthy()
     assert ras.last_check_result().health == HealthStateType.HEALTHY
+
+
+def test_per_event_escalation_threshold(monkeypatch, tmp_path):
+    """Per-event-name reboot thresholds override the global one
+    (reference: per-Xid thresholds, xid/threshold.go)."""
+    import datetime
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.apiv1.types import Event, EventType, RepairActionType, utcnow
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    cfg = Config(
+        data_dir=str(tmp_path),
+        ras_event_thresholds={"amdgpu_ring_timeout": 1},
+    )
+    core = build_core(cfg, in_memory_db=True, kmsg_writable=False, record_reboot=False)
+    try:
+        comp = core.registry.get("accelerator-amd-error-ras")
+        bucket = core.event_store.bucket("accelerator-amd-error-ras")
+        now = utcnow()
+        # event first seen 2h ago, ONE reboot since; global threshold (2)
+        # would still suggest reboot, but the override escalates at 1
+        for dt in (120, 5):
+            bucket.insert(
+                Event(
+                    time=now - datetime.timedelta(minutes=dt),
+                    component=comp.name,
+                    name="amdgpu_ring_timeout",
+                    type=EventType.CRITICAL,
+                    message="ring gfx_0.0.0 timeout",
+                )
+            )
+        core.reboot_event_store._bucket.insert(
+            Event(
+                time=now - datetime.timedelta(minutes=60),
+                component="os",
+                name="reboot",
+                type=EventType.WARNING,
+                message="reboot detected",
+            )
+        )
+        comp.get_now = lambda: now
+        cr = comp.trigger_check()
+        assert cr.health == "Unhealthy"
+        assert cr.suggested_actions.repair_actions == [
+            RepairActionType.HARDWARE_INSPECTION
+        ], cr.suggested_actions
+    finally:
+        core.close()
