@@ -83,6 +83,10 @@ def run(
         "v2 (gRPC, reference protobuf wire format), or auto (v2 with "
         "fallback to v1 when the manager reports it unsupported)"
     ),
+    ras_event_thresholds: str = typer.Option(
+        "", help='per-event escalation overrides as JSON, e.g. '
+        '\'{"amdgpu_ring_timeout": 1}\' (reference: xid-thresholds)'
+    ),
     poll_interval_seconds: float = typer.Option(60.0),
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
@@ -156,6 +160,15 @@ def run(
         ]
     cfg.temperature_margin_threshold_c = temperature_margin_celsius
     cfg.ras_reboot_threshold = ras_reboot_threshold
+    if ras_event_thresholds:
+        try:
+            cfg.ras_event_thresholds = {
+                str(k): int(v)
+                for k, v in json.loads(ras_event_thresholds).items()
+            }
+        except (ValueError, AttributeError) as e:
+            typer.echo(f"invalid --ras-event-thresholds: {e}", err=True)
+            raise typer.Exit(code=1)
     cfg.expected_ib_ports = expected_ib_ports
     cfg.expected_ib_rate_gbps = expected_ib_rate_gbps
     cfg.poll_interval_seconds = poll_interval_seconds
