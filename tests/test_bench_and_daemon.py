@@ -45,6 +45,13 @@ def test_bench_single_rank_mock():
     assert d["steps"] == 20 and d["warmup"] == 2
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["data"] == "synthetic"
+    # the full driver-contract key set (the round-end harness parses these)
+    required = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"}
+    assert required <= set(d), required - set(d)
+    assert d["vs_baseline"] is None  # the reference publishes no numbers
+    assert isinstance(d["config"], dict) and d["config"].get("model")
 
 
 def test_bench_two_ranks_gloo_mock():
