@@ -196,6 +196,13 @@ def parse_duration(v: Any) -> float:
 def load_specs(path: str) -> List[Spec]:
     with open(path) as f:
         raw = yaml.safe_load(f) or []
+    if not isinstance(raw, list):
+        raise ValueError(
+            f"plugin specs file must be a YAML list, got {type(raw).__name__}"
+        )
+    for i, d in enumerate(raw):
+        if not isinstance(d, dict):
+            raise ValueError(f"plugin spec #{i} is not a mapping")
     specs = [Spec.from_dict(d) for d in raw]
     names = set()
     for s in specs:

@@ -128,3 +128,30 @@ def test_eventstore_returns_desc_order(mem_db, offsets):
     # table accumulates across examples — purge everything)
     bucket.purge(int((base + datetime.timedelta(days=5)).timestamp()))
     store.close()
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    doc=st.recursive(
+        st.one_of(st.none(), st.booleans(), st.integers(), st.text(max_size=12)),
+        lambda children: st.one_of(
+            st.lists(children, max_size=4),
+            st.dictionaries(st.text(max_size=8), children, max_size=4),
+        ),
+        max_leaves=12,
+    )
+)
+def test_plugin_spec_loader_never_crashes(tmp_path_factory, doc):
+    """Arbitrary YAML documents either load or raise ValueError/TypeError —
+    never an unhandled crash (the specs file is operator-supplied)."""
+    import yaml as _yaml
+
+    from gpud_amd.pkg.custom_plugins import load_specs
+
+    p = tmp_path_factory.mktemp("specs") / "s.yaml"
+    p.write_text(_yaml.safe_dump(doc))
+    try:
+        specs = load_specs(str(p))
+        assert isinstance(specs, list)
+    except ValueError:
+        pass  # clean rejection is fine; anything else propagates and fails
