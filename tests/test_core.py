@@ -260,3 +260,32 @@ def test_recorder_self_telemetry(mem_db):
     assert "gpud_component_check_duration_seconds_count" in names
     comp_labels = {m.component for m in scraped}
     assert "gpud" in comp_labels and "cpu" in comp_labels
+
+
+def test_eventstore_find_by_name_since_bounds(mem_db):
+    """Name filter + since bound are both honored, and the returned order
+    is newest-first (the API contract /v1/events relies on)."""
+    import datetime
+
+    from gpud_amd.apiv1.types import Event, utcnow
+    from gpud_amd.pkg.eventstore import Store
+
+    rw, ro = mem_db
+    store = Store(rw, ro)
+    b = store.bucket("bounds", disable_purge=True)
+    now = utcnow()
+    for mins, name in [(50, "a"), (40, "b"), (30, "a"), (20, "b"), (10, "a")]:
+        b.insert(
+            Event(
+                time=now - datetime.timedelta(minutes=mins),
+                name=name,
+                type="Info",
+                message=f"{name}@{mins}",
+            )
+        )
+    got = b.find_by_name_since(name="a", since=now - datetime.timedelta(minutes=35))
+    assert [e.message for e in got] == ["a@10", "a@30"]
+    got = b.find_by_name_since(name="b", since=now - datetime.timedelta(minutes=999))
+    assert [e.message for e in got] == ["b@20", "b@40"]
+    assert b.find_by_name_since(name="zzz", since=now - datetime.timedelta(days=1)) == []
+    store.close()
