@@ -601,3 +601,27 @@ def test_bad_envs_ttl_and_live_daemon_scope(mock_core, tmp_path, monkeypatch):
     assert "daemon" in cr.reason
     monkeypatch.delenv("AMD_SERIALIZE_KERNEL")
     assert comp.trigger_check().health == base.health
+
+
+def test_diag_components_report_unhealthy_without_extension(monkeypatch, tmp_path):
+    """On a GPU host (/dev/kfd present) with the HIP extension missing the
+    diag components must be UNHEALTHY, never a silent eager fallback (the
+    round-end 'native code not loaded' check)."""
+    from gpud_amd.components.accelerator import diag as diag_mod
+
+    def broken_load():
+        raise ImportError("extension not built")
+
+    monkeypatch.setattr(diag_mod, "_load_diag", broken_load)
+    monkeypatch.setattr(diag_mod, "gpu_present", lambda: True)
+    from gpud_amd.components.base import GPUdInstance
+
+    comps = [f(GPUdInstance()) for f in diag_mod.init_funcs()]
+    assert comps
+    for comp in comps:
+        cr = comp.trigger_check()
+        assert cr.health == "Unhealthy", (comp.name, cr.reason)
+        blob = (cr.reason + cr.error).lower()
+        # kernels: missing extension is named; fabric: the real binary runs
+        # and reports the missing device — loud either way
+        assert "extension" in blob or "not built" in blob or "fabric" in blob
