@@ -280,3 +280,65 @@ def test_daemon_boot_with_smi_injection_flags():
         except ProcessLookupError:
             pass
         proc.wait(timeout=15)
+
+
+def test_daemon_plugin_churn_and_deregister(tmp_path):
+    """Plugin registered from a specs file at boot serves trigger-checks,
+    deregisters via DELETE /v1/components, and the daemon keeps serving."""
+    import textwrap
+
+    import httpx
+
+    specs = tmp_path / "specs.yaml"
+    specs.write_text(textwrap.dedent("""
+    - plugin_name: churn
+      plugin_type: component
+      run_mode: manual
+      health_state_plugin:
+        steps:
+          - run_bash_script: {script: "echo ok"}
+    """))
+    port = _free_port()
+    env = {**os.environ, "GPUD_AMDSMI_MOCK": "1", "PYTHONPATH": REPO}
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run", "--in-memory-db",
+            "--address", f"127.0.0.1:{port}", "--log-level", "warning",
+            "--plugin-specs-file", str(specs),
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        c = Client(f"https://127.0.0.1:{port}")
+        assert c.wait_healthz(30)
+        h = httpx.Client(verify=False, timeout=15)
+        base = f"https://127.0.0.1:{port}"
+        for _ in range(5):
+            r = h.get(
+                base + "/v1/components/trigger-check",
+                params={"componentName": "custom-plugin-churn"},
+            )
+            assert r.status_code == 200, r.text
+            assert r.json()["states"][0]["health"] == "Healthy"
+        r = h.delete(
+            base + "/v1/components",
+            params={"componentName": "custom-plugin-churn"},
+        )
+        assert r.status_code == 200, r.text
+        assert "custom-plugin-churn" not in c.get_components()
+        # built-in components cannot be deregistered
+        r = h.delete(base + "/v1/components", params={"componentName": "cpu"})
+        assert r.status_code in (400, 403), r.text
+        assert "cpu" in c.get_components()
+        h.close()
+        c.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        proc.wait(timeout=15)
