@@ -281,3 +281,24 @@ def test_ras_reboot_threshold_reaches_component(monkeypatch, tmp_path):
         assert comp.reboot_threshold == 5
     finally:
         core.close()
+
+
+def test_version_and_help_render():
+    from typer.testing import CliRunner
+
+    from gpud_amd import __version__
+    from gpud_amd.cli.main import app
+
+    r = CliRunner().invoke(app, ["version"])
+    assert r.exit_code == 0 and __version__ in r.output
+    r = CliRunner().invoke(app, ["--help"])
+    assert r.exit_code == 0
+    for cmd in ("run", "scan", "diagnose", "bundle", "custom-plugins",
+                "inject-fault", "update-check", "up", "down", "release"):
+        assert cmd in r.output, cmd
+    r = CliRunner().invoke(app, ["run", "--help"])
+    assert r.exit_code == 0
+    flat = "".join(r.output.split())  # rich wraps long flags across lines
+    for flag in ("--gpu-uuids-with-gpu-lost", "--session-protocol",
+                 "--expected-compute-partition", "--ras-event-thresholds"):
+        assert flag in flat, flag
