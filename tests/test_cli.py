@@ -296,7 +296,7 @@ def test_version_and_help_render():
     for cmd in ("run", "scan", "diagnose", "bundle", "custom-plugins",
                 "inject-fault", "update-check", "up", "down", "release"):
         assert cmd in r.output, cmd
-    r = CliRunner().invoke(app, ["run", "--help"])
+    r = CliRunner().invoke(app, ["run", "--help"], env={"COLUMNS": "250"})
     assert r.exit_code == 0
     flat = "".join(r.output.split())  # rich wraps long flags across lines
     for flag in ("--gpu-uuids-with-gpu-lost", "--session-protocol",
