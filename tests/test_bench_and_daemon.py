@@ -342,3 +342,33 @@ def test_daemon_plugin_churn_and_deregister(tmp_path):
         except ProcessLookupError:
             pass
         proc.wait(timeout=15)
+
+
+def test_bench_four_ranks_gloo_mock():
+    """4-rank weak-scaling contract: one JSON line from rank 0, n_gpus=4,
+    MAX-over-ranks aggregation (the driver's SCALE run shape)."""
+    port = _free_port()
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "4",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            "bench.py", "--mock", "--gpus", "4", "--steps", "8",
+            "--warmup", "2",
+        ],
+        capture_output=True,
+        text=True,
+        timeout=300,
+        cwd=REPO,
+        env={**os.environ, "MASTER_ADDR": "127.0.0.1"},
+    )
+    assert out.returncode == 0, out.stderr[-1500:]
+    json_lines = [
+        l for l in out.stdout.strip().splitlines() if l.startswith("{")
+    ]
+    assert len(json_lines) == 1, "exactly one JSON line from rank 0"
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 4
+    assert d["scaling"] == "weak"
+    assert d["config"]["gpus_per_rank"] == 1 or d["data"] == "synthetic"
+    assert d["value"] > 0
