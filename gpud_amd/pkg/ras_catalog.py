@@ -444,6 +444,39 @@ CATALOG: List[Detail] = [
         _APP,
     ),
     _d(
+        "amdgpu_ip_resume_failed",
+        r"amdgpu.*resume of IP block <[^>]+> failed",
+        "An IP block failed to resume after reset/suspend — the device is "
+        "in a partial state; reboot to reinitialize",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_ip_suspend_failed",
+        r"amdgpu.*suspend of IP block <[^>]+> failed",
+        "An IP block failed to suspend cleanly — the following resume is "
+        "suspect",
+        EventType.WARNING,
+        _REBOOT,
+    ),
+    _d(
+        "kfd_process_vm_failed",
+        r"(?:kfd|amdgpu).*Failed to create process VM",
+        "KFD could not create a process VM — compute process start "
+        "failure; often follows earlier VM faults or exhausted resources",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "host_swiotlb_full",
+        r"swiotlb buffer is full",
+        "SWIOTLB bounce-buffer exhaustion — DMA is being bounced (IOMMU/"
+        "mapping config); expect severe transfer slowdowns",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
         "host_soft_lockup",
         r"BUG: soft lockup - CPU#\d+ stuck",
         "CPU soft lockup — a kernel thread monopolized a CPU; node health "

@@ -560,6 +560,16 @@ def test_every_catalog_entry_matches_a_representative_line():
             "x16 link at 0000:00:01.1",
         "host_hung_task":
             "INFO: task python:4242 blocked for more than 122 seconds.",
+        "amdgpu_ip_resume_failed":
+            "[drm:amdgpu_device_ip_resume_phase2 [amdgpu]] *ERROR* amdgpu: "
+            "resume of IP block <sdma_v4_4_2> failed -110",
+        "amdgpu_ip_suspend_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: suspend of IP block <gfx_v9_4_3> "
+            "failed -22",
+        "kfd_process_vm_failed":
+            "kfd kfd: amdgpu: Failed to create process VM object",
+        "host_swiotlb_full":
+            "sdhci: swiotlb buffer is full (sz: 262144 bytes)",
     })
     missing = [d.name for d in CATALOG if d.name not in representatives]
     assert not missing, f"entries without representative lines: {missing}"
