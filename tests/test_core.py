@@ -289,3 +289,37 @@ def test_eventstore_find_by_name_since_bounds(mem_db):
     assert [e.message for e in got] == ["b@20", "b@40"]
     assert b.find_by_name_since(name="zzz", since=now - datetime.timedelta(days=1)) == []
     store.close()
+
+
+def test_metrics_store_component_and_since_filters(mem_db):
+    import datetime
+
+    from gpud_amd.pkg.metrics.scraper import ScrapedMetric
+    from gpud_amd.pkg.metrics.store import MetricsStore
+
+    rw, ro = mem_db
+    store = MetricsStore(rw, ro)
+    now = datetime.datetime.now(datetime.timezone.utc)
+    old = now - datetime.timedelta(hours=2)
+
+    def m(comp, name, value, ts):
+        return ScrapedMetric(
+            unix_ms=int(ts.timestamp() * 1000),
+            component=comp,
+            name=name,
+            labels={},
+            value=value,
+        )
+
+    store.record([m("cpu", "cpu_usage", 10.0, old),
+                  m("cpu", "cpu_usage", 20.0, now),
+                  m("memory", "memory_used", 5.0, now)])
+    everything = store.read()
+    assert set(everything) == {"cpu", "memory"}
+    only_cpu = store.read(components=["cpu"])
+    assert set(only_cpu) == {"cpu"} and len(only_cpu["cpu"]) == 2
+    recent = store.read(since=now - datetime.timedelta(minutes=30))
+    assert len(recent["cpu"]) == 1 and recent["cpu"][0].value == 20.0
+    both = store.read(since=now - datetime.timedelta(minutes=30),
+                      components=["memory"])
+    assert set(both) == {"memory"}
