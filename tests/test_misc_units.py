@@ -527,3 +527,23 @@ def test_asn_lookup_and_normalization(monkeypatch):
     assert asn.normalize_asn_name("GOOGLE-CLOUD-PLATFORM") == "google"
     assert asn.normalize_asn_name("Hetzner Online GmbH") == "hetzner"
     assert asn.normalize_asn_name("Some University") == "some university"
+
+
+def test_version_file_malformed_tolerated(tmp_path):
+    """Garbage in target_version must not crash the run-loop check."""
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.update import check_version_file
+
+    cfg = Config(data_dir=str(tmp_path))
+    import os
+
+    os.makedirs(cfg.data_dir, exist_ok=True)
+    for garbage in ("", "\n\n", "  v1.2.3  \n", "\x00\xff", "a" * 10_000):
+        with open(cfg.target_version_path, "w", errors="replace") as f:
+            f.write(garbage)
+        out = check_version_file(cfg)  # str or None, never an exception
+        assert out is None or isinstance(out, str)
+    # whitespace is trimmed
+    with open(cfg.target_version_path, "w") as f:
+        f.write("  9.9.9  \n")
+    assert check_version_file(cfg) == "9.9.9"
