@@ -580,3 +580,24 @@ def test_every_catalog_entry_matches_a_representative_line():
         assert res[0].name == d.name, (
             f"{d.name}: line attributed to {res[0].name} — ordering shadow"
         )
+
+
+def test_pstore_scanner_tolerates_garbage(tmp_path, mem_db):
+    """Unreadable/binary pstore records must not break the panic scan."""
+    from gpud_amd.pkg.pstore import Scanner
+
+    d = tmp_path / "pstore"
+    d.mkdir()
+    (d / "dmesg-ramoops-0").write_bytes(b"\x00\xff\xfe garbage \x80")
+    (d / "console-ramoops-0").write_text(
+        "Kernel panic - not syncing: Fatal exception\n"
+    )
+    (d / "unreadable").write_text("x")
+    (d / "unreadable").chmod(0o000)
+    rw, ro = mem_db
+    sc = Scanner(pstore_dir=str(d), db_rw=rw, db_ro=ro)
+    found = sc.scan()
+    assert any("panic" in (m or "").lower() for _n, m, _t in found) or any(
+        "Kernel panic" in str(x) for x in found
+    )
+    (d / "unreadable").chmod(0o644)
