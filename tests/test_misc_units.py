@@ -547,3 +547,22 @@ def test_version_file_malformed_tolerated(tmp_path):
     with open(cfg.target_version_path, "w") as f:
         f.write("  9.9.9  \n")
     assert check_version_file(cfg) == "9.9.9"
+
+
+def test_machine_info_wire_keys(monkeypatch):
+    """machine-info serializes the reference's camelCase wire keys
+    (api/v1 MachineInfo — control planes parse these)."""
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd import smi
+    from gpud_amd.pkg.machine_info import get_machine_info
+
+    d = get_machine_info(smi.new()).to_dict()
+    assert {"gpudVersion", "hostname", "bootID", "machineID",
+            "kernelVersion", "operatingSystem", "uptime", "gpuInfo",
+            "cpuInfo", "memoryInfo", "diskInfo", "nicInfo",
+            "gpuDriverVersion", "cudaVersion"} <= set(d)
+    gi = d["gpuInfo"]
+    assert gi["manufacturer"] == "AMD"
+    assert gi["architecture"].startswith("gfx950")
+    assert len(gi["gpus"]) == 8  # mock default
+    assert d["cudaVersion"]  # carries the ROCm version in the wire slot
