@@ -281,3 +281,21 @@ def test_trigger_tag_unknown_tag_empty(daemon):
     core, server, client = daemon
     out = client.trigger_check(tag="no-such-tag")
     assert out.get("components", []) == [] or out.get("states", []) == []
+
+
+def test_gzip_large_response(daemon):
+    """Accept-Encoding: gzip compresses the full states payload (the
+    GZip middleware the reference enables for /v1 routes)."""
+    import httpx
+
+    core, server, client = daemon
+    h = httpx.Client(verify=False, timeout=15)
+    r = h.get(
+        server.base_url + "/v1/states",
+        headers={"Accept-Encoding": "gzip"},
+    )
+    assert r.status_code == 200
+    # httpx transparently decompresses; the header proves the wire encoding
+    assert r.headers.get("content-encoding") == "gzip"
+    assert len(r.json()) > 10
+    h.close()
