@@ -299,3 +299,20 @@ def test_gzip_large_response(daemon):
     assert r.headers.get("content-encoding") == "gzip"
     assert len(r.json()) > 10
     h.close()
+
+
+def test_client_timeout_configurable(daemon):
+    """Client(timeout=) reaches the transport (operators tune this for
+    long-running trigger-tag diag requests)."""
+    from gpud_amd.client import Client
+
+    core, server, client = daemon
+    c = Client(server.base_url, timeout=0.001)
+    import httpx
+
+    with pytest.raises((httpx.TimeoutException, httpx.TransportError)):
+        c.get_health_states()
+    c.close()
+    c2 = Client(server.base_url, timeout=30)
+    assert c2.healthz()
+    c2.close()
