@@ -316,3 +316,16 @@ def test_client_timeout_configurable(daemon):
     c2 = Client(server.base_url, timeout=30)
     assert c2.healthz()
     c2.close()
+
+
+def test_states_subset_filter_exact(daemon):
+    """components= filter returns exactly the requested subset, nothing
+    else (the control plane relies on this for targeted polls)."""
+    core, server, client = daemon
+    comps = client.get_components()
+    subset = sorted(comps)[:5]
+    st = client.get_health_states(components=subset)
+    assert set(st) == set(s for s in subset if s in st)
+    assert set(st) <= set(subset)
+    for name, states in st.items():
+        assert states and states[0].health
