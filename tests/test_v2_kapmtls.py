@@ -370,3 +370,35 @@ def test_v2_auto_fallback_on_unimplemented(mock_core):
         agent.stop()
     finally:
         server.stop(grace=None)
+
+
+def test_v2_reconnects_after_manager_death(mock_core):
+    """A dead manager stream triggers backoff reconnect attempts, and the
+    agent thread survives to keep trying (reference keepAliveV2 loop)."""
+    import time
+
+    from gpud_amd.session import Session
+    from gpud_amd.session.v2 import V2Session, serve_fake_manager
+
+    server, service, port = serve_fake_manager(wire="proto")
+    dispatcher = Session(
+        mock_core,
+        endpoint="unused",
+        open_reader=lambda: iter(()),
+        send_response=lambda f: None,
+    )
+    agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}")
+    agent.start()
+    try:
+        assert service.wait_hello(15)
+        service.to_send.put({"request_id": "a", "get_health_states": {}})
+        assert service.responses.get(timeout=15)["req_id"] == "a"
+        server.stop(grace=None)
+        deadline = time.time() + 30
+        while agent.reconnects == 0 and time.time() < deadline:
+            time.sleep(0.2)
+        assert agent.reconnects >= 1
+        assert agent._thread.is_alive()
+    finally:
+        agent.stop()
+        server.stop(grace=None)
