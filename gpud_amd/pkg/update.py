@@ -66,6 +66,16 @@ def update_to_version(
         artifact = _fetch(f"{base_url}/{name}")
     except Exception as e:
         return f"download failed: {e}"
+    if root_pub is None:
+        # pinned root key: <dataDir>/root.pub makes verification mandatory
+        # (reference: release gen-key root key pinning; without a pin the
+        # update is trust-on-first-use like the reference's plain download)
+        pin_path = os.path.join(cfg.data_dir, "root.pub")
+        try:
+            with open(pin_path, "rb") as f:
+                root_pub = f.read().strip()
+        except OSError:
+            root_pub = None
     if root_pub is not None:
         try:
             sig = _fetch(f"{base_url}/{name}.sig")

@@ -674,3 +674,58 @@ def test_plugin_parser_extracts_embedded_json():
     cr = comp.trigger_check()
     assert cr.health == "Healthy", cr.reason
     assert cr.extra_info["result"] == "success"
+
+
+def test_update_pinned_root_key_enforced(tmp_path):
+    """A root.pub in the data dir makes signature verification MANDATORY:
+    unsigned/tampered releases are refused; a valid chain installs."""
+    import io
+    import os
+    import tarfile
+
+    from gpud_amd.pkg import distsign
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.update import update_to_version
+
+    buf = io.BytesIO()
+    with tarfile.open(fileobj=buf, mode="w:gz") as tf:
+        data = b"payload"
+        info = tarfile.TarInfo(name="gpud-amd/bin/gpud-amd")
+        info.size = len(data)
+        tf.addfile(info, io.BytesIO(data))
+    art = buf.getvalue()
+    (tmp_path / "gpud-amd_9.9.9.tar.gz").write_bytes(art)
+
+    cfg = Config(data_dir=str(tmp_path / "data"))
+    os.makedirs(cfg.data_dir, exist_ok=True)
+    root_seed, sign_seed = os.urandom(32), os.urandom(32)
+    _, root_pub = distsign.generate_keypair(root_seed)
+    _, sign_pub = distsign.generate_keypair(sign_seed)
+    (tmp_path / "data" / "root.pub").write_bytes(root_pub)
+
+    # pinned key + missing signature files => refused
+    err = update_to_version(
+        cfg, "9.9.9", base_url=f"file://{tmp_path}",
+        install_dir=str(tmp_path / "i1"),
+    )
+    assert err is not None and "signature" in err
+
+    # full valid chain => installs
+    name = "gpud-amd_9.9.9.tar.gz"
+    (tmp_path / f"{name}.sig").write_bytes(distsign.sign(art, sign_seed))
+    (tmp_path / f"{name}.pub").write_bytes(sign_pub)
+    (tmp_path / f"{name}.pub.sig").write_bytes(distsign.sign(sign_pub, root_seed))
+    err = update_to_version(
+        cfg, "9.9.9", base_url=f"file://{tmp_path}",
+        install_dir=str(tmp_path / "i2"),
+    )
+    assert err is None
+    assert (tmp_path / "i2" / "gpud-amd" / "bin" / "gpud-amd").exists()
+
+    # tampered artifact => refused even with the old signatures present
+    (tmp_path / name).write_bytes(art + b"x")
+    err = update_to_version(
+        cfg, "9.9.9", base_url=f"file://{tmp_path}",
+        install_dir=str(tmp_path / "i3"),
+    )
+    assert err is not None and "FAILED" in err
