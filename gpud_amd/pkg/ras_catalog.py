@@ -104,7 +104,10 @@ CATALOG: List[Detail] = [
     # ---- hangs / timeouts -------------------------------------------------
     _d(
         "amdgpu_ring_timeout",
-        r"amdgpu.*\*ERROR\* ring (?P<ring>\S+) timeout",
+        # drm-log form: "[drm:amdgpu_job_timedout [amdgpu]] *ERROR* ring X
+        # timeout"; newer kernels use dev_err: "amdgpu 0000:..: ring X
+        # timeout, signaled seq=.." — accept both
+        r"amdgpu.*?(?:\*ERROR\* )?ring (?P<ring>\S+) timeout",
         "amdgpu command-ring timeout — a GPU job exceeded the scheduler "
         "timeout; usually an application-level hang (infinite kernel, "
         "deadlocked wave), occasionally a hardware fault if recurring",
