@@ -73,7 +73,17 @@ def update_to_version(
         pin_path = os.path.join(cfg.data_dir, "root.pub")
         try:
             with open(pin_path, "rb") as f:
-                root_pub = f.read().strip()
+                raw = f.read()
+            # a raw 32-byte key is used verbatim (stripping would corrupt
+            # keys that happen to start/end with whitespace bytes); longer
+            # files are treated as hex text
+            if len(raw) == 32:
+                root_pub = raw
+            else:
+                try:
+                    root_pub = bytes.fromhex(raw.decode().strip())
+                except (ValueError, UnicodeDecodeError):
+                    root_pub = raw.strip() or None
         except OSError:
             root_pub = None
     if root_pub is not None:
