@@ -91,6 +91,15 @@ def run(
     plugin_specs_file: str = typer.Option(""),
     endpoint: str = typer.Option("", help="control-plane endpoint"),
     token: str = typer.Option("", help="control-plane token"),
+    control_plane_ca_file: str = typer.Option(
+        "", help="CA bundle to verify the control-plane endpoint against "
+        "(default: system CAs)"
+    ),
+    control_plane_insecure_tls: bool = typer.Option(
+        False, help="DANGEROUS: skip TLS verification towards the control "
+        "plane (lab use only; the session carries the auth token and "
+        "accepts bootstrap/update requests)"
+    ),
     in_memory_db: bool = typer.Option(False, help="ephemeral state (testing)"),
     no_tls: bool = typer.Option(False, help="serve plain HTTP (testing)"),
     log_level: str = typer.Option("info"),
@@ -143,6 +152,8 @@ def run(
         data_dir, address, expected_gpu_count, plugin_specs_file, endpoint
     )
     cfg.token = token
+    cfg.control_plane_ca_file = control_plane_ca_file
+    cfg.control_plane_insecure_tls = control_plane_insecure_tls
     cfg.expected_xgmi_link_count = expected_xgmi_link_count
     cfg.expected_compute_partition = expected_compute_partition
     cfg.expected_memory_partition = expected_memory_partition
@@ -279,12 +290,22 @@ def run(
                     )
                     legacy.start()
 
+            v2_creds = None
+            if cfg.control_plane_ca_file and not cfg.control_plane_insecure_tls:
+                import grpc as _grpc
+
+                with open(cfg.control_plane_ca_file, "rb") as _caf:
+                    v2_creds = _grpc.ssl_channel_credentials(
+                        root_certificates=_caf.read()
+                    )
             session = V2Session(
                 dispatcher,
                 endpoint=grpc_endpoint,
                 machine_id=machine_id,
                 token=token,
                 on_unsupported=fallback,
+                credentials=v2_creds,
+                insecure=cfg.control_plane_insecure_tls,
             )
         else:
             session = Session(
@@ -752,7 +773,7 @@ def notify(
             json={"machineID": machine_id, "event": event},
             headers={"token": token},
             timeout=10,
-            verify=False,
+            verify=cfg.control_plane_verify(),
         )
         typer.echo(f"notify {event}: HTTP {r.status_code}")
     except httpx.HTTPError as e:

@@ -75,6 +75,22 @@ class Config:
     token: str = ""
     machine_id: str = ""
     session_protocol: str = "auto"  # v1 | v2 | auto (reference protocol.go)
+    # TLS verification towards the control plane (reference:
+    # session_v2.go:278 sets MinVersion/ServerName and never skips
+    # verification). Verified by default; an operator can pin a private CA
+    # bundle, or explicitly opt into insecure mode for lab setups.
+    control_plane_insecure_tls: bool = False
+    control_plane_ca_file: str = ""
+
+    def control_plane_verify(self):
+        """httpx-style ``verify`` value for control-plane connections:
+        a CA bundle path when pinned, False only when explicitly opted
+        into insecure mode, True (system CAs) otherwise."""
+        if self.control_plane_insecure_tls:
+            return False
+        if self.control_plane_ca_file:
+            return self.control_plane_ca_file
+        return True
 
     @property
     def state_path(self) -> str:

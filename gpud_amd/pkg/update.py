@@ -51,6 +51,51 @@ def _fetch(url: str, timeout: float = 60.0) -> bytes:
     return r.content
 
 
+def _safe_extract(tf: tarfile.TarFile, target: str) -> Optional[str]:
+    """Extract a release archive, confining every member (and every link
+    target) to ``target``. Rejects absolute names, traversal via any
+    ``..`` segment (including ``a/../../x``), device/FIFO members, and
+    symlinks/hardlinks whose resolved destination escapes the install dir.
+    Uses tarfile's data filter as a second layer when available (backported
+    to this interpreter line). Returns an error string, or None."""
+    root = os.path.realpath(target)
+
+    def _confined(path: str) -> bool:
+        resolved = os.path.realpath(os.path.join(root, path))
+        return resolved == root or resolved.startswith(root + os.sep)
+
+    for member in tf.getmembers():
+        name = member.name
+        if name.startswith("/") or os.path.isabs(name):
+            return f"unsafe path in archive: {name}"
+        if ".." in name.split("/"):
+            return f"unsafe path in archive: {name}"
+        if not _confined(name):
+            return f"unsafe path in archive: {name}"
+        if member.isdev():
+            return f"device member in archive: {name}"
+        if member.issym() or member.islnk():
+            link = member.linkname
+            if os.path.isabs(link):
+                return f"unsafe link target in archive: {name} -> {link}"
+            # symlinks resolve relative to their containing directory;
+            # hardlink targets are relative to the archive root
+            if member.issym():
+                link = os.path.join(os.path.dirname(name), link)
+            resolved = os.path.normpath(link)
+            if ".." in resolved.split(os.sep) or not _confined(resolved):
+                return f"unsafe link target in archive: {name} -> {member.linkname}"
+    try:
+        tf.extractall(target, filter="data")
+    except TypeError:
+        # no extraction-filter support on this interpreter: the manual
+        # validation above already rejected every escaping member
+        tf.extractall(target)
+    except tarfile.FilterError as e:  # type: ignore[attr-defined]
+        return f"unsafe member in archive: {e}"
+    return None
+
+
 def update_to_version(
     cfg: Config,
     version: str,
@@ -69,11 +114,16 @@ def update_to_version(
     if root_pub is None:
         # pinned root key: <dataDir>/root.pub makes verification mandatory
         # (reference: release gen-key root key pinning; without a pin the
-        # update is trust-on-first-use like the reference's plain download)
+        # update is trust-on-first-use like the reference's plain download).
+        # Fails CLOSED: a present-but-unreadable or present-but-unparseable
+        # pin aborts the update instead of downgrading to unverified.
         pin_path = os.path.join(cfg.data_dir, "root.pub")
-        try:
-            with open(pin_path, "rb") as f:
-                raw = f.read()
+        if os.path.lexists(pin_path):
+            try:
+                with open(pin_path, "rb") as f:
+                    raw = f.read()
+            except OSError as e:
+                return f"pinned root key {pin_path} unreadable: {e}"
             # a raw 32-byte key is used verbatim (stripping would corrupt
             # keys that happen to start/end with whitespace bytes); longer
             # files are treated as hex text
@@ -83,9 +133,12 @@ def update_to_version(
                 try:
                     root_pub = bytes.fromhex(raw.decode().strip())
                 except (ValueError, UnicodeDecodeError):
-                    root_pub = raw.strip() or None
-        except OSError:
-            root_pub = None
+                    root_pub = raw.strip()
+            if not root_pub or len(root_pub) != 32:
+                return (
+                    f"pinned root key {pin_path} is not a 32-byte ed25519 "
+                    "key; refusing unverified update"
+                )
     if root_pub is not None:
         try:
             sig = _fetch(f"{base_url}/{name}.sig")
@@ -102,11 +155,9 @@ def update_to_version(
             tmp.write(artifact)
             tmp.flush()
             with tarfile.open(tmp.name, "r:gz") as tf:
-                for member in tf.getmembers():
-                    # refuse path traversal
-                    if member.name.startswith(("/", "..")):
-                        return f"unsafe path in archive: {member.name}"
-                tf.extractall(target)
+                err = _safe_extract(tf, target)
+                if err:
+                    return err
     except (tarfile.TarError, OSError) as e:
         return f"unpack failed: {e}"
     write_target_version(cfg, version)

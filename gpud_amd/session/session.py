@@ -50,6 +50,7 @@ class Session:
         send_response: Optional[Callable[[dict], None]] = None,
         sleep_fn: Callable[[float], None] = time.sleep,
         jitter_fn: Callable[[], float] = lambda: random.uniform(0.5, 1.5),
+        verify=None,
     ):
         self.core = core
         self.endpoint = endpoint.rstrip("/")
@@ -62,7 +63,14 @@ class Session:
         self._jitter = jitter_fn
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
-        self._client = httpx.Client(verify=False, timeout=None)
+        # verified TLS by default: this channel carries the auth token and
+        # accepts bootstrap/setPluginSpecs/update — a MITM must not be able
+        # to impersonate the control plane. Insecure mode exists only via
+        # the explicit config flag (Config.control_plane_verify()).
+        if verify is None:
+            cfg = getattr(core, "config", None)
+            verify = cfg.control_plane_verify() if cfg is not None else True
+        self._client = httpx.Client(verify=verify, timeout=None)
         self.reconnects = 0
 
     # -- transport -----------------------------------------------------------
@@ -436,14 +444,53 @@ class Session:
         res = run_bash(script, timeout_seconds=timeout)
         return {"exit_code": res.exit_code, "output": res.output[-4096:]}
 
+    # Config keys the control plane may update at runtime (reference
+    # restricts updateConfig to specific known keys —
+    # pkg/session/session.go:223-233). Command overrides (reboot_command
+    # etc.), data paths and TLS settings are deliberately NOT settable over
+    # the session: those would turn a config push into code execution or a
+    # downgrade of the channel's own security.
+    UPDATABLE_CONFIG_KEYS = frozenset(
+        {
+            "events_retention_days",
+            "metrics_retention_days",
+            "compact_period_hours",
+            "poll_interval_seconds",
+            "metrics_sync_interval_seconds",
+            "enabled_components",
+            "disabled_components",
+            "expected_gpu_count",
+            "expected_xgmi_link_count",
+            "expected_compute_partition",
+            "expected_memory_partition",
+            "expected_ib_ports",
+            "expected_ib_rate_gbps",
+            "latency_targets",
+            "temperature_margin_threshold_c",
+            "ras_reboot_threshold",
+            "ras_event_thresholds",
+            "zombie_degraded_threshold",
+            "zombie_unhealthy_threshold",
+            "nfs_host_root",
+            "kernel_modules_to_check",
+            "libraries_to_check",
+            "mount_points",
+        }
+    )
+
     def _m_updateConfig(self, payload: dict) -> dict:
-        applied = []
+        applied, rejected = [], []
         cfg = self.core.config
         for key, value in payload.items():
-            if hasattr(cfg, key):
+            if key in self.UPDATABLE_CONFIG_KEYS and hasattr(cfg, key):
                 setattr(cfg, key, value)
                 applied.append(key)
-        return {"applied": applied}
+            else:
+                rejected.append(key)
+        out = {"applied": applied}
+        if rejected:
+            out["rejected"] = rejected
+        return out
 
     def _m_updateToken(self, payload: dict) -> dict:
         token = payload.get("token", "")

@@ -55,11 +55,17 @@ class V2Session:
         credentials: Optional[grpc.ChannelCredentials] = None,
         wire: str = "proto",
         on_unsupported=None,
+        insecure: bool = False,
     ):
         self.dispatcher = dispatcher
         self.endpoint = endpoint
         self.machine_id = machine_id
         self.token = token
+        # TLS by default (reference: session_v2.go:278 — verified TLS, no
+        # InsecureSkipVerify); a plaintext channel needs the explicit
+        # insecure flag (tests, lab setups)
+        if credentials is None and not insecure:
+            credentials = grpc.ssl_channel_credentials()
         self.credentials = credentials
         self.wire = wire
         # "auto" protocol support (reference: session_keepalive.go:15 —

@@ -42,6 +42,7 @@ def test_v2_grpc_roundtrip(mock_core):
             machine_id="machine-7",
             token="tok-7",
             wire="json",
+            insecure=True,  # local plaintext test server
         )
         agent.start()
         hello = service.wait_hello(10)
@@ -153,7 +154,7 @@ def test_v2_protobuf_wire_contract(mock_core):
             open_reader=lambda: iter(()),
             send_response=lambda f: None,
         )
-        agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}")
+        agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}", insecure=True)
         assert agent.wire == "proto"  # the default is the reference framing
         agent.start()
         hello = service.wait_hello(10)
@@ -248,6 +249,8 @@ def test_daemon_boot_with_v2_session(tmp_path):
             "--log-level", "warning",
             "--endpoint", f"127.0.0.1:{grpc_port}",
             "--session-protocol", "v2",
+            # the fake manager is a local plaintext gRPC server
+            "--control-plane-insecure-tls",
         ],
         cwd=repo, env=env,
         stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
@@ -364,6 +367,7 @@ def test_v2_auto_fallback_on_unimplemented(mock_core):
             dispatcher,
             endpoint=f"127.0.0.1:{port}",
             on_unsupported=fell_back.set,
+            insecure=True,
         )
         agent.start()
         assert fell_back.wait(15), "fallback was not triggered"
@@ -387,7 +391,7 @@ def test_v2_reconnects_after_manager_death(mock_core):
         open_reader=lambda: iter(()),
         send_response=lambda f: None,
     )
-    agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}")
+    agent = V2Session(dispatcher, endpoint=f"127.0.0.1:{port}", insecure=True)
     agent.start()
     try:
         assert service.wait_hello(15)
