@@ -34,6 +34,7 @@
 #include <cstdint>
 #include <stdexcept>
 #include <string>
+#include <type_traits>
 #include <vector>
 
 namespace py = pybind11;
@@ -985,6 +986,136 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v3_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// GEMM v5: register-staged K-loop — NO LDS, NO barriers. Every wave loads
+// its own MFMA fragments straight from global memory (both operands are
+// K-contiguous, so a 16x16x32 fragment is 16 contiguous bytes per lane:
+// lane l reads row base+(l&15), k-block (l>>4)*8 — no transpose, no LDS
+// round trip), software-pipelined one quadrant / one B-tile ahead. With no
+// cross-wave LDS coupling there is no s_barrier and no workgroup-level
+// vmcnt drain anywhere in the K-loop — the structural stall that caps the
+// glds templates (cdna_hip_programming.md §5 "the step-3 structure's
+// ceiling": the path past the plain-HIP plateau is a K-loop whose prefetch
+// loads stay in flight across tile boundaries, waits counted per wave and
+// never drained to 0). Here hipcc itself emits the counted per-wave
+// s_waitcnt vmcnt(N) at each fragment's first consumer because every load
+// is an ordinary visible load (no glds in flight to force vmcnt(0) — §5
+// "Three .s-level traps" (b) in reverse).
+//
+// Cross-wave reuse moves from LDS to the cache hierarchy: the 4 waves of a
+// wave_m half re-read the same A rows through their CU's L1, and the 2
+// waves of a wave_n band share B likewise; unique per-CU traffic per
+// K-tile is the same 64 KiB a glds template stages. Register budget:
+// acc 128 + bfrag 2x32 + afrag 2x16 = 224 VGPRs -> 2 waves/SIMD at
+// __launch_bounds__(512, 2) (per-lane file 512; MI355X_MICROARCH.md
+// register table).
+// ---------------------------------------------------------------------------
+
+template <bool SETPRIO = true>
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v5_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;  // 0..1 -> rows [wave_m*128, +128)
+  const int wave_n = wave & 3;   // 0..3 -> cols [wave_n*64, +64)
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  // per-lane fragment base: row (lane&15), k-block (lane>>4)*8 within the
+  // 32-deep K half an mfma_f32_16x16x32_bf16 consumes
+  const size_t lane_off = (size_t)(lane & 15) * K + (lane >> 4) * 8;
+  const __hip_bfloat16* a0 = A + (size_t)(brow + wave_m * 128) * K + lane_off;
+  const __hip_bfloat16* b0 = Bt + (size_t)(bcol + wave_n * 64) * K + lane_off;
+
+  // static priority for the younger dispatch half (guide T5 static form:
+  // per-burst setprio flips cost 3 spilled VGPRs here; the static form is
+  // register-free). The condition must be wave-uniform via readfirstlane,
+  // else s_setprio lands under an exec mask and applies to every wave.
+  if (SETPRIO && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  f32x4 acc[8][4] = {};
+  bf16x8 bfrag[2][4][2];  // [tile ping-pong][col group][k half]
+  bf16x8 afrag[2][2][2];  // [quadrant ping-pong][row frag][k half]
+
+  // NB: all array indices below are compile-time constants (unrolled loops
+  // + the kt-by-2 buffer unroll) — a runtime buffer index would push these
+  // ext_vector arrays to scratch (guide §5.4 rule 20).
+  auto load_a = [&](int kt, int q, int buf) {
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr)
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh)
+        afrag[buf][rr][kh] = *(const bf16x8*)(
+            a0 + (size_t)(q * 32 + rr * 16) * K + kt * TK + kh * 32);
+  };
+  auto load_b = [&](int kt, int c, int buf) {
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh)
+      bfrag[buf][c][kh] = *(const bf16x8*)(
+          b0 + (size_t)(c * 16) * K + kt * TK + kh * 32);
+  };
+
+  // prologue: tile 0's B fragments + quadrant 0, then enter steady state
+#pragma unroll
+  for (int c = 0; c < 4; ++c) load_b(0, c, 0);
+  load_a(0, 0, 0);
+
+  // one K-tile: 4 quadrant phases x 16 MFMAs; phase p prefetches quadrant
+  // p+1 (next tile's q0 at p==3) and 2 of the next tile's 8 B fragments.
+  // CUR/NXT are template-constant so the fragment arrays stay in registers.
+  auto tile_body = [&](auto cur_c, int kt, bool has_next) {
+    constexpr int CUR = decltype(cur_c)::value;
+    constexpr int NXT = CUR ^ 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      if (p < 3) {
+        load_a(kt, p + 1, (p + 1) & 1);
+      } else if (has_next) {
+        load_a(kt + 1, 0, 0);
+      }
+      if (has_next) load_b(kt + 1, p, NXT);
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+#pragma unroll
+          for (int c = 0; c < 4; ++c) {
+            acc[p * 2 + rr][c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[p & 1][rr][kh], bfrag[CUR][c][kh], acc[p * 2 + rr][c],
+                0, 0, 0);
+          }
+        }
+      }
+    }
+  };
+
+  // ntiles is even for every accepted size (size % 256 == 0 -> K/64 even)
+  for (int kt = 0; kt < ntiles; kt += 2) {
+    tile_body(std::integral_constant<int, 0>{}, kt, true);
+    tile_body(std::integral_constant<int, 1>{}, kt + 1, kt + 2 < ntiles);
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
 py::dict gemm_stress_bf16_v3(int size, int iters) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -1195,6 +1326,79 @@ py::dict gemm_stress_bf16_v2_impl(int size, int iters, bool setprio,
 
 py::dict gemm_stress_bf16_v2(int size, int iters) {
   return gemm_stress_bf16_v2_impl(size, iters, true);
+}
+
+py::dict gemm_stress_bf16_v5_impl(int size, int iters, bool setprio) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  auto* kern = setprio ? gemm_bf16_v5_kernel<true> : gemm_bf16_v5_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
+                     N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c,
+                       M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-regstage-nobarrier";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
+py::dict gemm_stress_bf16_v5(int size, int iters) {
+  return gemm_stress_bf16_v5_impl(size, iters, true);
+}
+
+py::dict gemm_stress_bf16_v5_nosp(int size, int iters) {
+  return gemm_stress_bf16_v5_impl(size, iters, false);
 }
 
 // A/B seam: identical kernel with the s_setprio(1) MFMA-burst hint compiled
@@ -1486,6 +1690,12 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2_panel", &gemm_stress_bf16_v2_panel,
         py::arg("size") = 8192, py::arg("iters") = 5, py::arg("panel") = 16,
         "A/B variant of v2 with L2 panel supertiling (8/16/32, 0=off)");
+  m.def("gemm_stress_bf16_v5", &gemm_stress_bf16_v5, py::arg("size") = 8192,
+        py::arg("iters") = 8,
+        "register-staged no-LDS no-barrier bf16 GEMM stress (v5)");
+  m.def("gemm_stress_bf16_v5_nosp", &gemm_stress_bf16_v5_nosp,
+        py::arg("size") = 8192, py::arg("iters") = 8,
+        "v5 with the s_setprio hint compiled out (A/B seam)");
   m.def("gemm_stress_bf16_v3", &gemm_stress_bf16_v3, py::arg("size") = 8192,
         py::arg("iters") = 5,
         "bf16 GEMM stress, quadrant-phase deep pipeline (no boundary drain)");
