@@ -18,6 +18,12 @@ MFMA TFLOPS/GPU" — the reference (leptonai/gpud) publishes no numbers
 
 Launch: python bench.py --gpus N --steps K --warmup W
 (N>1 via torch.distributed.run, one rank per GPU over RCCL).
+
+--single-process measures the daemon's REAL deployment shape instead: ONE
+process polls all N GPUs per cycle (the daemon monitors every GPU from one
+process — SURVEY.md §7 "overhead flat to 8 GPUs"). With --mock it sets the
+mock backend to N GPUs so the 8-GPU single-process cycle cost is measurable
+on a CPU-only host; on a GPU box it monitors min(N, visible) live GPUs.
 """
 
 import argparse
@@ -94,12 +100,54 @@ class _RankFilteredSMI:
         pass
 
 
+class _SubsetSMI(_RankFilteredSMI):
+    """Expose the FIRST n GPUs of the instance (single-process mode: one
+    process monitors n GPUs per cycle, the daemon's real deployment shape)."""
+
+    def __init__(self, inst, n: int):
+        self._inst = inst
+        uuids = inst.device_uuids()
+        self._uuids = uuids[: max(1, n)]
+        self.uuid = self._uuids[0] if self._uuids else ""
+
+    def devices(self):
+        devs = self._inst.devices()
+        return {u: devs[u] for u in self._uuids if u in devs}
+
+    def device_uuids(self):
+        return list(self._uuids)
+
+    def device_count(self):
+        return len(self._uuids)
+
+    def snapshot_all(self):
+        from gpud_amd.smi import Instance
+
+        devs = self._inst.devices()
+        fi = self._inst.failure_injector
+        out = {}
+        for u in self._uuids:
+            if u not in devs:
+                continue
+            snap = devs[u].snapshot()
+            if fi is not None:
+                snap = Instance._apply_injection(u, snap, fi)
+            out[u] = snap
+        return out
+
+
 def main() -> int:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--mock", action="store_true", help="CPU-only: mock SMI backend")
+    p.add_argument(
+        "--single-process",
+        action="store_true",
+        help="one process polls all --gpus GPUs per cycle (the daemon's real "
+        "deployment shape) instead of one rank per GPU",
+    )
     p.add_argument(
         "--fault-replay",
         action="store_true",
@@ -111,9 +159,21 @@ def main() -> int:
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if args.single_process:
+        world = 1  # one process monitors all GPUs; n_gpus reported separately
 
     if args.mock:
         os.environ["GPUD_AMDSMI_MOCK"] = "1"
+        if args.single_process:
+            # mock exactly the requested node shape (e.g. 8 GPUs, 1 process)
+            os.environ["GPUD_AMDSMI_MOCK_GPUS"] = str(max(1, args.gpus))
+
+    if args.single_process and int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        print(
+            json.dumps({"error": "--single-process is incompatible with torchrun"}),
+            file=sys.stderr,
+        )
+        return 1
 
     import torch
     import torch.distributed as dist
@@ -136,7 +196,12 @@ def main() -> int:
     # filter to this rank's GPU BEFORE building the core so every component
     # is constructed against the single-GPU view
     inst = smi_pkg.new()
-    if inst.exists and inst.device_count() > 0:
+    if inst.exists and inst.device_count() > 0 and args.single_process:
+        n_want = max(1, args.gpus)
+        smi_for_core = _SubsetSMI(inst, n_want)
+        n_gpus_seen = smi_for_core.device_count()
+        data_source = "mock" if args.mock else "amdsmi"
+    elif inst.exists and inst.device_count() > 0:
         smi_for_core = _RankFilteredSMI(inst, local_rank)
         n_gpus_seen = 1
         data_source = "mock" if args.mock else "amdsmi"
@@ -274,19 +339,24 @@ def main() -> int:
             "metric": "poll_cycle_p50_ms",
             "value": round(p50, 4),
             "unit": "ms",
-            "n_gpus": world,
+            "n_gpus": n_gpus_seen if args.single_process else world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(mean_ms, 4),
             "higher_is_better": False,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp64",
+            # telemetry poll metric: no compute dtype (a dtype label here
+            # invites misreading the poll latency as a numeric benchmark)
+            "dtype": None,
             "data": "synthetic" if data_source == "mock" else "live-telemetry",
             "config": {
                 "model": "gpud-amd accelerator poll cycle",
+                "mode": "single-process" if args.single_process else "rank-per-gpu",
                 "components_per_cycle": len(accel_components),
-                "gpus_per_rank": 1 if data_source == "amdsmi" else n_gpus_seen,
+                "gpus_per_rank": n_gpus_seen
+                if args.single_process
+                else (1 if data_source == "amdsmi" else n_gpus_seen),
                 "data_source": data_source,
                 "poll_cycle_p99_ms": round(p99, 4),
                 "daemon_cpu_percent": round(cpu_pct_max, 2),

@@ -52,6 +52,35 @@ def test_bench_single_rank_mock():
     assert required <= set(d), required - set(d)
     assert d["vs_baseline"] is None  # the reference publishes no numbers
     assert isinstance(d["config"], dict) and d["config"].get("model")
+    # a poll-latency metric has no compute dtype (an "fp64"
+    # label here invited misreading)
+    assert d["dtype"] is None
+
+
+def test_bench_single_process_multi_gpu_mock():
+    """--single-process: ONE process polls all N GPUs per cycle — the
+    daemon's real deployment shape. The mock backend
+    is sized to N so an 8-GPU cycle is measurable on a CPU-only host."""
+    values = {}
+    for n in (1, 8):
+        out = subprocess.run(
+            [
+                sys.executable, "bench.py", "--mock", "--single-process",
+                "--gpus", str(n), "--steps", "25", "--warmup", "2",
+            ],
+            capture_output=True, text=True, timeout=240, cwd=REPO,
+        )
+        assert out.returncode == 0, out.stderr[-800:]
+        d = json.loads(out.stdout.strip().splitlines()[-1])
+        assert d["n_gpus"] == n
+        assert d["config"]["mode"] == "single-process"
+        assert d["config"]["gpus_per_rank"] == n
+        assert d["value"] > 0
+        values[n] = d["value"]
+    # one cycle over 8 GPUs costs more than over 1, but sub-linearly
+    # (shared-session snapshot; SURVEY §7 "overhead flat to 8 GPUs")
+    assert values[8] > values[1]
+    assert values[8] < values[1] * 8
 
 
 def test_bench_two_ranks_gloo_mock():
