@@ -101,6 +101,10 @@ def run(
         "accepts bootstrap/update requests)"
     ),
     in_memory_db: bool = typer.Option(False, help="ephemeral state (testing)"),
+    kmsg_path: str = typer.Option(
+        "/dev/kmsg", help="kmsg source/sink; a regular file switches the "
+        "fault-injection loop to a file seam (rate-limited-kmsg hosts)"
+    ),
     no_tls: bool = typer.Option(False, help="serve plain HTTP (testing)"),
     log_level: str = typer.Option("info"),
     # SMI-level failure injection for exercising the daemon on healthy
@@ -183,6 +187,7 @@ def run(
     cfg.expected_ib_ports = expected_ib_ports
     cfg.expected_ib_rate_gbps = expected_ib_rate_gbps
     cfg.poll_interval_seconds = poll_interval_seconds
+    cfg.kmsg_path = kmsg_path
     if not in_memory_db:
         os.makedirs(cfg.data_dir, exist_ok=True)
 

@@ -47,6 +47,11 @@ class Config:
     docker_socket: str = "/var/run/docker.sock"
     nfs_host_root: str = ""
     plugin_specs_file: str = ""
+    # kmsg source/sink path. Pointing this at a regular file turns the
+    # fault-injection loop into a file seam (FileSeamWriter + poll-follow
+    # watcher) for environments whose /dev/kmsg write path is rate-limited
+    # — same role as the reference's nsenter-style command override seams.
+    kmsg_path: str = "/dev/kmsg"
     kernel_modules_to_check: List[str] = field(default_factory=list)
     libraries_to_check: Dict[str, List[str]] = field(default_factory=dict)
     mount_points: List[str] = field(default_factory=lambda: ["/"])

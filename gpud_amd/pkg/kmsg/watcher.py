@@ -50,11 +50,41 @@ class Watcher:
         self._callbacks: List[Callable[[Message], None]] = []
         self._cb_lock = threading.Lock()
 
+    def _is_device(self) -> bool:
+        """True for the real /dev/kmsg (one record per read(2), EPIPE
+        semantics); False for a regular-file seam (FileSeamWriter output,
+        line-framed, poll-followed)."""
+        import stat
+
+        try:
+            return stat.S_ISCHR(os.stat(self.path).st_mode)
+        except OSError:
+            return self.path == KMSG_PATH
+
     # -- one-shot ring read (reference watcher.go ReadAll) ------------------
 
     def read_all(self, limit: int = 100_000) -> List[Message]:
         boot = boot_wall_time()
         out: List[Message] = []
+        if not self._is_device():
+            # file seam: line-framed records
+            try:
+                with open(self.path, "r", errors="replace") as f:
+                    for line in f:
+                        line = line.rstrip("\n")
+                        if not line:
+                            continue
+                        if line[:1] == " " and out:
+                            parse_continuation(line, out[-1])
+                            continue
+                        m = parse_line(line, boot)
+                        if m is not None:
+                            out.append(m)
+                        if len(out) >= limit:
+                            break
+            except OSError as e:
+                logger.warning("cannot open %s: %s", self.path, e)
+            return out
         try:
             fd = os.open(self.path, os.O_RDONLY | os.O_NONBLOCK)
         except OSError as e:
@@ -93,6 +123,16 @@ class Watcher:
     def start(self, from_start: bool = False) -> None:
         if self._thread is not None:
             return
+        if not self._is_device():
+            # file seam: poll-follow from the current end (the file may not
+            # exist yet — the first poll picks it up)
+            self._seam_from_start = from_start
+            self._thread = threading.Thread(
+                target=self._follow_file_loop, daemon=True,
+                name="gpud-kmsg-watch",
+            )
+            self._thread.start()
+            return
         try:
             # blocking fd for the follow loop
             self._follow_fd = os.open(self.path, os.O_RDONLY)
@@ -105,6 +145,51 @@ class Watcher:
             target=self._follow_loop, daemon=True, name="gpud-kmsg-watch"
         )
         self._thread.start()
+
+    def _deliver(self, m: Message) -> None:
+        with self._cb_lock:
+            cbs = list(self._callbacks)
+        for cb in cbs:
+            try:
+                cb(m)
+            except Exception:
+                logger.exception("kmsg callback failed")
+
+    def _follow_file_loop(self) -> None:
+        boot = boot_wall_time()
+        pos = None  # unknown until the file exists
+        buf = ""
+        last: Optional[Message] = None
+        while not self._stop.is_set():
+            try:
+                with open(self.path, "r", errors="replace") as f:
+                    if pos is None:
+                        f.seek(0, 0 if getattr(self, "_seam_from_start", False) else 2)
+                        pos = f.tell()
+                    else:
+                        end = f.seek(0, 2)
+                        if end < pos:
+                            pos = 0  # truncated/rotated: restart
+                        f.seek(pos)
+                    chunk = f.read()
+                    pos = f.tell()
+            except OSError:
+                chunk = ""
+            if chunk:
+                buf += chunk
+                *lines, buf = buf.split("\n")
+                for line in lines:
+                    if not line:
+                        continue
+                    if line[:1] == " " and last is not None:
+                        parse_continuation(line, last)
+                        continue
+                    m = parse_line(line, boot)
+                    if m is not None:
+                        last = m
+                        self._deliver(m)
+            else:
+                self._stop.wait(0.1)
 
     def _follow_loop(self) -> None:
         boot = boot_wall_time()

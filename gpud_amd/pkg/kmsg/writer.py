@@ -44,6 +44,35 @@ class Writer:
         return None
 
 
+class FileSeamWriter(Writer):
+    """Writes kmsg-RECORD-format lines (``pri,seq,ts_us,-;msg``) to a
+    regular file. The injection seam for environments that rate-limit
+    /dev/kmsg writes: point the daemon's --kmsg-path
+    at a file and the full inject-fault -> watcher -> error-ras loop runs
+    without touching the kernel ring. /dev/kmsg itself formats records on
+    READ, so a file seam must produce the read format the parser expects."""
+
+    def __init__(self, path: str):
+        super().__init__(path=path)
+        self._seq = 0
+
+    def write(self, message: str, priority: int = 2) -> Optional[str]:
+        import time as _time
+
+        from .parser import boot_wall_time
+
+        ts_us = int((_time.time() - boot_wall_time()) * 1e6)
+        try:
+            with open(self.path, "a") as f:
+                for i in range(0, max(len(message), 1), MAX_PAYLOAD):
+                    self._seq += 1
+                    chunk = message[i : i + MAX_PAYLOAD]
+                    f.write(f"{priority},{self._seq},{ts_us},-;{chunk}\n")
+        except OSError as e:
+            return f"cannot write {self.path}: {e}"
+        return None
+
+
 class NoopWriter(Writer):
     """Used where /dev/kmsg is not writable (tests, unprivileged runs)."""
 

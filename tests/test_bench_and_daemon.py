@@ -171,6 +171,75 @@ def test_daemon_boot_via_cli():
             proc.wait(timeout=5)
 
 
+def test_fault_injection_kmsg_file_seam(tmp_path):
+    """The full inject-fault -> kmsg -> error-ras -> set-healthy loop
+    through the --kmsg-path FILE seam (this loop must
+    be assertable even where /dev/kmsg writes are rate-limited)."""
+    import signal
+    import time as _time
+
+    seam = tmp_path / "kmsg-seam"
+    seam.write_text("")
+    port = _free_port()
+    env = {
+        **os.environ,
+        "GPUD_AMDSMI_MOCK": "1",
+        "GPUD_AMDSMI_MOCK_GPUS": "1",
+        "PYTHONPATH": REPO,
+    }
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "gpud_amd", "run",
+            "--in-memory-db", "--address", f"127.0.0.1:{port}",
+            "--log-level", "warning",
+            "--kmsg-path", str(seam),
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        start_new_session=True,
+    )
+    try:
+        from gpud_amd.client import Client
+
+        client = Client(f"https://127.0.0.1:{port}")
+        assert client.wait_healthz(30), "daemon did not become healthy"
+
+        client.inject_fault(ras_event_name="amdgpu_ring_timeout")
+        # the record must have reached the seam file in read-format
+        deadline = _time.time() + 10
+        detected = False
+        st = None
+        while _time.time() < deadline:
+            st = client.get_health_states(
+                components=["accelerator-amd-error-ras"]
+            )["accelerator-amd-error-ras"][0]
+            if st.health == "Unhealthy":
+                detected = True
+                break
+            client.trigger_check(component="accelerator-amd-error-ras")
+            _time.sleep(0.3)
+        assert detected, f"seam injection not detected: {st and st.reason}"
+        assert "amdgpu_ring_timeout" in st.reason
+        assert ";" in seam.read_text()  # read-format record landed
+
+        client.set_healthy(["accelerator-amd-error-ras"])
+        st = client.get_health_states(
+            components=["accelerator-amd-error-ras"]
+        )["accelerator-amd-error-ras"][0]
+        assert st.health == "Healthy", st.reason
+        client.close()
+    finally:
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            os.killpg(proc.pid, signal.SIGKILL)
+            proc.wait(timeout=5)
+
+
 def test_bench_fault_replay_mock():
     out = subprocess.run(
         [
