@@ -322,6 +322,7 @@ def run(
     from ..pkg.gpud_manager import PackageController
 
     pkg_controller = PackageController(cfg)
+    core.pkg_controller = pkg_controller  # /admin/packages reads live status
     pkg_controller.start()
 
     stop = {"flag": False}

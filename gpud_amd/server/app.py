@@ -274,6 +274,11 @@ def create_app(core: DaemonCore, plugin_specs: Optional[list] = None) -> FastAPI
 
     @app.get("/admin/packages")
     def admin_packages(request: Request):
+        # prefer the live controller's rich status (reference: the manager's
+        # packages.PackageStatus shape); fall back to one-shot probes
+        ctl = getattr(core, "pkg_controller", None)
+        if ctl is not None:
+            return _negotiate(request, ctl.admin_statuses())
         from ..pkg.gpud_manager import package_statuses
 
         return _negotiate(

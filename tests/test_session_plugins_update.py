@@ -406,8 +406,11 @@ def test_package_manager(tmp_path):
 case "$1" in
   isInstalled) [ -f {marker} ] && exit 0 || exit 1 ;;
   install) touch {marker}; exit 0 ;;
+  start) exit 0 ;;
+  status) exit 0 ;;
   version) echo 1.2.3 ;;
   run) exit 0 ;;
+  *) exit 1 ;;
 esac
 """
     )
