@@ -1,15 +1,30 @@
-"""disk — mount-point usage and block devices.
+"""disk — mount-point usage, mount-target tracking, block devices.
 
-Reference: components/disk (statfs usage per mount point, findmnt/lsblk
-with nsenter overrides and retries — disk/component.go:600-623,175-181).
-Python-first: psutil statfs for usage; ``lsblk -J`` (override-able command)
-for the block-device tree.
+Reference: components/disk + pkg/disk. Mechanism parity items:
+
+  * statfs usage per mount point with used-percent thresholds
+    (disk/component.go free-space thresholds);
+  * **findmnt with retries** for every tracked mount target
+    (disk/component.go:600-623 — findmnt is occasionally flaky and returns
+    empty output which fails JSON parsing; that is transient, so up to 5
+    attempts with a retry interval, only a fully-exhausted budget is an
+    error), ``--target --json --df --bytes``, command override supported
+    (the reference's nsenter seam, pkg/disk/findmnt.go:35);
+  * **lsblk with flush-retry** and recursive tree flattening
+    (disk/component.go:175-181, pkg/disk/lsblk.go) — transient lsblk
+    failures retry, children are flattened into the device list, and
+    devices whose fstype/mountpoint are missing from lsblk are back-filled
+    via findmnt (pkg/disk's fstype fallback).
+
+Every external probe is a function field so tests inject fakes (the
+reference's injected-function-field pattern).
 """
 
 from __future__ import annotations
 
 import json
 import subprocess
+import time
 from typing import Callable, Dict, List, Optional
 
 import psutil
@@ -22,22 +37,145 @@ NAME = "disk"
 
 DEFAULT_USED_PERCENT_DEGRADED = 90.0
 DEFAULT_USED_PERCENT_UNHEALTHY = 98.0
+FINDMNT_RETRIES = 5  # reference: disk/component.go:595 "for attempt := range 5"
+LSBLK_RETRIES = 3
+DEFAULT_RETRY_INTERVAL_SECONDS = 1.0
 
 
-def list_block_devices(lsblk_command: str = "") -> Optional[List[Dict]]:
-    cmd = lsblk_command or "lsblk"
-    try:
-        out = subprocess.run(
-            [cmd, "-J", "-b", "-o", "NAME,TYPE,SIZE,MOUNTPOINT,FSTYPE"],
-            capture_output=True,
-            text=True,
-            timeout=15,
+def _run(cmd: List[str], timeout: float = 15.0) -> subprocess.CompletedProcess:
+    return subprocess.run(cmd, capture_output=True, text=True, timeout=timeout)
+
+
+# ---------------------------------------------------------------------------
+# findmnt (reference: pkg/disk/findmnt.go)
+# ---------------------------------------------------------------------------
+
+
+def find_mnt(
+    target: str,
+    findmnt_command: str = "",
+    run: Callable = _run,
+) -> Optional[Dict]:
+    """One findmnt invocation for a mount target. Returns the reference's
+    FindMntOutput shape ({target, filesystems:[{mounted_point, sources,
+    fstype, size_bytes, used_bytes, available_bytes, used_percent}]}) or
+    raises on command/parse failure (the caller retries)."""
+    cmd = (findmnt_command or "findmnt").split() + [
+        "--target", target, "--json", "--df", "--bytes",
+    ]
+    out = run(cmd)
+    if out.returncode != 0:
+        raise RuntimeError(f"findmnt exit {out.returncode}: {out.stderr[:200]}")
+    raw = json.loads(out.stdout)  # empty/garbled output raises -> retried
+    filesystems = []
+    for fs in raw.get("filesystems", []):
+        sources = fs.get("sources") or []
+        if not sources and fs.get("source"):
+            sources = [fs["source"]]
+        pct_raw = str(fs.get("use%", "0%")).rstrip("%")
+        try:
+            pct = float(pct_raw)
+        except ValueError:
+            pct = 0.0
+        filesystems.append(
+            {
+                "mounted_point": fs.get("target", ""),
+                "sources": sources,
+                "fstype": fs.get("fstype", ""),
+                "size_bytes": int(fs.get("size") or 0),
+                "used_bytes": int(fs.get("used") or 0),
+                "available_bytes": int(fs.get("avail") or 0),
+                "used_percent": pct,
+            }
         )
-        if out.returncode != 0:
-            return None
-        return json.loads(out.stdout).get("blockdevices", [])
-    except (OSError, subprocess.TimeoutExpired, json.JSONDecodeError):
+    return {"target": target, "filesystems": filesystems}
+
+
+def find_mnt_with_retries(
+    target: str,
+    findmnt_command: str = "",
+    retries: int = FINDMNT_RETRIES,
+    retry_interval: float = DEFAULT_RETRY_INTERVAL_SECONDS,
+    run: Callable = _run,
+    sleep: Callable = time.sleep,
+) -> Optional[Dict]:
+    """findmnt with the reference's transient-failure retry loop
+    (disk/component.go:600-623). Returns None only after the whole retry
+    budget is exhausted."""
+    for attempt in range(retries):
+        try:
+            return find_mnt(target, findmnt_command, run=run)
+        except (RuntimeError, OSError, ValueError, subprocess.TimeoutExpired,
+                json.JSONDecodeError):
+            if attempt + 1 < retries:
+                sleep(retry_interval)
+    return None
+
+
+# ---------------------------------------------------------------------------
+# lsblk (reference: pkg/disk/lsblk.go — JSON tree, flatten, fstype fallback)
+# ---------------------------------------------------------------------------
+
+_LSBLK_COLUMNS = "NAME,TYPE,SIZE,MOUNTPOINT,FSTYPE,PKNAME"
+
+
+def _flatten_devices(devs: List[Dict], parent: str = "") -> List[Dict]:
+    """Flatten lsblk's nested children tree (pkg/disk/lsblk_flatten.go):
+    every device carries its parent's name so usage can be attributed to
+    the physical device."""
+    flat: List[Dict] = []
+    for d in devs:
+        entry = {k: v for k, v in d.items() if k != "children"}
+        if parent and not entry.get("pkname"):
+            entry["pkname"] = parent
+        flat.append(entry)
+        if d.get("children"):
+            flat.extend(_flatten_devices(d["children"], d.get("name", "")))
+    return flat
+
+
+def list_block_devices(
+    lsblk_command: str = "",
+    findmnt_command: str = "",
+    retries: int = LSBLK_RETRIES,
+    retry_interval: float = DEFAULT_RETRY_INTERVAL_SECONDS,
+    run: Callable = _run,
+    sleep: Callable = time.sleep,
+) -> Optional[List[Dict]]:
+    """lsblk JSON with the flush-retry loop (transient empty/garbled output
+    retries — disk/component.go:175-181) and the fstype fallback: a mounted
+    device lsblk reports without an fstype is back-filled via findmnt
+    (pkg/disk's DefaultFsTypeFunc fallback)."""
+    cmd = (lsblk_command or "lsblk").split() + [
+        "-J", "-b", "-o", _LSBLK_COLUMNS,
+    ]
+    devices: Optional[List[Dict]] = None
+    for attempt in range(retries):
+        try:
+            out = run(cmd)
+            if out.returncode != 0:
+                raise RuntimeError(f"lsblk exit {out.returncode}")
+            devices = json.loads(out.stdout).get("blockdevices", [])
+            break
+        except (OSError, RuntimeError, subprocess.TimeoutExpired,
+                json.JSONDecodeError, ValueError):
+            if attempt + 1 < retries:
+                sleep(retry_interval)
+    if devices is None:
         return None
+    flat = _flatten_devices(devices)
+    # fstype back-fill for mounted devices lsblk could not type
+    for d in flat:
+        if d.get("mountpoint") and not d.get("fstype"):
+            try:
+                mnt = find_mnt(d["mountpoint"], findmnt_command, run=run)
+            except Exception:  # noqa: BLE001 — fallback only
+                continue
+            for fs in mnt["filesystems"]:
+                if fs["fstype"]:
+                    d["fstype"] = fs["fstype"]
+                    break
+    return flat
 
 
 class DiskComponent(TickerComponent):
@@ -47,9 +185,21 @@ class DiskComponent(TickerComponent):
         self.mount_points = list(inst.mount_points or ["/"])
         self.mount_targets = list(inst.mount_targets or [])
         self._lsblk_command = inst.lsblk_command
+        self._findmnt_command = getattr(inst, "findmnt_command", "")
+        self.retry_interval = DEFAULT_RETRY_INTERVAL_SECONDS
+        # injected function fields (reference test pattern)
         self.get_block_devices: Callable = lambda: list_block_devices(
-            self._lsblk_command
+            self._lsblk_command,
+            self._findmnt_command,
+            retry_interval=self.retry_interval,
         )
+        self.find_mnt: Callable = lambda target: find_mnt_with_retries(
+            target,
+            self._findmnt_command,
+            retry_interval=self.retry_interval,
+        )
+        # cached last probes (queryable via extra_info)
+        self.mount_target_usages: Dict[str, Dict] = {}
 
     @property
     def name(self) -> str:
@@ -61,7 +211,7 @@ class DiskComponent(TickerComponent):
     def check(self) -> CheckResult:
         degraded, unhealthy, missing = [], [], []
         extra = {}
-        for mp in self.mount_points + self.mount_targets:
+        for mp in self.mount_points:
             try:
                 u = psutil.disk_usage(mp)
             except OSError:
@@ -82,6 +232,39 @@ class DiskComponent(TickerComponent):
                 unhealthy.append(mp)
             elif u.percent >= DEFAULT_USED_PERCENT_DEGRADED:
                 degraded.append(mp)
+
+        # mount-target tracking via findmnt-with-retries (reference:
+        # MountTargetUsages — a target findmnt cannot resolve after the
+        # retry budget is recorded, logged, and does not flip health,
+        # matching disk/component.go:626)
+        failed_targets = []
+        for target in self.mount_targets:
+            mnt = self.find_mnt(target)
+            if mnt is None:
+                failed_targets.append(target)
+                continue
+            self.mount_target_usages[target] = mnt
+            for fs in mnt["filesystems"]:
+                extra[f"target.{target}.mounted_point"] = fs["mounted_point"]
+                extra[f"target.{target}.fstype"] = fs["fstype"]
+                extra[f"target.{target}.used_percent"] = f"{fs['used_percent']:.1f}"
+                self._gauges.set(
+                    "mount_target_used_percent",
+                    "Mount-target used percent (findmnt)",
+                    fs["used_percent"],
+                    mount_point=target,
+                )
+                break
+        if failed_targets:
+            extra["findmnt_failed_targets"] = ", ".join(failed_targets)
+
+        # block-device tree (flush-retried lsblk)
+        devs = self.get_block_devices()
+        if devs is not None:
+            extra["block_devices"] = str(len(devs))
+            disks = [d for d in devs if d.get("type") == "disk"]
+            extra["disks"] = str(len(disks))
+
         if unhealthy or missing:
             parts = []
             if unhealthy:

@@ -75,6 +75,14 @@ class Config:
     ras_event_thresholds: Dict[str, int] = field(default_factory=dict)
     zombie_degraded_threshold: int = 1000
     zombie_unhealthy_threshold: int = 2000
+    # D-state (uninterruptible sleep) persistence tracking (reference:
+    # components/os/threshold.go): consecutive one-minute checks before a
+    # blocked process counts as persistent, and the name regexes that gate
+    # escalation to unhealthy+reboot (empty list = no escalation)
+    dstate_persistence_threshold: int = 5
+    dstate_name_regexes: List[str] = field(
+        default_factory=lambda: ["^amd", "^rocm"]
+    )
     # control plane
     endpoint: str = ""
     token: str = ""

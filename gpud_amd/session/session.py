@@ -471,6 +471,8 @@ class Session:
             "ras_event_thresholds",
             "zombie_degraded_threshold",
             "zombie_unhealthy_threshold",
+            "dstate_persistence_threshold",
+            "dstate_name_regexes",
             "nfs_host_root",
             "kernel_modules_to_check",
             "libraries_to_check",

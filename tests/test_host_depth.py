@@ -1,0 +1,373 @@
+"""Round-2 host-component mechanism depth: disk findmnt/lsblk retries and
+mount-target tracking, os D-state persistence tracker, containerd CRI probe
+(reference: components/disk/component.go:600-623,175-181,
+components/os/blocked_processes.go, containerd/component.go:283)."""
+
+import json
+import subprocess
+import types
+
+import pytest
+
+from gpud_amd.apiv1.types import HealthStateType, RepairActionType
+
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
+
+
+def _cp(stdout="", rc=0, stderr=""):
+    return subprocess.CompletedProcess(args=[], returncode=rc, stdout=stdout,
+                                       stderr=stderr)
+
+
+FINDMNT_JSON = json.dumps(
+    {
+        "filesystems": [
+            {
+                "target": "/var/lib/gpud",
+                "source": "/dev/nvme0n1p2",
+                "fstype": "ext4",
+                "size": 1000000,
+                "used": 250000,
+                "avail": 750000,
+                "use%": "25%",
+            }
+        ]
+    }
+)
+
+LSBLK_JSON = json.dumps(
+    {
+        "blockdevices": [
+            {
+                "name": "nvme0n1", "type": "disk", "size": 2000000,
+                "mountpoint": None, "fstype": None,
+                "children": [
+                    {"name": "nvme0n1p1", "type": "part", "size": 500000,
+                     "mountpoint": "/boot", "fstype": "vfat"},
+                    {"name": "nvme0n1p2", "type": "part", "size": 1500000,
+                     "mountpoint": "/", "fstype": None},
+                ],
+            }
+        ]
+    }
+)
+
+
+# ---------------------------------------------------------------------------
+# disk: findmnt with retries
+# ---------------------------------------------------------------------------
+
+
+def test_find_mnt_parses_output():
+    from gpud_amd.components.host.disk import find_mnt
+
+    out = find_mnt("/var/lib/gpud", run=lambda cmd, **kw: _cp(FINDMNT_JSON))
+    assert out["target"] == "/var/lib/gpud"
+    fs = out["filesystems"][0]
+    assert fs["mounted_point"] == "/var/lib/gpud"
+    assert fs["sources"] == ["/dev/nvme0n1p2"]
+    assert fs["fstype"] == "ext4"
+    assert fs["size_bytes"] == 1000000
+    assert fs["used_percent"] == 25.0
+
+
+def test_find_mnt_retries_transient_empty_output():
+    """findmnt occasionally returns empty output (unparseable JSON) — the
+    reference treats that as transient and retries up to 5 times
+    (disk/component.go:600-623)."""
+    from gpud_amd.components.host.disk import find_mnt_with_retries
+
+    calls = {"n": 0}
+    sleeps = []
+
+    def flaky(cmd, **kw):
+        calls["n"] += 1
+        if calls["n"] < 3:
+            return _cp("")  # empty output -> JSONDecodeError -> retry
+        return _cp(FINDMNT_JSON)
+
+    out = find_mnt_with_retries(
+        "/var/lib/gpud", run=flaky, sleep=sleeps.append, retry_interval=0.5
+    )
+    assert out is not None and out["filesystems"]
+    assert calls["n"] == 3
+    assert sleeps == [0.5, 0.5]
+
+
+def test_find_mnt_exhausted_budget_returns_none():
+    from gpud_amd.components.host.disk import find_mnt_with_retries
+
+    out = find_mnt_with_retries(
+        "/nope", run=lambda cmd, **kw: _cp("", rc=1), sleep=lambda s: None
+    )
+    assert out is None
+
+
+def test_find_mnt_command_override():
+    """The nsenter-style override is used verbatim as the command prefix
+    (reference: pkg/disk/findmnt.go FindMntWithCommand)."""
+    from gpud_amd.components.host.disk import find_mnt
+
+    seen = {}
+
+    def capture(cmd, **kw):
+        seen["cmd"] = cmd
+        return _cp(FINDMNT_JSON)
+
+    find_mnt("/x", findmnt_command="nsenter -t 1 -m findmnt", run=capture)
+    assert seen["cmd"][:5] == ["nsenter", "-t", "1", "-m", "findmnt"]
+    assert "--target" in seen["cmd"] and "/x" in seen["cmd"]
+
+
+# ---------------------------------------------------------------------------
+# disk: lsblk flush-retry + flatten + fstype backfill
+# ---------------------------------------------------------------------------
+
+
+def test_lsblk_flush_retry_and_flatten():
+    from gpud_amd.components.host.disk import list_block_devices
+
+    calls = {"n": 0}
+
+    def flaky(cmd, **kw):
+        if cmd[0] == "lsblk":
+            calls["n"] += 1
+            if calls["n"] == 1:
+                return _cp("garbage{")  # transient -> retried
+            return _cp(LSBLK_JSON)
+        return _cp(FINDMNT_JSON)  # findmnt backfill for "/"
+
+    devs = list_block_devices(run=flaky, sleep=lambda s: None)
+    assert calls["n"] == 2
+    names = [d["name"] for d in devs]
+    # children flattened with parent attribution
+    assert names == ["nvme0n1", "nvme0n1p1", "nvme0n1p2"]
+    assert devs[1]["pkname"] == "nvme0n1"
+    # fstype back-filled via findmnt for the mounted-but-untyped partition
+    p2 = devs[2]
+    assert p2["mountpoint"] == "/"
+    assert p2["fstype"] == "ext4"
+
+
+def test_lsblk_exhausted_returns_none():
+    from gpud_amd.components.host.disk import list_block_devices
+
+    assert list_block_devices(
+        run=lambda cmd, **kw: _cp("", rc=1), sleep=lambda s: None
+    ) is None
+
+
+def test_disk_component_tracks_mount_targets(mock_core):
+    from gpud_amd.components.host.disk import DiskComponent
+
+    comp = DiskComponent(mock_core.gpud_instance)
+    comp.mount_targets = ["/var/lib/gpud", "/data/missing"]
+    comp.find_mnt = lambda t: (
+        {"target": t, "filesystems": [{
+            "mounted_point": "/", "sources": ["/dev/sda1"], "fstype": "ext4",
+            "size_bytes": 10, "used_bytes": 5, "available_bytes": 5,
+            "used_percent": 50.0}]}
+        if t == "/var/lib/gpud" else None
+    )
+    comp.get_block_devices = lambda: [{"name": "sda", "type": "disk"}]
+    cr = comp.check()
+    # target usage recorded; a findmnt-failed target is logged in extra but
+    # does not flip health (reference behavior: logs only)
+    assert cr.extra_info["target./var/lib/gpud.fstype"] == "ext4"
+    assert cr.extra_info["findmnt_failed_targets"] == "/data/missing"
+    assert cr.health == HealthStateType.HEALTHY
+    assert comp.mount_target_usages["/var/lib/gpud"]["filesystems"]
+
+
+# ---------------------------------------------------------------------------
+# os: D-state persistence tracker
+# ---------------------------------------------------------------------------
+
+
+def test_blocked_tracker_persistence_and_wall_gate():
+    from gpud_amd.components.host.os_component import BlockedProcessTracker
+
+    t = BlockedProcessTracker()
+    # 5 consecutive checks in a burst (same wall time): NOT persistent —
+    # the wall-time gate resists trigger-check bursts
+    for _ in range(5):
+        upd = t.update(1000.0, [(42, "amd-smi")], persistence_threshold=5)
+    assert upd["persistent"] == []
+    # same 5 checks spaced a minute apart: persistent
+    t.reset()
+    for i in range(5):
+        upd = t.update(1000.0 + i * 60, [(42, "amd-smi")], persistence_threshold=5)
+    assert len(upd["persistent"]) == 1
+    bp = upd["persistent"][0]
+    assert bp.pid == 42 and bp.name == "amd-smi"
+    assert bp.consecutive_checks == 5
+    assert bp.blocked_seconds == 240
+
+
+def test_blocked_tracker_absence_grace():
+    """A tracked PID absent for ONE check keeps its counter (PID churn /
+    transient /proc read failure); two absences drop it (recovered)."""
+    from gpud_amd.components.host.os_component import BlockedProcessTracker
+
+    t = BlockedProcessTracker()
+    t.update(0.0, [(7, "dd")])
+    t.update(60.0, [(7, "dd")])
+    upd = t.update(120.0, [])  # absent once: kept
+    assert upd["cleared"] == []
+    upd = t.update(180.0, [(7, "dd")])  # back: counter continued
+    assert upd["persistent"] == []  # 3 checks < threshold 5
+    upd = t.update(240.0, [])
+    upd = t.update(300.0, [])  # absent twice: cleared
+    assert [c.pid for c in upd["cleared"]] == [7]
+
+
+def test_os_dstate_escalation_and_set_healthy(mock_core):
+    from gpud_amd.components.host.os_component import OSComponent
+
+    comp = OSComponent(mock_core.gpud_instance)
+    clock = {"t": 1000.0}
+    comp.get_time_now = lambda: clock["t"]
+    comp.get_process_states = lambda: {
+        "total": 100, "zombies": 0, "dstate": 1,
+        "blocked": [(1234, "amd-smi")],
+    }
+    # below the persistence threshold: healthy
+    for _ in range(4):
+        cr = comp.check()
+        clock["t"] += 60
+        assert cr.health == HealthStateType.HEALTHY, cr.reason
+    # 5th consecutive one-minute check: unhealthy + reboot suggestion
+    cr = comp.check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert "amd-smi" in cr.reason
+    assert cr.suggested_actions is not None
+    assert RepairActionType.REBOOT_SYSTEM in cr.suggested_actions.repair_actions
+    # set-healthy resets the tracker: the process must re-earn persistence
+    comp.set_healthy()
+    clock["t"] += 60
+    cr = comp.check()
+    assert cr.health == HealthStateType.HEALTHY
+
+
+def test_os_dstate_non_matching_name_degrades_only(mock_core):
+    """Escalation is name-gated: a persistent D-state dd degrades, only
+    management-process names (default ^amd/^rocm) go unhealthy."""
+    from gpud_amd.components.host.os_component import OSComponent
+
+    comp = OSComponent(mock_core.gpud_instance)
+    clock = {"t": 0.0}
+    comp.get_time_now = lambda: clock["t"]
+    comp.get_process_states = lambda: {
+        "total": 10, "zombies": 0, "dstate": 1, "blocked": [(9, "dd")],
+    }
+    for _ in range(6):
+        cr = comp.check()
+        clock["t"] += 60
+    assert cr.health == HealthStateType.DEGRADED
+    assert "dd" in cr.reason
+
+
+def test_os_dstate_reboot_escalates_to_hw_inspection(mock_core):
+    from gpud_amd.components.host.os_component import OSComponent
+
+    comp = OSComponent(mock_core.gpud_instance)
+    clock = {"t": 0.0}
+    comp.get_time_now = lambda: clock["t"]
+    comp.get_process_states = lambda: {
+        "total": 10, "zombies": 0, "dstate": 1,
+        "blocked": [(5, "rocm-smi")],
+    }
+    comp._reboot_store = types.SimpleNamespace(
+        reboot_count_since=lambda since: 2,
+        get_reboot_events=lambda since: [],
+    )
+    for _ in range(6):
+        cr = comp.check()
+        clock["t"] += 60
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert (
+        RepairActionType.HARDWARE_INSPECTION
+        in cr.suggested_actions.repair_actions
+    )
+
+
+# ---------------------------------------------------------------------------
+# containerd: CRI probe + GPU runtime config
+# ---------------------------------------------------------------------------
+
+
+def _pb_string(field_no, s):
+    b = s.encode()
+    return bytes([(field_no << 3) | 2, len(b)]) + b
+
+
+def test_cri_response_decode():
+    from gpud_amd.components.host.containerd import _decode_string_fields
+
+    payload = (
+        _pb_string(1, "0.1.0")
+        + _pb_string(2, "containerd")
+        + _pb_string(3, "1.7.27")
+        + _pb_string(4, "v1")
+    )
+    fields = _decode_string_fields(payload)
+    assert fields == {1: "0.1.0", 2: "containerd", 3: "1.7.27", 4: "v1"}
+
+
+def test_gpu_runtime_configuration_detection():
+    from gpud_amd.components.host.containerd import (
+        has_gpu_runtime_configuration,
+    )
+
+    v1 = '[plugins."io.containerd.grpc.v1.cri".containerd.runtimes.amd]\n'
+    v2 = '[plugins."io.containerd.cri.v1.runtime".containerd.runtimes.nvidia]\n'
+    assert has_gpu_runtime_configuration(v1)
+    assert has_gpu_runtime_configuration(v2)
+    assert not has_gpu_runtime_configuration(
+        '[plugins."io.containerd.grpc.v1.cri".containerd.runtimes.runc]'
+    )
+
+
+def test_containerd_component_cri_probe(mock_core):
+    from gpud_amd.components.host.containerd import ContainerdComponent
+
+    comp = ContainerdComponent(mock_core.gpud_instance)
+    comp.check_socket = lambda: True
+    comp.check_service = lambda: "active"
+    comp.get_cri_version = lambda: {
+        "version": "0.1.0", "runtime_name": "containerd",
+        "runtime_version": "1.7.27", "runtime_api_version": "v1",
+    }
+    comp.get_config = lambda: (
+        '[plugins."io.containerd.grpc.v1.cri".containerd.runtimes.amd]'
+    )
+    cr = comp.check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "CRI containerd 1.7.27" in cr.reason
+    assert cr.extra_info["cri_runtime_api_version"] == "v1"
+    assert cr.extra_info["gpu_runtime_configured"] == "true"
+
+    # CRI unreachable on an active containerd: healthy, reason says so
+    comp.get_cri_version = lambda: None
+    cr = comp.check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "CRI is not enabled" in cr.reason
+
+
+@pytest.fixture()
+def mock_core(monkeypatch, tmp_path):
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    core = build_core(
+        Config(data_dir=str(tmp_path)),
+        in_memory_db=True,
+        kmsg_writable=False,
+        record_reboot=False,
+    )
+    yield core
+    core.close()
