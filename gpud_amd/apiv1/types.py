@@ -425,6 +425,7 @@ class MachineInfo:
     gpu_driver_version: str = ""
     cuda_version: str = ""  # carries ROCm version on AMD (wire-compat slot)
     container_runtime_version: str = ""
+    tailscale_version: str = ""  # reference: MachineInfo.TailscaleVersion
     kernel_version: str = ""
     os_image: str = ""
     operating_system: str = ""
@@ -449,6 +450,8 @@ class MachineInfo:
             d["cudaVersion"] = self.cuda_version
         if self.container_runtime_version:
             d["containerRuntimeVersion"] = self.container_runtime_version
+        if self.tailscale_version:
+            d["tailscaleVersion"] = self.tailscale_version
         if self.kernel_version:
             d["kernelVersion"] = self.kernel_version
         if self.os_image:

@@ -116,7 +116,10 @@ def find_mnt_with_retries(
 # lsblk (reference: pkg/disk/lsblk.go — JSON tree, flatten, fstype fallback)
 # ---------------------------------------------------------------------------
 
-_LSBLK_COLUMNS = "NAME,TYPE,SIZE,MOUNTPOINT,FSTYPE,PKNAME"
+_LSBLK_COLUMNS = (
+    "NAME,TYPE,SIZE,MOUNTPOINT,FSTYPE,PKNAME,ROTA,SERIAL,WWN,VENDOR,MODEL,"
+    "REV,FSUSED"
+)
 
 
 def _flatten_devices(devs: List[Dict], parent: str = "") -> List[Dict]:
