@@ -40,6 +40,29 @@ LINK_DOWN = 0
 # cannot blink the node healthy between polls
 FLAP_AUTO_CLEAR = datetime.timedelta(minutes=10)
 
+# per-product expected xGMI link counts on a fully-populated node
+# (reference: infiniband threshold_default.go per-product port defaults).
+# MI3xx OAM modules expose 7 point-to-point xGMI links on an 8-GPU mesh;
+# the default applies only when >1 GPU is visible (a single-GPU box
+# legitimately reports no peer links).
+PRODUCT_EXPECTED_LINKS = (
+    ("MI355", 7),
+    ("MI350", 7),
+    ("MI325", 7),
+    ("MI300", 7),
+)
+
+
+def expected_links_for_product(product: str, n_gpus: int) -> int:
+    """Default expected up-link count by product name; 0 = no expectation."""
+    if n_gpus < 2:
+        return 0
+    for marker, links in PRODUCT_EXPECTED_LINKS:
+        if marker.lower() in (product or "").lower():
+            # a partially populated node has fewer peers than links
+            return min(links, n_gpus - 1)
+    return 0
+
 
 class XGMIComponent(TickerComponent, SmiComponentMixin):
     def __init__(self, inst: GPUdInstance):
@@ -56,6 +79,19 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
         )
         self._last_states: Dict[str, List[int]] = {}
         self._last_traffic: Dict = {}
+        # SQLite link-state history: drop/flap evaluation survives daemon
+        # restarts (reference: infiniband store/store.go:57-319)
+        self.link_store = None
+        if inst.db_rw is not None and inst.db_ro is not None:
+            try:
+                from ...pkg.link_store import LinkStore
+
+                self.link_store = LinkStore(
+                    inst.db_rw, inst.db_ro, table_prefix="xgmi"
+                )
+            except Exception:
+                self.link_store = None
+        self.flap_auto_clear_seconds = FLAP_AUTO_CLEAR.total_seconds()
         self.get_snapshots: Callable = (
             self._shared.get if self._shared is not None else lambda: {}
         )
@@ -76,6 +112,25 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
     def events(self, since: datetime.datetime):
         return self._bucket.get(since) if self._bucket is not None else []
 
+    def can_set_healthy(self) -> bool:
+        return True
+
+    def set_healthy(self) -> None:
+        """Tombstone the link history: cleared drop/flap findings stay
+        cleared across restarts (reference: store tombstone semantics).
+        The event-bucket flap window is trimmed too so the auto-clear
+        window does not resurface the cleared finding."""
+        if self.link_store is not None:
+            self.link_store.set_tombstone()
+        if self._bucket is not None:
+            try:
+                import time as _time
+
+                self._bucket.purge(int(_time.time()) + 1)
+            except Exception:
+                pass
+        self._last_states = {}
+
     def check(self) -> CheckResult:
         guard = self.smi_guard()
         if guard is not None:
@@ -86,6 +141,12 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
         missing_links: List[str] = []
         extra: Dict[str, str] = {}
         any_links = False
+        expected = self.expected_links
+        if expected <= 0:
+            expected = expected_links_for_product(
+                getattr(self._smi, "product_name", "") or "", len(snaps)
+            )
+        store_rows: List[Dict] = []
         for uuid, snap in snaps.items():
             x = snap.get("xgmi_link_status")
             if x:
@@ -123,7 +184,15 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
                                 )
                             )
                 self._last_states[uuid] = states
-                if self.expected_links > 0 and up < self.expected_links:
+                store_rows.extend(
+                    {
+                        "device": uuid,
+                        "port": i,
+                        "state": "active" if s == LINK_UP else "down",
+                    }
+                    for i, s in enumerate(states)
+                )
+                if expected > 0 and up < expected:
                     missing_links.append(uuid)
             err = snap.get("xgmi_error_status")
             if err is not None and int(err) != 0:
@@ -184,6 +253,26 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
         except Exception:
             pass  # traffic counters are best-effort
 
+        # persist the sweep + evaluate drops/flaps from the SQLite history
+        # (reference: infiniband store — restart-surviving flap detection)
+        store_flaps = []
+        store_drops = []
+        if self.link_store is not None and store_rows:
+            try:
+                self.link_store.insert(store_rows)
+                ev = self.link_store.evaluate(
+                    drop_sticky_window=10 * 60.0,
+                    flap_auto_clear_window=self.flap_auto_clear_seconds,
+                )
+                store_drops = ev["drops"]
+                store_flaps = ev["flaps"]
+            except Exception:
+                pass
+        if store_drops:
+            extra["link_drops"] = "; ".join(e.reason for e in store_drops)
+        if store_flaps:
+            extra["link_flaps"] = "; ".join(e.reason for e in store_flaps)
+
         if down_by_uuid or err_by_uuid or missing_links:
             parts = []
             if down_by_uuid:
@@ -197,7 +286,7 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
                 )
             if missing_links:
                 parts.append(
-                    f"fewer than {self.expected_links} links up on: "
+                    f"fewer than {expected} links up on: "
                     + ", ".join(missing_links)
                 )
             return CheckResult(
@@ -212,6 +301,31 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
                         RepairActionType.HARDWARE_INSPECTION,
                     ],
                 ),
+            )
+        # store-evaluated drops/flaps: a persistent drop that has not
+        # stabilized, or a flapping link inside its auto-clear window,
+        # keeps the component unhealthy/degraded even though the links are
+        # up right now (reference: infiniband drop sticky window + flap
+        # auto-clear window)
+        if store_drops:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason="xGMI link drop history: "
+                + "; ".join(e.reason for e in store_drops),
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="persistently down xGMI link",
+                    repair_actions=[RepairActionType.HARDWARE_INSPECTION],
+                ),
+            )
+        if store_flaps:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.DEGRADED,
+                reason="xGMI link flap history: "
+                + "; ".join(e.reason for e in store_flaps),
+                extra_info=extra,
             )
         # flap auto-clear: recent down events keep the state Degraded even
         # after the link recovered (reference infiniband flap store)

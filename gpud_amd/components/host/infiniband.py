@@ -105,6 +105,27 @@ class InfinibandComponent(TickerComponent):
             self.sysfs_root
         )
         self._last_counters: Dict[str, Dict[str, int]] = {}
+        # SQLite ibports history (reference: infiniband/store/store.go —
+        # drop/flap evaluation survives daemon restarts)
+        self.link_store = None
+        if inst.db_rw is not None and inst.db_ro is not None:
+            try:
+                from ...pkg.link_store import LinkStore
+
+                self.link_store = LinkStore(
+                    inst.db_rw, inst.db_ro, table_prefix="ibports"
+                )
+            except Exception:
+                self.link_store = None
+        self.flap_auto_clear_seconds = FLAP_AUTO_CLEAR.total_seconds()
+
+    def can_set_healthy(self) -> bool:
+        return True
+
+    def set_healthy(self) -> None:
+        if self.link_store is not None:
+            self.link_store.set_tombstone()
+        self._last_counters = {}
 
     @property
     def name(self) -> str:
@@ -179,6 +200,35 @@ class InfinibandComponent(TickerComponent):
                     )
                 )
             self._last_counters[key] = dict(cur)
+        # persist the sweep; evaluate drops/flaps from the history store
+        store_drops, store_flaps = [], []
+        if self.link_store is not None and ports:
+            try:
+                self.link_store.insert(
+                    [
+                        {
+                            "device": p["device"],
+                            "port": int(str(p["port"]).strip() or 0),
+                            "state": "active" if p["active"] else "down",
+                            "rate_gb_sec": p["rate_gbps"] / 8.0,
+                            "total_link_downed": p["counters"].get(
+                                "link_downed", 0
+                            ),
+                        }
+                        for p in ports
+                    ]
+                )
+                ev = self.link_store.evaluate(
+                    drop_sticky_window=10 * 60.0,
+                    flap_auto_clear_window=self.flap_auto_clear_seconds,
+                )
+                store_drops, store_flaps = ev["drops"], ev["flaps"]
+            except Exception:
+                pass
+        if store_drops:
+            extra["port_drops"] = "; ".join(e.reason for e in store_drops)
+        if store_flaps:
+            extra["port_flaps"] = "; ".join(e.reason for e in store_flaps)
         if down or slow or (self.expected_ports > 0 and len(active) < self.expected_ports):
             parts = []
             if down:
@@ -200,6 +250,26 @@ class InfinibandComponent(TickerComponent):
                     description="degraded RDMA fabric ports",
                     repair_actions=[RepairActionType.HARDWARE_INSPECTION],
                 ),
+            )
+        if store_drops:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason="IB port drop history: "
+                + "; ".join(e.reason for e in store_drops),
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="persistently down IB port",
+                    repair_actions=[RepairActionType.HARDWARE_INSPECTION],
+                ),
+            )
+        if store_flaps:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.DEGRADED,
+                reason="IB port flap history: "
+                + "; ".join(e.reason for e in store_flaps),
+                extra_info=extra,
             )
         # auto-clear window for recent flaps (reference store semantics)
         if self._bucket is not None:
