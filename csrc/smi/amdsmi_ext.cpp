@@ -764,6 +764,11 @@ struct Snapshot {
   // throttle residency accumulators from gpu_metrics (violation analog)
   uint64_t thr_acc_counter = 0, thr_prochot = 0, thr_ppt = 0, thr_socket = 0,
            thr_vr = 0, thr_hbm = 0;
+  // per-XCC (per-XCD) instantaneous gfx busy from xcp_stats — the CDNA
+  // per-engine-cluster utilization breakdown (8 XCDs on MI355X); a sick
+  // XCD shows as an outlier against its siblings
+  uint32_t xcc_busy[AMDSMI_MAX_NUM_XCP * AMDSMI_MAX_NUM_XCC] = {0};
+  uint32_t n_xcc = 0;
 };
 
 // Static per-device values (temperature limits, power caps, max clocks, the
@@ -866,6 +871,15 @@ void take_snapshot(amdsmi_processor_handle h, Snapshot& s) {
     s.gfx_activity = gm.average_gfx_activity;
     s.umc_activity = gm.average_umc_activity;
     s.mm_activity = gm.average_mm_activity;
+    // per-XCC busy: xcp_stats gfx_busy_inst, UINT32_MAX = N/A sentinel
+    for (uint32_t p = 0; p < AMDSMI_MAX_NUM_XCP; ++p) {
+      for (uint32_t x = 0; x < AMDSMI_MAX_NUM_XCC; ++x) {
+        uint32_t v = gm.xcp_stats[p].gfx_busy_inst[x];
+        if (v == UINT32_MAX) continue;
+        if (s.n_xcc < AMDSMI_MAX_NUM_XCP * AMDSMI_MAX_NUM_XCC)
+          s.xcc_busy[s.n_xcc++] = v;
+      }
+    }
     s.ok_power = true;
     s.power_w = gm_valid16(gm.current_socket_power)
                     ? gm.current_socket_power
@@ -995,6 +1009,11 @@ py::dict snapshot_to_dict(const Snapshot& s) {
     a["gfx_activity_pct"] = s.gfx_activity;
     a["umc_activity_pct"] = s.umc_activity;
     a["mm_activity_pct"] = s.mm_activity;
+    if (s.n_xcc > 0) {
+      py::list xb;
+      for (uint32_t i = 0; i < s.n_xcc; ++i) xb.append(s.xcc_busy[i]);
+      a["xcc_busy_pct"] = xb;
+    }
     d["activity"] = a;
   }
   if (s.ok_vram) {

@@ -4,6 +4,17 @@ Reference: components/accelerator/nvidia/gpm (NVML GPM dual-sample deltas:
 SM occupancy, tensor/fp utilization — gpm/component.go:34-43). AMD-first:
 amdsmi gpu_metrics current clocks + engine activity plus per-process CU
 occupancy, which is the CDNA-native occupancy signal (256 CUs per MI355X).
+
+Per-pipe breakdown on CDNA4: amdsmi does NOT expose an MFMA-pipe
+utilization counter (gpu_metrics' ``mm_activity`` is MultiMedia, not
+matrix; MFMA busy cycles are rocprof PMCs — SQ_VALU_MFMA_BUSY_CYCLES —
+which need a profiling session, not a telemetry ioctl). The per-engine
+signals that DO exist and are surfaced here: per-XCC (per-XCD)
+instantaneous gfx busy from xcp_stats (8 XCDs per MI355X — an unhealthy
+XCD shows as an outlier against its siblings; exported per-XCC plus a
+max-min spread gauge) and the gfx/umc/mm activity set. The ACTIVE MFMA
+utilization measurement lives in the diag component's MFMA stress kernels
+(PMC-verified 97.5% MfmaUtil — profiles/diag_kernels_r1.txt).
 """
 
 from __future__ import annotations
@@ -59,6 +70,32 @@ class GPMComponent(TickerComponent, SmiComponentMixin):
                     "accelerator_amd_gpm_current_uclk_mhz",
                     "Instant memory clock from gpu_metrics",
                     float(gm.get("current_uclk_mhz", 0)),
+                    uuid=uuid,
+                )
+            act = snap.get("activity") or {}
+            xcc = act.get("xcc_busy_pct") or []
+            if xcc:
+                for i, busy in enumerate(xcc):
+                    self._gauges.set(
+                        "accelerator_amd_gpm_xcc_busy_percent",
+                        "Per-XCC (per-XCD) instantaneous gfx busy",
+                        float(busy),
+                        uuid=uuid,
+                        xcc=str(i),
+                    )
+                spread = max(xcc) - min(xcc)
+                self._gauges.set(
+                    "accelerator_amd_gpm_xcc_busy_spread_percent",
+                    "max-min spread of per-XCC busy (outlier XCD signal)",
+                    float(spread),
+                    uuid=uuid,
+                )
+                extra[f"{uuid}.xcc_busy"] = ",".join(str(b) for b in xcc)
+            if "mm_activity_pct" in act:
+                self._gauges.set(
+                    "accelerator_amd_gpm_mm_activity_percent",
+                    "MultiMedia engine activity (NOT matrix/MFMA)",
+                    float(act["mm_activity_pct"]),
                     uuid=uuid,
                 )
         # per-process CU occupancy (the SM-occupancy analog on 256-CU CDNA4);

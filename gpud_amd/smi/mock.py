@@ -208,6 +208,8 @@ class MockBackend:
             "gfx_activity_pct": s["gfx_activity"],
             "umc_activity_pct": s["umc_activity"],
             "mm_activity_pct": s["mm_activity"],
+            # per-XCC (per-XCD) busy from xcp_stats: 8 XCDs on MI355X
+            "xcc_busy_pct": [s["gfx_activity"]] * 8,
         }
 
     def vram_usage(self, i: int) -> Dict[str, Any]:

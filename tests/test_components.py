@@ -265,3 +265,22 @@ def test_check_duration_histogram_populated(mock_core):
     scraped = mock_core.metrics_scraper.scrape()
     names = {m.name for m in scraped}
     assert "gpud_component_check_duration_seconds_count" in names
+
+
+def test_gpm_per_xcc_busy(mock_core):
+    """gpm exports the per-XCC (per-XCD) busy breakdown and its spread —
+    the CDNA per-pipe utilization analog (amdsmi has no
+    MFMA-pipe counter; xcp_stats per-XCD busy is the real signal)."""
+    comp = mock_core.registry.get("accelerator-amd-gpm")
+    cr = comp.trigger_check()
+    assert cr.health == "Healthy"
+    xcc_keys = [k for k in (cr.extra_info or {}) if k.endswith(".xcc_busy")]
+    assert xcc_keys, cr.extra_info
+    # the mock reports 8 XCDs per GPU
+    assert len((cr.extra_info[xcc_keys[0]]).split(",")) == 8
+    from prometheus_client import generate_latest
+
+    text = generate_latest(mock_core.metrics_registry).decode()
+    assert "accelerator_amd_gpm_xcc_busy_percent" in text
+    assert "accelerator_amd_gpm_xcc_busy_spread_percent" in text
+    assert "accelerator_amd_gpm_mm_activity_percent" in text
