@@ -302,6 +302,58 @@ def create_app(core: DaemonCore, plugin_specs: Optional[list] = None) -> FastAPI
             out[names.get(tid, str(tid))] = traceback.format_stack(frame)
         return _negotiate(request, out)
 
+    @app.get("/admin/pprof/profile")
+    def pprof_profile(request: Request, seconds: float = 5.0):
+        """CPU profile analog of the reference's /admin/pprof/profile
+        (pkg/server/server.go:435-440): a statistical sampler over
+        sys._current_frames() — samples every thread's stack at 10 ms for
+        the requested window and reports functions by inclusive sample
+        count (py-spy-style, no instrumentation overhead between samples)."""
+        import collections
+        import sys
+        import time as _time
+
+        seconds = max(0.1, min(seconds, 30.0))
+        interval = 0.01
+        own = {__import__("threading").get_ident()}
+        counts: "collections.Counter[str]" = collections.Counter()
+        leaf_counts: "collections.Counter[str]" = collections.Counter()
+        n_samples = 0
+        deadline = _time.monotonic() + seconds
+        while _time.monotonic() < deadline:
+            for tid, frame in sys._current_frames().items():
+                if tid in own:
+                    continue
+                f = frame
+                leaf = True
+                while f is not None:
+                    key = (
+                        f"{f.f_code.co_filename.rsplit('/', 1)[-1]}:"
+                        f"{f.f_code.co_name}"
+                    )
+                    counts[key] += 1
+                    if leaf:
+                        leaf_counts[key] += 1
+                        leaf = False
+                    f = f.f_back
+            n_samples += 1
+            _time.sleep(interval)
+        return _negotiate(
+            request,
+            {
+                "seconds": seconds,
+                "samples": n_samples,
+                "top_inclusive": [
+                    {"func": k, "samples": v}
+                    for k, v in counts.most_common(50)
+                ],
+                "top_self": [
+                    {"func": k, "samples": v}
+                    for k, v in leaf_counts.most_common(50)
+                ],
+            },
+        )
+
     @app.get("/admin/pprof/heap")
     def pprof_heap(request: Request):
         import tracemalloc

@@ -212,6 +212,20 @@ def test_admin_endpoints(daemon):
     assert any("gpud" in k or "MainThread" in k for k in r.json())
     r = httpx.get(server.base_url + "/admin/packages", verify=False)
     assert r.status_code == 200
+    # CPU-profile analog (reference: /admin/pprof/profile): sampled stacks
+    r = httpx.get(
+        server.base_url + "/admin/pprof/profile?seconds=0.3",
+        verify=False, timeout=30,
+    )
+    assert r.status_code == 200
+    prof = r.json()
+    assert prof["samples"] > 0
+    assert isinstance(prof["top_inclusive"], list)
+    # swagger UI route + openapi schema (reference: /swagger/*any)
+    r = httpx.get(server.base_url + "/swagger", verify=False)
+    assert r.status_code == 200 and "swagger" in r.text.lower()
+    r = httpx.get(server.base_url + "/openapi.json", verify=False)
+    assert r.status_code == 200 and "/v1/states" in r.text
 
 
 def test_malformed_requests_never_500(daemon):
