@@ -723,6 +723,73 @@ py::dict cper_entries(int index, uint32_t severity_mask, uint64_t cursor,
         default: break;
       }
       if (nt != nullptr) e["notify_type"] = std::string(nt);
+      // Section descriptors (UEFI spec appendix N: 72-byte descriptors
+      // following the 128-byte record header) — decoded for FRU-level
+      // attribution: section type, section severity, FRU id/text when the
+      // descriptor's validation bits say they are present.
+      {
+        const char* rec = reinterpret_cast<const char*>(hd);
+        const char* buf_end = data.data() + data.size();
+        const uint32_t rec_len = hd->record_length;
+        const size_t hdr_sz = sizeof(amdsmi_cper_hdr_t);
+        constexpr size_t kDescSz = 72;
+        py::list sections;
+        for (uint16_t sidx = 0; sidx < hd->sec_cnt && sidx < 16; ++sidx) {
+          const char* desc = rec + hdr_sz + (size_t)sidx * kDescSz;
+          if (desc + kDescSz > buf_end) break;
+          if (rec_len && desc + kDescSz > rec + rec_len) break;
+          py::dict s;
+          uint32_t soff = 0, slen = 0, ssev = 0;
+          std::memcpy(&soff, desc + 0, 4);
+          std::memcpy(&slen, desc + 4, 4);
+          std::memcpy(&ssev, desc + 48, 4);
+          uint8_t valid = *(const uint8_t*)(desc + 10);
+          s["offset"] = soff;
+          s["length"] = slen;
+          s["severity"] = ssev;
+          // section type GUID (mixed-endian text form)
+          const uint8_t* g = (const uint8_t*)(desc + 16);
+          char gs[40];
+          std::snprintf(gs, sizeof(gs),
+                        "%02x%02x%02x%02x-%02x%02x-%02x%02x-%02x%02x-"
+                        "%02x%02x%02x%02x%02x%02x",
+                        g[3], g[2], g[1], g[0], g[5], g[4], g[7], g[6],
+                        g[8], g[9], g[10], g[11], g[12], g[13], g[14],
+                        g[15]);
+          std::string guid(gs);
+          s["type_guid"] = guid;
+          // well-known UEFI CPER section types
+          const char* tname = nullptr;
+          if (guid == "a5bc1114-6f64-4ede-b863-3e83ed7c83b1")
+            tname = "memory";
+          else if (guid == "d995e954-bbc1-430f-ad91-b44dcb3c6f35")
+            tname = "pcie";
+          else if (guid == "9876ccad-47b4-4bdb-b65e-16f193c4f3db")
+            tname = "processor_generic";
+          else if (guid == "81212a96-09ed-4996-9471-8d729c8e69ed")
+            tname = "firmware";
+          else if (guid == "5b51fef7-c79d-4434-8f1b-aa62de3e2c64")
+            tname = "dmar_generic";
+          if (tname != nullptr) s["type_name"] = std::string(tname);
+          if (valid & 0x1) {  // FRU id GUID valid
+            const uint8_t* f = (const uint8_t*)(desc + 32);
+            char fs[40];
+            std::snprintf(fs, sizeof(fs),
+                          "%02x%02x%02x%02x-%02x%02x-%02x%02x-%02x%02x-"
+                          "%02x%02x%02x%02x%02x%02x",
+                          f[3], f[2], f[1], f[0], f[5], f[4], f[7], f[6],
+                          f[8], f[9], f[10], f[11], f[12], f[13], f[14],
+                          f[15]);
+            s["fru_id"] = std::string(fs);
+          }
+          if (valid & 0x2) {  // FRU text valid (20-byte ASCII)
+            s["fru_text"] =
+                std::string(desc + 52, strnlen(desc + 52, 20));
+          }
+          sections.append(s);
+        }
+        if (py::len(sections) > 0) e["sections"] = sections;
+      }
       entries.append(e);
     }
     if (rc != AMDSMI_STATUS_MORE_DATA) break;

@@ -152,6 +152,22 @@ class CPERComponent(TickerComponent, SmiComponentMixin):
                         else EventType.INFO
                     )
                     nt = e.get("notify_type", "?")
+                    # FRU-level attribution from the decoded section
+                    # descriptors (UEFI CPER appendix N): FRU text names
+                    # the failing replaceable unit, section type names the
+                    # error class
+                    sec_bits = []
+                    for s in e.get("sections", []) or []:
+                        label = s.get("type_name") or (
+                            (s.get("type_guid") or "?")[:8]
+                        )
+                        fru = s.get("fru_text") or s.get("fru_id") or ""
+                        sec_bits.append(
+                            f"{label}" + (f" fru={fru}" if fru else "")
+                        )
+                    sec_txt = (
+                        " [" + "; ".join(sec_bits) + "]" if sec_bits else ""
+                    )
                     self._bucket.insert(
                         Event(
                             time=now,
@@ -163,6 +179,7 @@ class CPERComponent(TickerComponent, SmiComponentMixin):
                                 f"{uuid}: notify={nt} "
                                 f"record_id={uuid}:{rid or '?'} "
                                 f"sections={e.get('section_count', 0)}"
+                                f"{sec_txt}"
                             ),
                         )
                     )

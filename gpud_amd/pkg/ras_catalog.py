@@ -488,6 +488,285 @@ CATALOG: List[Detail] = [
         _REBOOT,
         critical=True,
     ),
+    # ---- round-2 expansion: amdgpu/KFD lifecycle depth ---------------------
+    _d(
+        "amdgpu_mode1_reset",
+        r"amdgpu.*GPU mode1 reset",
+        "Whole-ASIC mode-1 reset — the heaviest reset path (full chip, all "
+        "XCDs); every context on the device was lost",
+        EventType.CRITICAL,
+        _APP,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_atombios_hang",
+        r"amdgpu.*atombios stuck in loop",
+        "AtomBIOS command-table execution hung — VBIOS-level wedge; the "
+        "device usually needs a reset and the board firmware is suspect",
+        EventType.FATAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_flr_notification",
+        r"amdgpu.*FLR notification",
+        "SR-IOV function-level-reset notification from the host — the "
+        "hypervisor reset this virtual function; running work was lost",
+        EventType.CRITICAL,
+        _APP,
+    ),
+    _d(
+        "amdgpu_bo_va_update_failed",
+        r"amdgpu.*Couldn't update BO_VA",
+        "GPU VM mapping update failed (usually -ENOMEM under memory "
+        "pressure) — the submitting process sees submission errors",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "amdgpu_evict_resources_failed",
+        r"amdgpu.*evicting device resources failed",
+        "Device resource eviction failed during suspend/reset preparation — "
+        "the following reset/resume is suspect",
+        EventType.WARNING,
+        _REBOOT,
+    ),
+    _d(
+        "amdgpu_deferred_error",
+        r"amdgpu.*deferred (?:hardware )?error",
+        "RAS deferred (latent) hardware error logged — not yet consumed; "
+        "pages will be poisoned on touch. Watch for poison-consumption "
+        "events and bad-page growth",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "kfd_migrate_failed",
+        r"(?:kfd|amdgpu).*fail(?:ed)? to migrate",
+        "SVM range migration between VRAM and host memory failed — the "
+        "compute process may stall or abort",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "kfd_restore_queues_failed",
+        r"(?:kfd|amdgpu).*[Ff]ailed to restore queues",
+        "KFD could not restore a process's queues after eviction — the "
+        "process's GPU work is stopped",
+        EventType.CRITICAL,
+        _APP,
+        critical=True,
+    ),
+    _d(
+        "kfd_reset_wavefronts",
+        r"(?:kfd|amdgpu).*[Rr]esetting wave fronts",
+        "KFD reset in-flight wavefronts (hang recovery on a compute queue) "
+        "— the owning process's kernels were killed",
+        EventType.CRITICAL,
+        _APP,
+        critical=True,
+    ),
+    _d(
+        "kfd_unmap_queue_failed",
+        r"(?:kfd|amdgpu).*[Ff]ailed to unmap (?:legacy )?queue",
+        "KFD queue unmap failed — frequently the precursor of a HWS hang "
+        "and a following GPU reset",
+        EventType.CRITICAL,
+        _APP,
+        critical=True,
+    ),
+    # ---- host kernel-integrity signatures (Xid-table host analogs) ---------
+    _d(
+        "host_kernel_panic",
+        r"Kernel panic - not syncing",
+        "Kernel panic on the current boot ring (pstore carries prior-boot "
+        "panics) — node integrity lost",
+        EventType.FATAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "host_kernel_bug",
+        r"BUG: (?:unable to handle|kernel NULL pointer)",
+        "Kernel page-fault BUG (NULL dereference / unhandled fault) — a "
+        "kernel code path crashed; node stability is compromised",
+        EventType.FATAL,
+        _REBOOT,
+        critical=True,
+    ),
+    _d(
+        "host_kernel_oops",
+        r"Oops: [0-9a-f]{4}",
+        "Kernel oops logged — a kernel thread died; taint follows and later "
+        "failures are suspect until reboot",
+        EventType.CRITICAL,
+        _REBOOT,
+    ),
+    _d(
+        "host_rcu_stall",
+        r"rcu.*detected (?:stalls|expedited stalls)",
+        "RCU stall — a CPU stopped responding to the RCU state machine for "
+        "seconds; driver wedges and hard IRQ storms look like this",
+        EventType.CRITICAL,
+        _REBOOT,
+    ),
+    _d(
+        "host_hard_lockup",
+        r"Watchdog detected hard LOCKUP",
+        "NMI watchdog hard lockup — a CPU stopped servicing interrupts; "
+        "usually firmware/hardware level trouble",
+        EventType.FATAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "host_apei_hardware_error",
+        r"\{\d+\}\[Hardware Error\]",
+        "APEI/GHES firmware-reported hardware error (the kernel-side print "
+        "of a CPER record) — correlate with the cper component's records",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_memory_failure",
+        r"Memory failure: 0x",
+        "Kernel memory-failure handling (hwpoison) ran on a page — ECC "
+        "uncorrectable host memory; track recurrence per DIMM via EDAC",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_list_corruption",
+        r"list_(?:del|add) corruption",
+        "Kernel list corruption detected — memory corruption inside the "
+        "kernel; frequently a driver bug or failing DIMM",
+        EventType.FATAL,
+        _REBOOT + _HW,
+        critical=True,
+    ),
+    _d(
+        "host_irq_nobody_cared",
+        r"irq \d+: nobody cared",
+        "An IRQ line fired with no handler claiming it — the kernel "
+        "disabled the line; devices behind it stop interrupting",
+        EventType.WARNING,
+        _REBOOT,
+    ),
+    _d(
+        "host_page_alloc_failure",
+        r"page allocation failure",
+        "Kernel page allocation failure (fragmentation/pressure) — DMA "
+        "buffer allocations may be failing",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "host_cpu_thermal_throttle",
+        r"[Cc]ore temperature above threshold",
+        "Host CPU thermal throttling engaged — check chassis cooling; GPU "
+        "thermals on the same node are suspect",
+        EventType.WARNING,
+        _HW,
+    ),
+    # ---- storage / filesystem / network health -----------------------------
+    _d(
+        "host_io_error",
+        r"I/O error, dev",
+        "Block-layer I/O error — failing disk or transport; check the "
+        "device's SMART state",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_filesystem_error",
+        r"(?:EXT4-fs error|XFS \([^)]*\): (?:Internal error|Corruption)|"
+        r"BTRFS error)",
+        "Filesystem-level error/corruption — data integrity at risk; fsck "
+        "and underlying device inspection needed",
+        EventType.CRITICAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "host_filesystem_readonly",
+        r"Remounting filesystem read-only",
+        "A filesystem remounted itself read-only after errors — writes are "
+        "failing node-wide on that mount",
+        EventType.FATAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "nvme_io_timeout",
+        r"nvme nvme\d+: (?:I/O .*timeout|controller is down|Device not ready)",
+        "NVMe command timeout / controller failure — local storage is "
+        "degraded or lost",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_nfs_not_responding",
+        r"nfs: server .* not responding",
+        "NFS server stopped responding — mounts hang and D-state process "
+        "counts rise (see the os component's tracker)",
+        EventType.WARNING,
+        _APP,
+    ),
+    _d(
+        "host_netdev_watchdog",
+        r"NETDEV WATCHDOG",
+        "Network transmit queue timeout — NIC or driver wedge on a host "
+        "interface",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    # ---- IOMMU / PCIe depth -------------------------------------------------
+    _d(
+        "iommu_io_page_fault",
+        r"AMD-Vi.*IO_PAGE_FAULT",
+        "IOMMU DMA page fault — a device (often a GPU after a bad mapping "
+        "or reset) performed DMA to an unmapped address",
+        EventType.CRITICAL,
+        _APP,
+    ),
+    _d(
+        "pcie_dpc_containment",
+        r"DPC: containment event",
+        "PCIe Downstream Port Containment fired — the link was cut to "
+        "contain an uncorrected error; devices below dropped off",
+        EventType.FATAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "pcie_link_down",
+        r"pciehp.*Link Down",
+        "PCIe hotplug reported Link Down — a device (possibly a GPU) left "
+        "the bus",
+        EventType.CRITICAL,
+        _HW,
+        critical=True,
+    ),
+    # ---- ROCm user-space crash signatures ----------------------------------
+    _d(
+        "amd_hip_segfault_in_libamdhip",
+        r"segfault at .* in libamdhip64",
+        "Process crashed inside the HIP runtime — correlate with preceding "
+        "amdgpu/KFD events; repeated crashes across processes implicate the "
+        "node, a single app implicates the app",
+        EventType.CRITICAL,
+        _APP,
+    ),
+    _d(
+        "amd_rocm_lib_segfault",
+        r"segfault at .* in (?:librocblas|libhipblaslt|libMIOpen|"
+        r"librocsolver|librocfft)",
+        "Process crashed inside a ROCm math library — usually an "
+        "application/library-version issue; node-wide recurrence points at "
+        "the GPU or driver",
+        EventType.WARNING,
+        _APP,
+    ),
 ]
 
 _CATALOG_BY_NAME: Dict[str, Detail] = {d.name: d for d in CATALOG}
@@ -566,6 +845,25 @@ INJECTABLE: Dict[str, str] = {
     ),
     "amdgpu_fallen_off_bus": (
         "amdgpu 0000:0a:00.0: amdgpu: GPU has fallen off the bus"
+    ),
+    "amdgpu_deferred_error": (
+        "amdgpu 0000:0a:00.0: amdgpu: 3 deferred hardware errors detected "
+        "in UMC block"
+    ),
+    "iommu_io_page_fault": (
+        "AMD-Vi: Event logged [IO_PAGE_FAULT domain=0x0035 "
+        "address=0x7f8100000000 flags=0x0070]"
+    ),
+    "pcie_dpc_containment": (
+        "pcieport 0000:00:01.1: DPC: containment event, status:0x1f01 "
+        "source:0x0000"
+    ),
+    "host_rcu_stall": (
+        "rcu: INFO: rcu_sched detected stalls on CPUs/tasks: { 12-.... } "
+        "(detected by 0, t=60002 jiffies)"
+    ),
+    "kfd_reset_wavefronts": (
+        "kfd kfd: amdgpu: Resetting wave fronts on dev 0xb3c"
     ),
 }
 

@@ -570,6 +570,75 @@ def test_every_catalog_entry_matches_a_representative_line():
             "kfd kfd: amdgpu: Failed to create process VM object",
         "host_swiotlb_full":
             "sdhci: swiotlb buffer is full (sz: 262144 bytes)",
+        # ---- round-2 expansion ------------------------------------------
+        "amdgpu_mode1_reset":
+            "amdgpu 0000:0a:00.0: amdgpu: GPU mode1 reset",
+        "amdgpu_atombios_hang":
+            "[drm:amdgpu_atom_execute_table [amdgpu]] *ERROR* atombios "
+            "stuck in loop for more than 20secs aborting",
+        "amdgpu_flr_notification":
+            "amdgpu 0000:00:07.0: amdgpu: Got AMDGPU_HOST_FLR notification",
+        "amdgpu_bo_va_update_failed":
+            "[drm:amdgpu_gem_va_ioctl [amdgpu]] *ERROR* Couldn't update "
+            "BO_VA (-12)",
+        "amdgpu_evict_resources_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: evicting device resources failed",
+        "kfd_migrate_failed":
+            "kfd kfd: amdgpu: qcm fence wait loop timeout; failed to "
+            "migrate svm range",
+        "kfd_restore_queues_failed":
+            "kfd kfd: amdgpu: Failed to restore queues of pasid 0x8002",
+        "kfd_unmap_queue_failed":
+            "kfd kfd: amdgpu: Failed to unmap legacy queue, mqd gone",
+        "host_kernel_panic":
+            "Kernel panic - not syncing: Fatal exception in interrupt",
+        "host_kernel_bug":
+            "BUG: kernel NULL pointer dereference, address: "
+            "0000000000000008",
+        "host_kernel_oops":
+            "Oops: 0002 [#1] PREEMPT SMP NOPTI",
+        "host_hard_lockup":
+            "NMI watchdog: Watchdog detected hard LOCKUP on cpu 3",
+        "host_apei_hardware_error":
+            "{1}[Hardware Error]: Hardware error from APEI Generic "
+            "Hardware Error Source: 1",
+        "host_memory_failure":
+            "Memory failure: 0x3c5e00: recovery action for dirty LRU page: "
+            "Recovered",
+        "host_list_corruption":
+            "list_del corruption, ffff88810deadbe0->next is LIST_POISON1",
+        "host_irq_nobody_cared":
+            'irq 16: nobody cared (try booting with the "irqpoll" option)',
+        "host_page_alloc_failure":
+            "python: page allocation failure: order:5, "
+            "mode:0x40cc0(GFP_KERNEL|__GFP_COMP)",
+        "host_cpu_thermal_throttle":
+            "CPU12: Core temperature above threshold, cpu clock throttled",
+        "host_io_error":
+            "blk_update_request: I/O error, dev sda, sector 123456 op "
+            "0x0:(READ)",
+        "host_filesystem_error":
+            "EXT4-fs error (device nvme0n1p2): ext4_lookup:1855: inode "
+            "#1234: comm python: deleted inode referenced",
+        "host_filesystem_readonly":
+            "Aborting journal on device nvme0n1p2-8. Remounting filesystem "
+            "read-only",
+        "nvme_io_timeout":
+            "nvme nvme0: I/O 123 QID 4 timeout, aborting",
+        "host_nfs_not_responding":
+            "nfs: server 10.0.0.5 not responding, timed out",
+        "host_netdev_watchdog":
+            "NETDEV WATCHDOG: eth0 (mlx5_core): transmit queue 5 timed out",
+        "pcie_link_down":
+            "pcieport 0000:00:01.1: pciehp: Slot(0): Link Down",
+        "amd_hip_segfault_in_libamdhip":
+            "python[4242]: segfault at 0 ip 00007f1234567890 sp "
+            "00007ffc12345678 error 4 in libamdhip64.so.7[7f1230000000+"
+            "400000]",
+        "amd_rocm_lib_segfault":
+            "python[4242]: segfault at 10 ip 00007f1234567890 sp "
+            "00007ffc12345678 error 4 in librocblas.so.4[7f1200000000+"
+            "8000000]",
     })
     missing = [d.name for d in CATALOG if d.name not in representatives]
     assert not missing, f"entries without representative lines: {missing}"

@@ -625,3 +625,39 @@ def test_diag_components_report_unhealthy_without_extension(monkeypatch, tmp_pat
         # kernels: missing extension is named; fabric: the real binary runs
         # and reports the missing device — loud either way
         assert "extension" in blob or "not built" in blob or "fabric" in blob
+
+
+def test_cper_section_fru_attribution(mock_core):
+    """Decoded CPER section descriptors carry section type + FRU text into
+    the event message (FRU-level attribution)."""
+    backend = mock_core.smi_instance._b
+    comp = mock_core.registry.get("accelerator-amd-cper")
+    comp.trigger_check()
+    backend.state[0]["cper"].append(
+        {
+            "severity": 1,
+            "severity_name": "fatal",
+            "record_id": "bb02",
+            "notify_type": "MCE",
+            "section_count": 2,
+            "sections": [
+                {
+                    "type_guid": "a5bc1114-6f64-4ede-b863-3e83ed7c83b1",
+                    "type_name": "memory",
+                    "severity": 1,
+                    "fru_text": "OAM3 HBM stack 2",
+                },
+                {
+                    "type_guid": "d995e954-bbc1-430f-ad91-b44dcb3c6f35",
+                    "type_name": "pcie",
+                    "severity": 0,
+                },
+            ],
+        }
+    )
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    evs = comp.events(utcnow() - datetime.timedelta(minutes=5))
+    msg = next(e.message for e in evs if "bb02" in e.message)
+    assert "memory fru=OAM3 HBM stack 2" in msg
+    assert "pcie" in msg
