@@ -20,6 +20,7 @@
 #include <amd_smi/amdsmi.h>
 
 #include <chrono>
+#include <cstdlib>
 #include <cstring>
 #include <map>
 #include <mutex>
@@ -61,6 +62,28 @@ std::mutex g_mu;
 // never blocks the interpreter; per-cycle cost is nil (one caller
 // dominates each poll cycle via the shared snapshot).
 std::mutex g_call_mu;
+
+// GPUD_AMDSMI_NO_CALL_MUTEX=1 disables the serialization ABOVE the library
+// — an instrumentation knob for the ASan/TSan race hunt (docs/ROADMAP.md
+// "Observed-but-unresolved" heap corruption): running the concurrency
+// stress with the mutex off under ASan distinguishes a race in THIS
+// binding / libamd_smi from corruption elsewhere. Never set in production.
+bool call_mutex_disabled() {
+  static const bool disabled = [] {
+    const char* v = std::getenv("GPUD_AMDSMI_NO_CALL_MUTEX");
+    return v != nullptr && v[0] == '1';
+  }();
+  return disabled;
+}
+
+// lock_guard that honors the kill switch
+struct CallLock {
+  std::unique_lock<std::mutex> lk;
+  explicit CallLock(std::mutex& m) {
+    if (!call_mutex_disabled()) lk = std::unique_lock<std::mutex>(m);
+  }
+};
+
 bool g_initialized = false;
 std::vector<amdsmi_processor_handle> g_handles;
 
@@ -90,7 +113,7 @@ void enumerate_locked() {
 
 void smi_init() {
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   std::lock_guard<std::mutex> lk(g_mu);
   if (g_initialized) return;
   check(amdsmi_init(AMDSMI_INIT_AMD_GPUS), "amdsmi_init");
@@ -100,7 +123,7 @@ void smi_init() {
 
 void smi_shutdown() {
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   std::lock_guard<std::mutex> lk(g_mu);
   if (!g_initialized) return;
   amdsmi_shut_down();
@@ -130,7 +153,7 @@ std::string device_uuid(int index) {
   unsigned int len = AMDSMI_MAX_STRING_LENGTH;
   char buf[AMDSMI_MAX_STRING_LENGTH] = {0};
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   check(amdsmi_get_gpu_device_uuid(h, &len, buf), "amdsmi_get_gpu_device_uuid");
   return std::string(buf);
 }
@@ -140,7 +163,7 @@ std::string device_bdf(int index) {
   uint64_t bdfid = 0;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_bdf_id(h, &bdfid), "amdsmi_get_gpu_bdf_id");
   }
   // bdfid packs: [63:32] domain, [15:8] bus, [7:3] device, [2:0] function
@@ -159,7 +182,7 @@ py::dict asic_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_asic_info(h, &info), "amdsmi_get_gpu_asic_info");
   }
   py::dict d;
@@ -180,7 +203,7 @@ py::dict board_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_board_info(h, &info), "amdsmi_get_gpu_board_info");
   }
   py::dict d;
@@ -198,7 +221,7 @@ py::dict driver_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_driver_info(h, &info), "amdsmi_get_gpu_driver_info");
   }
   py::dict d;
@@ -214,7 +237,7 @@ py::dict vbios_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_vbios_info(h, &info), "amdsmi_get_gpu_vbios_info");
   }
   py::dict d;
@@ -231,7 +254,7 @@ py::dict vram_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_vram_info(h, &info), "amdsmi_get_gpu_vram_info");
   }
   py::dict d;
@@ -251,7 +274,7 @@ int64_t temp_metric(int index, int sensor_type, int metric) {
   auto h = handle_at(index);
   int64_t v = 0;
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   check(amdsmi_get_temp_metric(
             h, static_cast<amdsmi_temperature_type_t>(sensor_type),
             static_cast<amdsmi_temperature_metric_t>(metric), &v),
@@ -268,7 +291,7 @@ py::dict power_info(int index) {
   amdsmi_status_t st_cap;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_power_info(h, &info), "amdsmi_get_power_info");
     st_cap = amdsmi_get_power_cap_info(h, 0, &cap);
   }
@@ -293,7 +316,7 @@ py::dict clock_info(int index, int clk_type) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_clock_info(h, static_cast<amdsmi_clk_type_t>(clk_type),
                                 &info),
           "amdsmi_get_clock_info");
@@ -313,7 +336,7 @@ py::dict activity(int index) {
   std::memset(&u, 0, sizeof(u));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_activity(h, &u), "amdsmi_get_gpu_activity");
   }
   py::dict d;
@@ -329,7 +352,7 @@ py::dict vram_usage(int index) {
   std::memset(&u, 0, sizeof(u));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_vram_usage(h, &u), "amdsmi_get_gpu_vram_usage");
   }
   py::dict d;
@@ -344,7 +367,7 @@ py::dict ecc_count_total(int index) {
   std::memset(&ec, 0, sizeof(ec));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_total_ecc_count(h, &ec),
           "amdsmi_get_gpu_total_ecc_count");
   }
@@ -361,7 +384,7 @@ py::dict ecc_count_block(int index, uint64_t block) {
   std::memset(&ec, 0, sizeof(ec));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_ecc_count(h, static_cast<amdsmi_gpu_block_t>(block),
                                    &ec),
           "amdsmi_get_gpu_ecc_count");
@@ -379,7 +402,7 @@ py::dict bad_page_info(int index) {
   amdsmi_status_t st;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     st = amdsmi_get_gpu_bad_page_info(h, &num, nullptr);
   }
   py::dict d;
@@ -387,7 +410,7 @@ py::dict bad_page_info(int index) {
   std::vector<amdsmi_retired_page_record_t> recs(num);
   if (num > 0) {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_bad_page_info(h, &num, recs.data()),
           "amdsmi_get_gpu_bad_page_info(records)");
   }
@@ -403,7 +426,7 @@ py::dict bad_page_info(int index) {
   amdsmi_status_t st_thr;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     st_thr = amdsmi_get_gpu_bad_page_threshold(h, &threshold);
   }
   d["total"] = num;
@@ -420,7 +443,7 @@ py::list process_list(int index) {
   amdsmi_status_t st;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     st = amdsmi_get_gpu_process_list(h, &n, nullptr);
   }
   py::list out;
@@ -431,7 +454,7 @@ py::list process_list(int index) {
   std::memset(procs.data(), 0, sizeof(amdsmi_proc_info_t) * n);
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_process_list(h, &n, procs.data()),
           "amdsmi_get_gpu_process_list");
   }
@@ -455,7 +478,7 @@ py::dict violation_status(int index) {
   std::memset(&v, 0, sizeof(v));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_violation_status(h, &v), "amdsmi_get_violation_status");
   }
   py::dict d;
@@ -484,7 +507,7 @@ py::dict xgmi_link_status(int index) {
   std::memset(&s, 0, sizeof(s));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_gpu_xgmi_link_status(h, &s),
           "amdsmi_get_gpu_xgmi_link_status");
   }
@@ -501,7 +524,7 @@ int xgmi_error_status(int index) {
   auto h = handle_at(index);
   amdsmi_xgmi_status_t st;
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   check(amdsmi_gpu_xgmi_error_status(h, &st), "amdsmi_gpu_xgmi_error_status");
   return static_cast<int>(st);
 }
@@ -512,7 +535,7 @@ py::dict xgmi_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_xgmi_info(h, &info), "amdsmi_get_xgmi_info");
   }
   py::dict d;
@@ -529,7 +552,7 @@ py::dict link_metrics(int index) {
   std::memset(&lm, 0, sizeof(lm));
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_link_metrics(h, &lm), "amdsmi_get_link_metrics");
   }
   py::dict d;
@@ -562,7 +585,7 @@ py::dict energy_count(int index) {
   float res = 0.f;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     check(amdsmi_get_energy_count(h, &acc, &res, &ts),
           "amdsmi_get_energy_count");
   }
@@ -582,7 +605,7 @@ py::dict pcie_info(int index) {
   std::memset(&info, 0, sizeof(info));
   {
     py::gil_scoped_release nogil;
-    std::lock_guard<std::mutex> call_lk(g_call_mu);
+    CallLock call_lk(g_call_mu);
     check(amdsmi_get_pcie_info(h, &info), "amdsmi_get_pcie_info");
   }
   py::dict d;
@@ -615,7 +638,7 @@ py::dict partition_info(int index) {
   char buf[64] = {0};
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     if (amdsmi_get_gpu_compute_partition(h, buf, sizeof(buf)) ==
         AMDSMI_STATUS_SUCCESS) {
       // re-acquire handled after block
@@ -627,7 +650,7 @@ py::dict partition_info(int index) {
   char mbuf[64] = {0};
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     if (amdsmi_get_gpu_memory_partition(h, mbuf, sizeof(mbuf)) !=
         AMDSMI_STATUS_SUCCESS)
       mbuf[0] = '\0';
@@ -639,7 +662,7 @@ py::dict partition_info(int index) {
   amdsmi_status_t prc;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     prc = amdsmi_get_gpu_accelerator_partition_profile(h, &prof, part_ids);
   }
   if (prc == AMDSMI_STATUS_SUCCESS) {
@@ -672,7 +695,7 @@ py::dict cper_entries(int index, uint32_t severity_mask, uint64_t cursor,
     amdsmi_status_t rc;
     {
       py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
       rc = amdsmi_get_gpu_cper_entries(h, severity_mask, data.data(),
                                        &buf_size, hdrs.data(), &entry_count,
                                        &cursor);
@@ -1144,7 +1167,7 @@ bool power_management_enabled(int index) {
   auto h = handle_at(index);
   bool enabled = false;
   py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
   check(amdsmi_is_gpu_power_management_enabled(h, &enabled),
         "amdsmi_is_gpu_power_management_enabled");
   return enabled;
@@ -1155,7 +1178,7 @@ py::dict metrics_snapshot(int index) {
   Snapshot s;
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     take_snapshot(h, s);
   }
   return snapshot_to_dict(s);
@@ -1230,7 +1253,7 @@ py::list metrics_snapshot_all() {
   std::vector<Snapshot> snaps(handles.size());
   {
     py::gil_scoped_release nogil;
-  std::lock_guard<std::mutex> call_lk(g_call_mu);
+  CallLock call_lk(g_call_mu);
     if (handles.size() <= 1) {
       for (size_t i = 0; i < handles.size(); ++i)
         take_snapshot(handles[i], snaps[i]);
