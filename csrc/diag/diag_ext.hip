@@ -987,6 +987,164 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v3_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// GEMM v6: counted-wait chunk pipeline — the v2 glds structure with the
+// K-tile boundary vmcnt(0) drain REMOVED. The 64 KiB tile is staged as 4
+// chunks of 16 slots (C0 = B first-halves, C1 = A rows 0-63/128-191,
+// C2 = B second-halves, C3 = A rows 64-127/192-255), one chunk per phase,
+// every wave contributing 2 glds per chunk IN THE SAME ORDER — so after
+// each wave waits s_waitcnt vmcnt(4) and passes a barrier, every chunk
+// issued >=3 phases earlier is certified landed for ALL waves (vmcnt is
+// per-wave; the uniform issue order + barrier extends it workgroup-wide).
+// Phase p of a tile consumes (A half H=p>>1, B half C=p&1): 4x2x2 = 16
+// MFMAs whose operands were staged 3-4 phases earlier — loads stay in
+// flight across every barrier and nothing in the steady-state loop ever
+// drains to vmcnt(0) (cdna_hip_programming.md §5 T3+T4: "counted-vs-drain0
+// = +38%@4k / +73%@8k"; the v2 structure keeps a once-per-tile drain that
+// its prefetch covers at 4096 but not at 8192 where loads come from HBM).
+// The last tile keeps staging (wrapping to tile 0, never read) so the
+// wait arithmetic stays uniform instead of under-waiting when the
+// prefetch queue drains.
+// ---------------------------------------------------------------------------
+
+template <bool SETPRIO = true>
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v6_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  __shared__ __hip_bfloat16 lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  // chunk -> 16 image slots (1 KiB each, 8 rows of 128 B). Slot s of image
+  // img covers rows [8s, 8s+8). B cols live as Bt rows.
+  //   C0: B cols {64n+0..31}  = B slots {0-3, 8-11, 16-19, 24-27}
+  //   C1: A rows 0-63,128-191 = A slots {0-7, 16-23}
+  //   C2: B cols {64n+32..63} = B slots {4-7, 12-15, 20-23, 28-31}
+  //   C3: A rows 64-127,192-255 = A slots {8-15, 24-31}
+  // wave w stages slots chunk[2w], chunk[2w+1] — identical order per wave.
+  auto chunk_slot = [&](int chunk, int j) -> int {
+    const int k = wave * 2 + j;  // 0..15 within the chunk
+    switch (chunk) {
+      case 0: return (k >> 2) * 8 + (k & 3);          // B slots
+      case 2: return (k >> 2) * 8 + 4 + (k & 3);      // B slots
+      case 1: return (k >> 3) * 16 + (k & 7);         // A slots
+      default: return (k >> 3) * 16 + 8 + (k & 7);    // A slots
+    }
+  };
+
+  auto stage_chunk = [&](int buf, int kt, int chunk) {
+    const int k0 = kt * TK;
+    const int img = (chunk == 1 || chunk == 3) ? 0 : 1;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int s_img = chunk_slot(chunk, j);
+      const int D = s_img * 1024 + lane * 16;
+      const int L = swz(D);
+      const int row = L >> 7;
+      const int k = (L & 127) >> 1;
+      const __hip_bfloat16* g =
+          img == 0 ? A + (size_t)(brow + row) * K + k0 + k
+                   : Bt + (size_t)(bcol + row) * K + k0 + k;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)(
+              &lds[buf][img * TM * TK] + s_img * 512),
+          16, 0, 0);
+    }
+  };
+
+  auto read_a = [&](int buf, int rbase, int kh) -> bf16x8 {
+    const int r = rbase + (lane & 15);
+    const int off = swz(r * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][0] + off);
+  };
+  auto read_b = [&](int buf, int cbase, int kh) -> bf16x8 {
+    const int c = cbase + (lane & 15);
+    const int off = swz(c * 128 + kh * 64 + ((lane >> 4) * 16));
+    return *(const bf16x8*)((const char*)&lds[buf][TM * TK] + off);
+  };
+
+  if (SETPRIO && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  accfrag_t acc[8][4] = {};
+  bf16x8 afrag[4][2];  // phase-local: 4 row frags x 2 k-halves
+  bf16x8 bfrag[2][2];  // phase-local: 2 col frags x 2 k-halves
+
+  // prologue: tile 0 fully staged and drained once (the only vmcnt(0))
+#pragma unroll
+  for (int c = 0; c < 4; ++c) stage_chunk(0, 0, c);
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    // the last tile stages tile 0 again (into the buffer it no longer
+    // reads) purely to keep every wave's outstanding-load count uniform
+    const int kt_next = (kt + 1 < ntiles) ? kt + 1 : 0;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      stage_chunk(nxt, kt_next, p);
+      // certify the chunk issued 3 phases ago: 2 glds per chunk, keep the
+      // latest 2 chunks (4 loads) in flight — never 0 in the main loop
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const int H = p >> 1;   // A half: rr 4H..4H+3
+      const int Cc = p & 1;   // B half: c 2Cc..2Cc+1
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh)
+          afrag[rr][kh] =
+              read_a(cur, wave_m * 128 + H * 64 + rr * 16, kh);
+      }
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh)
+          bfrag[c][kh] = read_b(cur, wave_n * 64 + Cc * 32 + c * 16, kh);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int kh = 0; kh < 2; ++kh) {
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+          for (int c = 0; c < 2; ++c) {
+            acc[H * 4 + rr][Cc * 2 + c] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    afrag[rr][kh], bfrag[c][kh],
+                    acc[H * 4 + rr][Cc * 2 + c], 0, 0, 0);
+          }
+        }
+      }
+    }
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM v5: register-staged K-loop — NO LDS, NO barriers. Every wave loads
 // its own MFMA fragments straight from global memory (both operands are
 // K-contiguous, so a 16x16x32 fragment is 16 contiguous bytes per lane:
@@ -1393,6 +1551,79 @@ py::dict gemm_stress_bf16_v5_impl(int size, int iters, bool setprio) {
   return d;
 }
 
+py::dict gemm_stress_bf16_v6_impl(int size, int iters, bool setprio) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  auto* kern = setprio ? gemm_bf16_v6_kernel<true> : gemm_bf16_v6_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
+                     N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c,
+                       M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-chunk-counted-vmcnt";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
+py::dict gemm_stress_bf16_v6(int size, int iters) {
+  return gemm_stress_bf16_v6_impl(size, iters, true);
+}
+
+py::dict gemm_stress_bf16_v6_nosp(int size, int iters) {
+  return gemm_stress_bf16_v6_impl(size, iters, false);
+}
+
 py::dict gemm_stress_bf16_v5(int size, int iters) {
   return gemm_stress_bf16_v5_impl(size, iters, true);
 }
@@ -1690,6 +1921,13 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2_panel", &gemm_stress_bf16_v2_panel,
         py::arg("size") = 8192, py::arg("iters") = 5, py::arg("panel") = 16,
         "A/B variant of v2 with L2 panel supertiling (8/16/32, 0=off)");
+  m.def("gemm_stress_bf16_v6", &gemm_stress_bf16_v6, py::arg("size") = 8192,
+        py::arg("iters") = 8,
+        "chunk-pipelined counted-vmcnt bf16 GEMM stress (v6, no boundary "
+        "drain)");
+  m.def("gemm_stress_bf16_v6_nosp", &gemm_stress_bf16_v6_nosp,
+        py::arg("size") = 8192, py::arg("iters") = 8,
+        "v6 with the static setprio hint compiled out (A/B seam)");
   m.def("gemm_stress_bf16_v5", &gemm_stress_bf16_v5, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "register-staged no-LDS no-barrier bf16 GEMM stress (v5)");

@@ -24,11 +24,16 @@ g++ -O1 -g -fno-omit-frame-pointer -fsanitize=address -shared -fPIC \
 echo "asan build rc=$?" >> gpurun_out/asan_build.log
 
 ASAN_LIB=$(g++ -print-file-name=libasan.so)
-export ASAN_OPTIONS="detect_leaks=0:abort_on_error=1:disable_coredump=0:unmap_shadow_on_exit=1"
+# libstdc++ must be preloaded right after libasan: libamd_smi throws (and
+# catches) std::ios_failure during KFD discovery, and without libstdc++ in
+# the preload chain ASan's lazy __cxa_throw interceptor cannot resolve the
+# real symbol ("real___cxa_throw != 0" CHECK abort at init)
+STDCXX=$(ldconfig -p | awk '/libstdc\+\+\.so\.6 \(libc6,x86-64\)/{print $NF; exit}')
+export ASAN_OPTIONS="detect_leaks=0:abort_on_error=1:disable_coredump=0:unmap_shadow_on_exit=1:verify_asan_link_order=0"
 
 run_stress() {
   local label="$1"
-  timeout 600 env LD_PRELOAD="${ASAN_LIB}" python - > "gpurun_out/asan_${label}.log" 2>&1 <<'PYEOF'
+  timeout 600 env LD_PRELOAD="${ASAN_LIB} ${STDCXX}" python - > "gpurun_out/asan_${label}.log" 2>&1 <<'PYEOF'
 import concurrent.futures as cf
 import os, random, sys, time
 sys.path.insert(0, "/root/repo")
@@ -85,4 +90,4 @@ run_stress "mutex_on"
 # arm 2: mutex disabled — concurrent library entry, the actual race hunt
 GPUD_AMDSMI_NO_CALL_MUTEX=1 run_stress "mutex_off"
 
-tail -5 gpurun_out/asan_mutex_on.log gpurun_out/asan_mutex_off.log
+tail -n 5 gpurun_out/asan_mutex_on.log; tail -n 5 gpurun_out/asan_mutex_off.log
