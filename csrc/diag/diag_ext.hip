@@ -1145,6 +1145,139 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v6_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// GEMM v7: hand-scheduled asm K-loop on the v2 choreography — the guide's
+// named lever past the plain-HIP plateau (cdna_hip_programming.md §5 "What
+// does break it": a hand-scheduled K-loop with MFMA <-> load interleave and
+// counted waits). Memory layout, swizzle, slot assignment, barrier and
+// drain placement are byte-identical to v2; what changes is WHO schedules
+// the phase body: each phase is one asm statement (generated by
+// scripts/gen_v7_asm.py into gemm_v7_body.h) placing the next-phase
+// ds_read_b128s and the next-tile global_load_lds pieces one-per-two-MFMAs
+// inside the MFMA stream instead of hipcc's read-burst-then-MFMA-burst.
+//
+// Addressing exploits a provable property of the swizzle: in
+//   x = r*128 + kh*64 + (lane>>4)*16,  r = base + q*32 + rr*16 + (lane&15)
+// every term occupies disjoint bit ranges and bits 8..9 of x (the swz()
+// XOR selectors) come from (lane&15)*128 alone — so swz(x) = x ^ mask(lane)
+// and ONE per-lane base VGPR serves every fragment read of the kernel with
+// 16-bit immediate offsets (A: q*4096+rr*2048+kh*64; B likewise off its
+// own base). The glds pieces use the SADDR form (wave-uniform 64-bit SGPR
+// base + per-lane 32-bit offset VGPR) so the slot walk is one v_add_u32.
+// ---------------------------------------------------------------------------
+
+#include "gemm_v7_body.h"
+
+template <bool SETPRIO = true>
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  __shared__ __hip_bfloat16 lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  // per-lane LDS fragment-read base (swz mask is lane-only, see header)
+  const uint32_t lds0 = (uint32_t)(uintptr_t)&lds[0][0];
+  const uint32_t swz_lane =
+      (uint32_t)swz((lane & 15) * 128 + ((lane >> 4) * 16));
+
+  // glds geometry (v2 stage_slots equivalences): my wave owns slots
+  // wave*8 .. wave*8+7 — image A for waves 0-3, image B for 4-7
+  const int s_img0 = (wave * 8) & 31;
+  const int glds_img = wave >> 2;
+  const int D0 = s_img0 * 1024 + lane * 16;
+  const int L0 = swz(D0);
+  const int row0 = L0 >> 7;
+  const int kcol0 = (L0 & 127) >> 1;
+  const __hip_bfloat16* gptr =
+      glds_img == 0 ? A + (size_t)brow * K : Bt + (size_t)bcol * K;
+  uint64_t gbase;
+  {
+    const uint64_t p = (uint64_t)(uintptr_t)gptr;
+    const uint32_t lo = __builtin_amdgcn_readfirstlane((uint32_t)p);
+    const uint32_t hi = __builtin_amdgcn_readfirstlane((uint32_t)(p >> 32));
+    gbase = ((uint64_t)hi << 32) | lo;
+  }
+  const uint32_t gstride =
+      __builtin_amdgcn_readfirstlane((uint32_t)(8u * (uint32_t)K * 2u));
+  const uint32_t voff_lane = ((uint32_t)row0 * (uint32_t)K + kcol0) * 2u;
+  const uint32_t gdest_base = __builtin_amdgcn_readfirstlane(
+      lds0 + (uint32_t)glds_img * 32768u + (uint32_t)s_img0 * 1024u);
+
+  if (SETPRIO && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  accfrag_t acc[8][4] = {};
+  bf16x8 afrag[2][2][2] = {};
+  bf16x8 bfrag[4][2] = {};
+
+  // prologue: stage tile 0 (all 8 of my slots) with the builtin glds —
+  // identical addressing to the asm pieces, drained once
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int s_img = s_img0 + j;
+    const int D = s_img * 1024 + lane * 16;
+    const int L = swz(D);
+    const int row = L >> 7;
+    const int k = (L & 127) >> 1;
+    const __hip_bfloat16* g = gptr + (size_t)row * K + k;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(
+            &lds[0][glds_img * TM * TK] + s_img * 512),
+        16, 0, 0);
+  }
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    // last tile wraps its (never-read) prefetch to tile 0 so the wait
+    // arithmetic stays uniform (same trick as v6)
+    const int kt_next = (kt + 1 < ntiles) ? kt + 1 : 0;
+    uint32_t aaddr = lds0 + (uint32_t)cur * 65536u +
+                     (uint32_t)wave_m * 16384u + swz_lane;
+    uint32_t baddr = lds0 + (uint32_t)cur * 65536u + 32768u +
+                     (uint32_t)wave_n * 8192u + swz_lane;
+    uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;
+    uint32_t gdest = gdest_base + (uint32_t)nxt * 65536u;
+    uint32_t mscratch;
+    V7_PHASE0(acc, afrag, bfrag, aaddr, baddr, voff, gdest, mscratch, gbase,
+              gstride);
+    V7_PHASE1(acc, afrag, bfrag, aaddr, voff, gdest, mscratch, gbase,
+              gstride);
+    V7_PHASE2(acc, afrag, bfrag, aaddr);
+    V7_PHASE3(acc, afrag, bfrag, aaddr);
+    // K-tile boundary: tile kt+1's 8 glds were issued in phases 0-1 with
+    // 2-3 phases of MFMA cover (v2 semantics)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM v5: register-staged K-loop — NO LDS, NO barriers. Every wave loads
 // its own MFMA fragments straight from global memory (both operands are
 // K-contiguous, so a 16x16x32 fragment is 16 contiguous bytes per lane:
@@ -1551,6 +1684,79 @@ py::dict gemm_stress_bf16_v5_impl(int size, int iters, bool setprio) {
   return d;
 }
 
+py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  auto* kern = setprio ? gemm_bf16_v7_kernel<true> : gemm_bf16_v7_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
+                     N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c,
+                       M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-asm-kloop";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
+py::dict gemm_stress_bf16_v7(int size, int iters) {
+  return gemm_stress_bf16_v7_impl(size, iters, true);
+}
+
+py::dict gemm_stress_bf16_v7_nosp(int size, int iters) {
+  return gemm_stress_bf16_v7_impl(size, iters, false);
+}
+
 py::dict gemm_stress_bf16_v6_impl(int size, int iters, bool setprio) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -1921,6 +2127,12 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v2_panel", &gemm_stress_bf16_v2_panel,
         py::arg("size") = 8192, py::arg("iters") = 5, py::arg("panel") = 16,
         "A/B variant of v2 with L2 panel supertiling (8/16/32, 0=off)");
+  m.def("gemm_stress_bf16_v7", &gemm_stress_bf16_v7, py::arg("size") = 8192,
+        py::arg("iters") = 8,
+        "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
+  m.def("gemm_stress_bf16_v7_nosp", &gemm_stress_bf16_v7_nosp,
+        py::arg("size") = 8192, py::arg("iters") = 8,
+        "v7 with the static setprio hint compiled out (A/B seam)");
   m.def("gemm_stress_bf16_v6", &gemm_stress_bf16_v6, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "chunk-pipelined counted-vmcnt bf16 GEMM stress (v6, no boundary "
