@@ -1183,10 +1183,13 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
   const int bcol = (blockIdx.x % tiles_n) * TN;
   const int ntiles = K / TK;
 
-  // per-lane LDS fragment-read base (swz mask is lane-only, see header)
+  // per-lane LDS fragment-read bases, one per MFMA k-half: kh*64 must live
+  // INSIDE the swizzled part (the mask XORs bit 6; adding 64 across a set
+  // mask bit would carry into bit 7)
   const uint32_t lds0 = (uint32_t)(uintptr_t)&lds[0][0];
-  const uint32_t swz_lane =
-      (uint32_t)swz((lane & 15) * 128 + ((lane >> 4) * 16));
+  const int lane_raw = (lane & 15) * 128 + ((lane >> 4) * 16);
+  const uint32_t swz_lane_k0 = (uint32_t)swz(lane_raw);
+  const uint32_t swz_lane_k1 = (uint32_t)swz(lane_raw + 64);
 
   // glds geometry (v2 stage_slots equivalences): my wave owns slots
   // wave*8 .. wave*8+7 — image A for waves 0-3, image B for 4-7
@@ -1243,19 +1246,23 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
     // last tile wraps its (never-read) prefetch to tile 0 so the wait
     // arithmetic stays uniform (same trick as v6)
     const int kt_next = (kt + 1 < ntiles) ? kt + 1 : 0;
-    uint32_t aaddr = lds0 + (uint32_t)cur * 65536u +
-                     (uint32_t)wave_m * 16384u + swz_lane;
-    uint32_t baddr = lds0 + (uint32_t)cur * 65536u + 32768u +
-                     (uint32_t)wave_n * 8192u + swz_lane;
+    const uint32_t abase =
+        lds0 + (uint32_t)cur * 65536u + (uint32_t)wave_m * 16384u;
+    const uint32_t bbase = lds0 + (uint32_t)cur * 65536u + 32768u +
+                           (uint32_t)wave_n * 8192u;
+    uint32_t aaddr0 = abase + swz_lane_k0;
+    uint32_t aaddr1 = abase + swz_lane_k1;
+    uint32_t baddr0 = bbase + swz_lane_k0;
+    uint32_t baddr1 = bbase + swz_lane_k1;
     uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;
     uint32_t gdest = gdest_base + (uint32_t)nxt * 65536u;
     uint32_t mscratch;
-    V7_PHASE0(acc, afrag, bfrag, aaddr, baddr, voff, gdest, mscratch, gbase,
-              gstride);
-    V7_PHASE1(acc, afrag, bfrag, aaddr, voff, gdest, mscratch, gbase,
-              gstride);
-    V7_PHASE2(acc, afrag, bfrag, aaddr);
-    V7_PHASE3(acc, afrag, bfrag, aaddr);
+    V7_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+              gdest, mscratch, gbase, gstride);
+    V7_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
+              gbase, gstride);
+    V7_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+    V7_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
     // K-tile boundary: tile kt+1's 8 glds were issued in phases 0-1 with
     // 2-3 phases of MFMA cover (v2 semantics)
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
