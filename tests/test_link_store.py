@@ -210,3 +210,51 @@ def test_xgmi_component_uses_store(monkeypatch, tmp_path):
         assert cr.health == "Healthy", cr.reason
     finally:
         core.close()
+
+
+def test_infiniband_component_uses_store(monkeypatch, tmp_path):
+    """IB port drops surface through the SQLite history store and persist
+    until recovery + sticky window (the reference's ibports store)."""
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+
+    core = build_core(
+        Config(data_dir=str(tmp_path)), in_memory_db=True,
+        kmsg_writable=False, record_reboot=False,
+    )
+    try:
+        from gpud_amd.components.host.infiniband import InfinibandComponent
+
+        comp = InfinibandComponent(core.gpud_instance)
+        assert comp.link_store is not None
+        clock = Clock()
+        comp.link_store.now = clock
+        comp.link_store.min_insert_interval = 0.0
+
+        def ports(active):
+            return [{
+                "device": "mlx5_0", "port": "1", "active": active,
+                "rate_gbps": 400.0,
+                "counters": {"link_downed": 0 if active else 1,
+                             "link_error_recovery": 0, "symbol_error": 0,
+                             "port_rcv_errors": 0},
+            }]
+
+        # port down for 5+ minutes -> drop detected via the store
+        comp.get_ports = lambda: ports(False)
+        for _ in range(6):
+            cr = comp.check()
+            clock.advance(60)
+        assert cr.health == "Unhealthy"
+        # port recovers: still surfaced within the 10-min sticky window
+        comp.get_ports = lambda: ports(True)
+        cr = comp.check()
+        assert cr.health in ("Unhealthy", "Degraded"), cr.reason
+        assert "drop" in cr.reason.lower() or "flap" in cr.reason.lower()
+        # set-healthy tombstones the finding
+        comp.set_healthy()
+        cr = comp.check()
+        assert cr.health == "Healthy", cr.reason
+    finally:
+        core.close()
