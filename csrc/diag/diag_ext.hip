@@ -1167,7 +1167,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v6_kernel(
 
 #include "gemm_v7_body.h"
 
-template <bool SETPRIO = true>
+template <bool SETPRIO = false, int STYLE = 0>
 __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
     const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1257,12 +1257,30 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
     uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;
     uint32_t gdest = gdest_base + (uint32_t)nxt * 65536u;
     uint32_t mscratch;
-    V7_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
-              gdest, mscratch, gbase, gstride);
-    V7_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
-              gbase, gstride);
-    V7_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
-    V7_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    if constexpr (STYLE == 1) {
+      // in-stream s_setprio(1..0) per phase (dynamic priority form)
+      V7S_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V7S_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
+                 gbase, gstride);
+      V7S_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7S_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else if constexpr (STYLE == 2) {
+      // memory groups placed late (under MFMAs 8-16)
+      V7L_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V7L_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
+                 gbase, gstride);
+      V7L_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7L_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else {
+      V7_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                gdest, mscratch, gbase, gstride);
+      V7_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
+                gbase, gstride);
+      V7_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    }
     // K-tile boundary: tile kt+1's 8 glds were issued in phases 0-1 with
     // 2-3 phases of MFMA cover (v2 semantics)
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -1691,7 +1709,8 @@ py::dict gemm_stress_bf16_v5_impl(int size, int iters, bool setprio) {
   return d;
 }
 
-py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio) {
+py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
+                                  int style = 0) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
   if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
@@ -1709,7 +1728,10 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio) {
   hipEvent_t t0, t1;
   HIP_CHECK(hipEventCreate(&t0));
   HIP_CHECK(hipEventCreate(&t1));
-  auto* kern = setprio ? gemm_bf16_v7_kernel<true> : gemm_bf16_v7_kernel<false>;
+  auto* kern = setprio ? gemm_bf16_v7_kernel<true, 0>
+                       : gemm_bf16_v7_kernel<false, 0>;
+  if (style == 1) kern = gemm_bf16_v7_kernel<false, 1>;
+  else if (style == 2) kern = gemm_bf16_v7_kernel<false, 2>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
@@ -1757,11 +1779,21 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio) {
 }
 
 py::dict gemm_stress_bf16_v7(int size, int iters) {
+  // static setprio measured +3% @4096 but -9% @8192 (profiles log): the
+  // shipped default is setprio-OFF, the 8192 diag shape's winner
+  return gemm_stress_bf16_v7_impl(size, iters, false);
+}
+
+py::dict gemm_stress_bf16_v7_style(int size, int iters, int style) {
+  return gemm_stress_bf16_v7_impl(size, iters, false, style);
+}
+
+py::dict gemm_stress_bf16_v7_sp(int size, int iters) {
   return gemm_stress_bf16_v7_impl(size, iters, true);
 }
 
 py::dict gemm_stress_bf16_v7_nosp(int size, int iters) {
-  return gemm_stress_bf16_v7_impl(size, iters, false);
+  return gemm_stress_bf16_v7_impl(size, iters, false);  // alias of default
 }
 
 py::dict gemm_stress_bf16_v6_impl(int size, int iters, bool setprio) {
@@ -2137,6 +2169,12 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v7", &gemm_stress_bf16_v7, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
+  m.def("gemm_stress_bf16_v7_style", &gemm_stress_bf16_v7_style,
+        py::arg("size") = 8192, py::arg("iters") = 8, py::arg("style") = 0,
+        "v7 schedule-style seam: 0=base 1=in-burst setprio 2=late mem groups");
+  m.def("gemm_stress_bf16_v7_sp", &gemm_stress_bf16_v7_sp,
+        py::arg("size") = 8192, py::arg("iters") = 8,
+        "v7 with static young-half setprio (A/B seam)");
   m.def("gemm_stress_bf16_v7_nosp", &gemm_stress_bf16_v7_nosp,
         py::arg("size") = 8192, py::arg("iters") = 8,
         "v7 with the static setprio hint compiled out (A/B seam)");
