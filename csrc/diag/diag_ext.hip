@@ -1273,6 +1273,29 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
                  gbase, gstride);
       V7L_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
       V7L_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else if constexpr (STYLE == 3) {
+      // E: reordered p0 head — kh-grouped reads, incremental counted waits
+      V7E_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V7E_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1, voff, gdest, mscratch,
+                 gbase, gstride);
+      V7E_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7E_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else if constexpr (STYLE == 4) {
+      // F: all 8 glds slots issued in phase 0 (max cover before the
+      // boundary drain); phase 1 carries only its ds_read prefetches
+      V7F_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V7F_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7F_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7F_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else if constexpr (STYLE == 5) {
+      // X = E + F combined
+      V7X_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V7X_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7X_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7X_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
     } else {
       V7_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
                 gdest, mscratch, gbase, gstride);
@@ -1732,6 +1755,9 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
                        : gemm_bf16_v7_kernel<false, 0>;
   if (style == 1) kern = gemm_bf16_v7_kernel<false, 1>;
   else if (style == 2) kern = gemm_bf16_v7_kernel<false, 2>;
+  else if (style == 3) kern = gemm_bf16_v7_kernel<false, 3>;
+  else if (style == 4) kern = gemm_bf16_v7_kernel<false, 4>;
+  else if (style == 5) kern = gemm_bf16_v7_kernel<false, 5>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
@@ -2171,7 +2197,8 @@ PYBIND11_MODULE(_diag, m) {
         "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
   m.def("gemm_stress_bf16_v7_style", &gemm_stress_bf16_v7_style,
         py::arg("size") = 8192, py::arg("iters") = 8, py::arg("style") = 0,
-        "v7 schedule-style seam: 0=base 1=in-burst setprio 2=late mem groups");
+        "v7 schedule-style seam: 0=base 1=in-burst setprio 2=late mem "
+        "groups 3=reordered-p0-head 4=all-glds-p0 5=3+4");
   m.def("gemm_stress_bf16_v7_sp", &gemm_stress_bf16_v7_sp,
         py::arg("size") = 8192, py::arg("iters") = 8,
         "v7 with static young-half setprio (A/B seam)");
