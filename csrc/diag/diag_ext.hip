@@ -1326,6 +1326,161 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// GEMM v8: 3-buffer glds ring — the guide's "glds with >1 tile in flight
+// across the barrier" structure (cdna_hip_programming.md §5 "Pipelining
+// across barriers": 2-buf overlap +40%, 3-buf span +83% vs serial at
+// 1-block/CU occupancy and vgpr>=200 — exactly this kernel's regime).
+// v7's 2 buffers cap glds cover at ONE tile (~1.2 us), marginal against
+// loaded HBM latency; here every glds is issued TWO tiles before its
+// data is read and the K-tile boundary is a counted per-wave
+// s_waitcnt vmcnt(N) + lgkmcnt(0) + raw s_barrier — no vmcnt(0) drain
+// anywhere in the steady state (and no __syncthreads(): its fence would
+// emit vmcnt(0) and drain the in-flight glds this structure exists for).
+// Tile 256x128x64 (48 KiB/buffer, 3 buffers = 144 KiB of the 160 KiB
+// LDS); 8 waves as 4(M) x 2(N), wave tile 64x64, acc[4][4] = 64 VGPRs.
+// The staging split keeps per-wave glds counts uniform for the counted
+// wait: waves 0-3 stage the A image (8 slots/tile, boundary vmcnt(8)),
+// waves 4-7 stage B (4 slots, vmcnt(4)). Two hand-scheduled asm phases
+// per tile (one per MFMA k-half; bodies generated by
+// scripts/gen_v8_asm.py): phase 0 reads all 16 fragments (kh0 batch
+// first — lgkmcnt(8)) and carries ALL the tile's glds pieces (the v7
+// E+F lessons); phase 1 is a pure MFMA burst. Addressing identities are
+// v7's (swz() XOR mask is lane-only, so one swizzled per-lane base per
+// k-half serves every fragment read with 16-bit immediates).
+// ---------------------------------------------------------------------------
+
+#include "gemm_v8_body.h"
+
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v8_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 128, TK = 64;
+  constexpr int BUF = (TM + TN) * TK;  // bf16 elems: A image then B image
+  __shared__ __hip_bfloat16 lds[3 * BUF];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 1;  // 0..3 — 64-row band
+  const int wave_n = wave & 1;   // 0..1 — 64-col band
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  const uint32_t lds0 = (uint32_t)(uintptr_t)&lds[0];
+  const int lane_raw = (lane & 15) * 128 + ((lane >> 4) * 16);
+  const uint32_t swz_lane_k0 = (uint32_t)swz(lane_raw);
+  const uint32_t swz_lane_k1 = (uint32_t)swz(lane_raw + 64);
+
+  // staging geometry: A image = 32 slots (1 KiB each, 8 rows of 128 B),
+  // B image = 16 slots at +32 KiB. Waves 0-3 stage A slots wave*8..+8,
+  // waves 4-7 stage B slots (wave-4)*4..+4.
+  const int glds_img = wave >> 2;  // 0 = A, 1 = B
+  const int nglds = glds_img == 0 ? 8 : 4;
+  const int s_img0 = glds_img == 0 ? wave * 8 : (wave - 4) * 4;
+  const int D0 = s_img0 * 1024 + lane * 16;
+  const int L0 = swz(D0);
+  const int row0 = L0 >> 7;
+  const int kcol0 = (L0 & 127) >> 1;
+  const __hip_bfloat16* gptr =
+      glds_img == 0 ? A + (size_t)brow * K : Bt + (size_t)bcol * K;
+  uint64_t gbase;
+  {
+    const uint64_t p = (uint64_t)(uintptr_t)gptr;
+    const uint32_t lo = __builtin_amdgcn_readfirstlane((uint32_t)p);
+    const uint32_t hi = __builtin_amdgcn_readfirstlane((uint32_t)(p >> 32));
+    gbase = ((uint64_t)hi << 32) | lo;
+  }
+  const uint32_t gstride =
+      __builtin_amdgcn_readfirstlane((uint32_t)(8u * (uint32_t)K * 2u));
+  const uint32_t voff_lane = ((uint32_t)row0 * (uint32_t)K + kcol0) * 2u;
+  const uint32_t gdest_img = __builtin_amdgcn_readfirstlane(
+      lds0 + (uint32_t)glds_img * (uint32_t)(TM * TK * 2) +
+      (uint32_t)s_img0 * 1024u);  // + ring buffer offset at issue time
+
+  accfrag_t acc[4][4] = {};
+  bf16x8 afrag[4][2] = {};
+  bf16x8 bfrag[4][2] = {};
+
+  constexpr uint32_t BUFB = BUF * 2;  // bytes per ring buffer (49152)
+
+  // prologue: stage tiles 0 and 1 into ring buffers 0,1 with the builtin
+  // glds (identical addressing to the asm pieces), then a COUNTED wait —
+  // tile 0 certified, tile 1 left in flight across the first barrier.
+  for (int t = 0; t < 2; ++t) {
+    for (int j = 0; j < nglds; ++j) {
+      const int s_img = s_img0 + j;
+      const int D = s_img * 1024 + lane * 16;
+      const int L = swz(D);
+      const int row = L >> 7;
+      const int k = (L & 127) >> 1;
+      const __hip_bfloat16* g = gptr + (size_t)row * K + (size_t)t * TK + k;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)g,
+          (__attribute__((address_space(3))) unsigned int*)(
+              &lds[(size_t)t * BUF + glds_img * TM * TK] + s_img * 512),
+          16, 0, 0);
+    }
+  }
+  if (glds_img == 0)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  uint32_t cur_off = 0, nxt_off = BUFB, wr_off = 2 * BUFB;
+  for (int kt = 0; kt < ntiles; ++kt) {
+    // glds target: tile kt+2 into the retired ring buffer (dummy-wrapped
+    // to tile 0 past the end — written, never read, keeps every wave's
+    // outstanding-load count uniform)
+    const int kt2 = (kt + 2 < ntiles) ? kt + 2 : 0;
+    const uint32_t abase = lds0 + cur_off + (uint32_t)wave_m * 8192u;
+    const uint32_t bbase =
+        lds0 + cur_off + (uint32_t)(TM * TK * 2) + (uint32_t)wave_n * 8192u;
+    uint32_t aaddr0 = abase + swz_lane_k0;
+    uint32_t aaddr1 = abase + swz_lane_k1;
+    uint32_t baddr0 = bbase + swz_lane_k0;
+    uint32_t baddr1 = bbase + swz_lane_k1;
+    uint32_t voff = voff_lane + (uint32_t)kt2 * (TK * 2);
+    uint32_t gdest = gdest_img + wr_off;
+    uint32_t mscratch;
+    if (glds_img == 0) {
+      V8A_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V8_PHASE1(acc, afrag, bfrag);
+      // counted boundary: kt+1's 8 loads (issued LAST tile) certified,
+      // kt+2's 8 stay in flight across the barrier
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      V8B_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
+                 gdest, mscratch, gbase, gstride);
+      V8_PHASE1(acc, afrag, bfrag);
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    const uint32_t rot = cur_off;
+    cur_off = nxt_off;
+    nxt_off = wr_off;
+    wr_off = rot;
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 64 + rr * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[rr][c][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM v5: register-staged K-loop — NO LDS, NO barriers. Every wave loads
 // its own MFMA fragments straight from global memory (both operands are
 // K-contiguous, so a 16x16x32 fragment is 16 contiguous bytes per lane:
@@ -1822,6 +1977,70 @@ py::dict gemm_stress_bf16_v7_nosp(int size, int iters) {
   return gemm_stress_bf16_v7_impl(size, iters, false);  // alias of default
 }
 
+py::dict gemm_stress_bf16_v8(int size, int iters) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 128);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_bf16_v8_kernel, dim3(blocks), dim3(512), 0, 0, d_a,
+                     d_bt, d_c, M, N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_bf16_v8_kernel, dim3(blocks), dim3(512), 0, 0,
+                       d_a, d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256x128-3buf-glds-ring";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_bf16_v6_impl(int size, int iters, bool setprio) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -2195,6 +2414,10 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v7", &gemm_stress_bf16_v7, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
+  m.def("gemm_stress_bf16_v8", &gemm_stress_bf16_v8, py::arg("size") = 8192,
+        py::arg("iters") = 8,
+        "3-buffer glds-ring bf16 GEMM stress (v8): counted vmcnt boundary, "
+        "raw barrier, >1 tile in flight");
   m.def("gemm_stress_bf16_v7_style", &gemm_stress_bf16_v7_style,
         py::arg("size") = 8192, py::arg("iters") = 8, py::arg("style") = 0,
         "v7 schedule-style seam: 0=base 1=in-burst setprio 2=late mem "
