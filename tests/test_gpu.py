@@ -108,6 +108,14 @@ def test_diag_gemm_bf16_verified():
     res2 = _diag.gemm_stress_bf16_v2(size=4096, iters=2)
     assert res2["verified"], res2
     assert res2["tflops"] > 800, res2
+    # the shipped component default (v7X hand-scheduled asm K-loop) must
+    # verify and beat the plain-HIP v2 structure it replaced
+    res2x = _diag.gemm_stress_bf16_v7(size=4096, iters=2)
+    assert res2x["verified"], res2x
+    assert res2x["tflops"] > 950, res2x
+    # the 3-buffer ring (v8) stays as a structural seam: correct, slower
+    res8 = _diag.gemm_stress_bf16_v8(size=1024, iters=2)
+    assert res8["verified"], res8
     res3 = _diag.gemm_stress_mxfp8(size=4096, iters=2)
     assert res3["verified"], res3
     assert res3["tflops"] > 1400, res3
