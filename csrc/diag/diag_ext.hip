@@ -1296,6 +1296,16 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
       V7X_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1);
       V7X_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
       V7X_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
+    } else if constexpr (STYLE == 6 || STYLE == 7) {
+      // diagnostic skeletons (NOT health checks — wrong results by
+      // construction, no verification): the X schedule with the glds
+      // stream removed (6: barrier kept; 7: barrier removed too) to
+      // attribute the residual wave-park between ds_read latency /
+      // barrier skew / glds-boundary cover
+      V7N_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1);
+      V7N_PHASE1(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7N_PHASE2(acc, afrag, bfrag, aaddr0, aaddr1);
+      V7N_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
     } else {
       V7_PHASE0(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff,
                 gdest, mscratch, gbase, gstride);
@@ -1305,9 +1315,15 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
       V7_PHASE3(acc, afrag, bfrag, aaddr0, aaddr1);
     }
     // K-tile boundary: tile kt+1's 8 glds were issued in phases 0-1 with
-    // 2-3 phases of MFMA cover (v2 semantics)
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
+    // 2-3 phases of MFMA cover (v2 semantics). The skeletons have no
+    // glds in flight (6: barrier-only boundary; 7: free-running waves).
+    if constexpr (STYLE < 6) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    } else if constexpr (STYLE == 6) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
   }
 
   // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
@@ -1913,6 +1929,8 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   else if (style == 3) kern = gemm_bf16_v7_kernel<false, 3>;
   else if (style == 4) kern = gemm_bf16_v7_kernel<false, 4>;
   else if (style == 5) kern = gemm_bf16_v7_kernel<false, 5>;
+  else if (style == 6) kern = gemm_bf16_v7_kernel<false, 6>;
+  else if (style == 7) kern = gemm_bf16_v7_kernel<false, 7>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
@@ -1951,7 +1969,10 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   py::dict d;
   d["dtype"] = "bf16";
   d["size"] = size;
-  d["structure"] = "256sq-asm-kloop";
+  d["structure"] =
+      style >= 6 ? "256sq-asm-kloop-SKELETON (wrong results by design)"
+                 : "256sq-asm-kloop";
+  d["skeleton"] = style >= 6;
   d["tflops"] = flops / (ms * 1e-3) / 1e12;
   d["seconds_per_gemm"] = ms * 1e-3 / iters;
   d["verify_failures"] = (long)bad;
