@@ -1342,6 +1342,135 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// GEMM v7P: the X schedule with the phase-3 kh1 half CARRIED ACROSS THE
+// BARRIER. Skeleton attribution (gemm_skel log) decomposed v7X's ~32%
+// wave-park into ~145 TF barrier skew + ~290 TF glds boundary + ~400 TF
+// schedule/read-latency; the biggest single exposure is the post-barrier
+// phase-0 head, where every wave stalls on 6-12 LDS read returns with no
+// ready work. Here phase 3 runs only its kh0 burst before the boundary;
+// its kh1 burst (operands register-resident: afrag[1][..][1] + the OLD
+// B-fragment set) executes AFTER the barrier, giving each wave 8 MFMAs
+// (~128 XDL cycles) of latency-free work that covers the phase-0 read
+// head. Costs: B fragments double-buffered per tile (+32 VGPR -> ~238)
+// and the loop unrolls two K-tiles per iteration for the B-set
+// ping-pong. First iteration's carried MFMAs multiply zero-initialised
+// fragments (a 0*0 accumulate) so the pipeline needs no peel; the last
+// tile's carried half runs in V7P_TAIL after the loop.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 256, TN = 256, TK = 64;
+  __shared__ __hip_bfloat16 lds[2][2 * TM * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;
+  const int wave_n = wave & 3;
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;  // multiple of 4 for every valid size
+
+  const uint32_t lds0 = (uint32_t)(uintptr_t)&lds[0][0];
+  const int lane_raw = (lane & 15) * 128 + ((lane >> 4) * 16);
+  const uint32_t swz_lane_k0 = (uint32_t)swz(lane_raw);
+  const uint32_t swz_lane_k1 = (uint32_t)swz(lane_raw + 64);
+
+  const int s_img0 = (wave * 8) & 31;
+  const int glds_img = wave >> 2;
+  const int D0 = s_img0 * 1024 + lane * 16;
+  const int L0 = swz(D0);
+  const int row0 = L0 >> 7;
+  const int kcol0 = (L0 & 127) >> 1;
+  const __hip_bfloat16* gptr =
+      glds_img == 0 ? A + (size_t)brow * K : Bt + (size_t)bcol * K;
+  uint64_t gbase;
+  {
+    const uint64_t p = (uint64_t)(uintptr_t)gptr;
+    const uint32_t lo = __builtin_amdgcn_readfirstlane((uint32_t)p);
+    const uint32_t hi = __builtin_amdgcn_readfirstlane((uint32_t)(p >> 32));
+    gbase = ((uint64_t)hi << 32) | lo;
+  }
+  const uint32_t gstride =
+      __builtin_amdgcn_readfirstlane((uint32_t)(8u * (uint32_t)K * 2u));
+  const uint32_t voff_lane = ((uint32_t)row0 * (uint32_t)K + kcol0) * 2u;
+  const uint32_t gdest_base = __builtin_amdgcn_readfirstlane(
+      lds0 + (uint32_t)glds_img * 32768u + (uint32_t)s_img0 * 1024u);
+
+  accfrag_t acc[8][4] = {};
+  bf16x8 afrag[2][2][2] = {};
+  bf16x8 bfrag[2][4][2] = {};  // per-tile ping-pong (cross-barrier carry)
+
+  // prologue: stage tile 0, drain once (the only vmcnt(0) with no cover)
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int s_img = s_img0 + j;
+    const int D = s_img * 1024 + lane * 16;
+    const int L = swz(D);
+    const int row = L >> 7;
+    const int k = (L & 127) >> 1;
+    const __hip_bfloat16* g = gptr + (size_t)row * K + k;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(
+            &lds[0][glds_img * TM * TK] + s_img * 512),
+        16, 0, 0);
+  }
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __builtin_amdgcn_s_barrier();
+
+#define V7P_TILE(KT, BC, BO)                                                 \
+  {                                                                          \
+    const int kt = (KT);                                                     \
+    const int cur = kt & 1;                                                  \
+    const int nxt = cur ^ 1;                                                 \
+    const int kt_next = (kt + 1 < ntiles) ? kt + 1 : 0;                      \
+    const uint32_t abase =                                                   \
+        lds0 + (uint32_t)cur * 65536u + (uint32_t)wave_m * 16384u;           \
+    const uint32_t bbase = lds0 + (uint32_t)cur * 65536u + 32768u +          \
+                           (uint32_t)wave_n * 8192u;                         \
+    uint32_t aaddr0 = abase + swz_lane_k0;                                   \
+    uint32_t aaddr1 = abase + swz_lane_k1;                                   \
+    uint32_t baddr0 = bbase + swz_lane_k0;                                   \
+    uint32_t baddr1 = bbase + swz_lane_k1;                                   \
+    uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;                    \
+    uint32_t gdest = gdest_base + (uint32_t)nxt * 65536u;                    \
+    uint32_t mscratch;                                                       \
+    V7P_PHASE0(acc, afrag, bfrag[BC], bfrag[BO], aaddr0, aaddr1, baddr0,     \
+               baddr1, voff, gdest, mscratch, gbase, gstride);               \
+    V7X_PHASE1(acc, afrag, bfrag[BC], aaddr0, aaddr1);                       \
+    V7X_PHASE2(acc, afrag, bfrag[BC], aaddr0, aaddr1);                       \
+    V7P_PHASE3A(acc, afrag, bfrag[BC]);                                      \
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                         \
+    __builtin_amdgcn_s_barrier();                                            \
+  }
+
+  for (int kt = 0; kt < ntiles; kt += 2) {
+    V7P_TILE(kt, 0, 1);
+    V7P_TILE(kt + 1, 1, 0);
+  }
+#undef V7P_TILE
+  // the final tile's carried kh1 half (its B set is bfrag[1]: ntiles even)
+  V7P_TAIL(acc, afrag, bfrag[1]);
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int R = 0; R < 8; ++R) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[R][c][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM v8: 3-buffer glds ring — the guide's "glds with >1 tile in flight
 // across the barrier" structure (cdna_hip_programming.md §5 "Pipelining
 // across barriers": 2-buf overlap +40%, 3-buf span +83% vs serial at
@@ -1931,6 +2060,7 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   else if (style == 5) kern = gemm_bf16_v7_kernel<false, 5>;
   else if (style == 6) kern = gemm_bf16_v7_kernel<false, 6>;
   else if (style == 7) kern = gemm_bf16_v7_kernel<false, 7>;
+  else if (style == 8) kern = gemm_bf16_v7p_kernel;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
@@ -1969,10 +2099,12 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   py::dict d;
   d["dtype"] = "bf16";
   d["size"] = size;
+  const bool skel = (style == 6 || style == 7);
   d["structure"] =
-      style >= 6 ? "256sq-asm-kloop-SKELETON (wrong results by design)"
-                 : "256sq-asm-kloop";
-  d["skeleton"] = style >= 6;
+      skel ? "256sq-asm-kloop-SKELETON (wrong results by design)"
+           : (style == 8 ? "256sq-asm-kloop-xbarrier-pipelined"
+                         : "256sq-asm-kloop");
+  d["skeleton"] = skel;
   d["tflops"] = flops / (ms * 1e-3) / 1e12;
   d["seconds_per_gemm"] = ms * 1e-3 / iters;
   d["verify_failures"] = (long)bad;
@@ -2444,7 +2576,9 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v7_style", &gemm_stress_bf16_v7_style,
         py::arg("size") = 8192, py::arg("iters") = 8, py::arg("style") = 0,
         "v7 schedule-style seam: 0=base 1=in-burst setprio 2=late mem "
-        "groups 3=reordered-p0-head 4=all-glds-p0 5=3+4");
+        "groups 3=reordered-p0-head 4=all-glds-p0 5=3+4 "
+        "6=skeleton-no-glds 7=skeleton-no-glds-no-barrier "
+        "8=cross-barrier-pipelined (P)");
   m.def("gemm_stress_bf16_v7_sp", &gemm_stress_bf16_v7_sp,
         py::arg("size") = 8192, py::arg("iters") = 8,
         "v7 with static young-half setprio (A/B seam)");
