@@ -1358,6 +1358,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
 // tile's carried half runs in V7P_TAIL after the loop.
 // ---------------------------------------------------------------------------
 
+template <bool GFIRST = false>
 __global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
     const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1438,8 +1439,12 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
     uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;                    \
     uint32_t gdest = gdest_base + (uint32_t)nxt * 65536u;                    \
     uint32_t mscratch;                                                       \
-    V7P_PHASE0(acc, afrag, bfrag[BC], bfrag[BO], aaddr0, aaddr1, baddr0,     \
-               baddr1, voff, gdest, mscratch, gbase, gstride);               \
+    if constexpr (GFIRST)                                                    \
+      V7P2_PHASE0(acc, afrag, bfrag[BC], bfrag[BO], aaddr0, aaddr1, baddr0,  \
+                  baddr1, voff, gdest, mscratch, gbase, gstride);            \
+    else                                                                     \
+      V7P_PHASE0(acc, afrag, bfrag[BC], bfrag[BO], aaddr0, aaddr1, baddr0,   \
+                 baddr1, voff, gdest, mscratch, gbase, gstride);             \
     V7X_PHASE1(acc, afrag, bfrag[BC], aaddr0, aaddr1);                       \
     V7X_PHASE2(acc, afrag, bfrag[BC], aaddr0, aaddr1);                       \
     V7P_PHASE3A(acc, afrag, bfrag[BC]);                                      \
@@ -2060,7 +2065,8 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   else if (style == 5) kern = gemm_bf16_v7_kernel<false, 5>;
   else if (style == 6) kern = gemm_bf16_v7_kernel<false, 6>;
   else if (style == 7) kern = gemm_bf16_v7_kernel<false, 7>;
-  else if (style == 8) kern = gemm_bf16_v7p_kernel;
+  else if (style == 8) kern = gemm_bf16_v7p_kernel<false>;
+  else if (style == 9) kern = gemm_bf16_v7p_kernel<true>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
