@@ -1358,7 +1358,7 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7_kernel(
 // tile's carried half runs in V7P_TAIL after the loop.
 // ---------------------------------------------------------------------------
 
-template <bool GFIRST = false>
+template <bool GFIRST = false, bool NTSTORE = false>
 __global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
     const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
     float* __restrict__ C, int M, int N, int K) {
@@ -1460,7 +1460,10 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
   // the final tile's carried kh1 half (its B set is bfrag[1]: ntiles even)
   V7P_TAIL(acc, afrag, bfrag[1]);
 
-  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg.
+  // NTSTORE: C is written once and never re-read, but at 8192 its 256 MB
+  // of store traffic evicts the A+B working set (exactly the 256 MB LLC)
+  // — non-temporal stores keep the LLC for the operands.
 #pragma unroll
   for (int R = 0; R < 8; ++R) {
 #pragma unroll
@@ -1469,7 +1472,11 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v7p_kernel(
       for (int reg = 0; reg < 4; ++reg) {
         const int row = brow + wave_m * 128 + R * 16 + (lane >> 4) * 4 + reg;
         const int col = bcol + wave_n * 64 + c * 16 + (lane & 15);
-        C[(size_t)row * N + col] = acc[R][c][reg];
+        if constexpr (NTSTORE)
+          __builtin_nontemporal_store(acc[R][c][reg],
+                                      &C[(size_t)row * N + col]);
+        else
+          C[(size_t)row * N + col] = acc[R][c][reg];
       }
     }
   }
@@ -2067,6 +2074,7 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
   else if (style == 7) kern = gemm_bf16_v7_kernel<false, 7>;
   else if (style == 8) kern = gemm_bf16_v7p_kernel<false>;
   else if (style == 9) kern = gemm_bf16_v7p_kernel<true>;
+  else if (style == 10) kern = gemm_bf16_v7p_kernel<false, true>;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, 0, d_a, d_bt, d_c, M,
                      N, K);  // warmup
   HIP_CHECK(hipGetLastError());
