@@ -2127,11 +2127,13 @@ py::dict gemm_stress_bf16_v7_impl(int size, int iters, bool setprio,
 }
 
 py::dict gemm_stress_bf16_v7(int size, int iters) {
-  // Shipped default = style X (E reordered-p0-head + F all-glds-at-p0),
-  // the schedule-grid winner at both diag shapes (~1390-1400 TF vs 1352
-  // base, gemm_ab_v7c/v8 logs); static setprio stays off (measured -9%
-  // @8192 in the r1 A/B).
-  return gemm_stress_bf16_v7_impl(size, iters, false, 5);
+  // Shipped default = style P (cross-barrier-pipelined phase 3 on the
+  // E+F schedule), the 11-structure grid winner: ~1460-1490 TF @8192
+  // (PMC MfmaUtil 70%), ~1390-1430 @4096 (gemm_ab_v7p/v7p2/v7p3 logs).
+  // Rejected on A/B: in-burst + static setprio, late mem groups,
+  // glds-burst-first head, nt C stores, 3-buffer ring (v8), counted
+  // chunk pipeline (v6), register staging (v5).
+  return gemm_stress_bf16_v7_impl(size, iters, false, 8);
 }
 
 py::dict gemm_stress_bf16_v7_style(int size, int iters, int style) {

@@ -41,7 +41,7 @@ DEFAULT_BF16_TFLOPS_FLOOR = 1600.0
 DEFAULT_FP8_TFLOPS_FLOOR = 1600.0  # non-scaled fp8 runs at the bf16 rate
 DEFAULT_MXFP8_TFLOPS_FLOOR = 3400.0  # MX-scaled path, measured ~4790 TF
 DEFAULT_MXFP4_TFLOPS_FLOOR = 6500.0  # fp4 MX path, measured ~8730 TF
-DEFAULT_GEMM_TFLOPS_FLOOR = 950.0  # v7X asm-K-loop bf16 GEMM, measured ~1390
+DEFAULT_GEMM_TFLOPS_FLOOR = 1000.0  # v7P asm-K-loop bf16 GEMM, ~1390-1490
 DEFAULT_GEMM_FP8_TFLOPS_FLOOR = 1400.0  # 8-phase MX-fp8 GEMM, measured ~1940
 DEFAULT_HBM_GBPS_FLOOR = 4500.0
 DEFAULT_LDS_TBPS_FLOOR = 60.0
