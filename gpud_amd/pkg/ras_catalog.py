@@ -747,6 +747,151 @@ CATALOG: List[Detail] = [
         _HW,
         critical=True,
     ),
+    # ---- round-2b expansion: engine tests, firmware, fabric NICs, disks ----
+    # (appended AFTER the families above so first-match-wins keeps every
+    # earlier, more specific signature; each entry matcher-tested)
+    _d(
+        "amdgpu_ring_test_failed",
+        r"amdgpu.*ring (?P<ring>\S+) test failed",
+        "A ring's start-of-day test failed (gfx/sdma/vcn/jpeg engine did "
+        "not answer) — the engine is hung or the IP block failed init",
+        EventType.CRITICAL,
+        _REBOOT,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_ras_init_failed",
+        r"amdgpu.*RAS.*init.*fail",
+        "RAS subsystem failed to initialise — error telemetry is blind "
+        "until the driver reloads",
+        EventType.WARNING,
+        _REBOOT,
+    ),
+    _d(
+        "amdgpu_ras_ta_missing",
+        r"amdgpu.*ras ta ucode is not available",
+        "RAS trusted application firmware missing — RAS features degraded; "
+        "usually a ROCm/firmware packaging problem",
+        EventType.WARNING,
+    ),
+    _d(
+        "amdgpu_vbios_invalid",
+        r"amdgpu.*(?:Invalid VBIOS|BIOS signature incorrect)",
+        "Video BIOS image invalid or unsigned — board firmware corruption",
+        EventType.CRITICAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "amdgpu_gfxoff_failed",
+        r"amdgpu.*[Ff]ailed to (?:disable|enable) gfxoff",
+        "GFXOFF state transition failed — power-management degraded; "
+        "recurrence alongside SMU errors implicates the SMU firmware",
+        EventType.WARNING,
+    ),
+    _d(
+        "amdgpu_reg_write_failed",
+        r"amdgpu.*failed to write reg \S+ wait reg",
+        "Register write-and-confirm failed — the GFX core is not "
+        "responding (typically accompanies a hang/reset sequence)",
+        EventType.CRITICAL,
+        _REBOOT,
+    ),
+    _d(
+        "mlx5_device_error",
+        r"mlx5_core.*(?:health compromised|firmware internal error|"
+        r"assert_var|health buffer)",
+        "Fabric NIC (mlx5) device health error — firmware assert or "
+        "internal error; RDMA traffic on this port is suspect",
+        EventType.CRITICAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "mlx5_port_module_error",
+        r"mlx5_core.*Port module event\[error\]",
+        "Fabric NIC port module (cable/transceiver) error — check the "
+        "cable, transceiver seating and power budget",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "pcie_card_removed",
+        r"pciehp.*(?:Card not present|Surprise removal)",
+        "PCIe hotplug controller reports the card gone — a device "
+        "(possibly a GPU) dropped off the slot",
+        EventType.CRITICAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "host_thermal_critical_shutdown",
+        r"critical temperature reached.*shutting down",
+        "A thermal zone hit its critical trip point and the kernel is "
+        "shutting the machine down — cooling failure",
+        EventType.FATAL,
+        _HW,
+        critical=True,
+    ),
+    _d(
+        "host_disk_medium_error",
+        r"critical medium error, dev (?P<dev>\S+)",
+        "Disk medium error (unreadable sector) — the drive is failing; "
+        "check SMART and plan replacement",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_md_disk_failure",
+        r"md/raid.*Disk failure on (?P<dev>\S+)",
+        "Software-RAID member failed and was kicked from the array — "
+        "redundancy reduced; replace the member",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_jbd2_io_error",
+        r"JBD2: Detected IO errors",
+        "Journal layer saw I/O errors flushing data — filesystem may "
+        "degrade to read-only next; inspect the underlying device",
+        EventType.CRITICAL,
+        _HW,
+    ),
+    _d(
+        "host_acpi_error",
+        r"ACPI (?:BIOS )?Error[ :]",
+        "ACPI/BIOS error — firmware table or method problem; benign "
+        "recurrences are common but new onset after a BIOS update is not",
+        EventType.WARNING,
+    ),
+    _d(
+        "host_firmware_bug",
+        r"\[Firmware Bug\]:",
+        "Kernel flagged a platform firmware bug — record for correlation; "
+        "persistent new entries warrant a BIOS/BMC update",
+        EventType.WARNING,
+    ),
+    _d(
+        "host_clocksource_unstable",
+        r"clocksource.*Marking clocksource .* as unstable",
+        "TSC/clocksource marked unstable — timekeeping degraded, timers "
+        "and profiling results are suspect on this node",
+        EventType.WARNING,
+    ),
+    _d(
+        "host_tcp_oom",
+        r"TCP: out of memory",
+        "TCP stack out of memory — socket buffers exhausted under load; "
+        "tune tcp_mem or find the flood source",
+        EventType.WARNING,
+    ),
+    _d(
+        "host_conntrack_full",
+        r"nf_conntrack: .*table full, dropping packet",
+        "Connection-tracking table full — new connections are being "
+        "dropped; raise nf_conntrack_max or reduce connection churn",
+        EventType.WARNING,
+    ),
     # ---- ROCm user-space crash signatures ----------------------------------
     _d(
         "amd_hip_segfault_in_libamdhip",

@@ -639,6 +639,51 @@ def test_every_catalog_entry_matches_a_representative_line():
             "python[4242]: segfault at 10 ip 00007f1234567890 sp "
             "00007ffc12345678 error 4 in librocblas.so.4[7f1200000000+"
             "8000000]",
+        # round-2b expansion
+        "amdgpu_ring_test_failed":
+            "amdgpu 0000:0a:00.0: [drm] ring vcn_dec_0 test failed (-110)",
+        "amdgpu_ras_init_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS init failed (-22)",
+        "amdgpu_ras_ta_missing":
+            "amdgpu 0000:0a:00.0: amdgpu: RAS: optional ras ta ucode is "
+            "not available",
+        "amdgpu_vbios_invalid":
+            "amdgpu 0000:0a:00.0: Invalid VBIOS signature",
+        "amdgpu_gfxoff_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: Failed to disable gfxoff!",
+        "amdgpu_reg_write_failed":
+            "amdgpu 0000:0a:00.0: amdgpu: failed to write reg 28b4 wait "
+            "reg 28c6",
+        "mlx5_device_error":
+            "mlx5_core 0000:08:00.0: print_health_info:423:(pid 0): "
+            "firmware internal error detected",
+        "mlx5_port_module_error":
+            "mlx5_core 0000:08:00.0: Port module event[error]: module 0, "
+            "Cable error, Power budget exceeded",
+        "pcie_card_removed":
+            "pciehp 0000:00:01.1: pciehp: Slot(0): Card not present",
+        "host_thermal_critical_shutdown":
+            "thermal thermal_zone0: critical temperature reached (101 C), "
+            "shutting down",
+        "host_disk_medium_error":
+            "blk_update_request: critical medium error, dev sda, sector "
+            "1234567 op 0x0:(READ)",
+        "host_md_disk_failure":
+            "md/raid1:md0: Disk failure on sdb1, disabling device.",
+        "host_jbd2_io_error":
+            "JBD2: Detected IO errors while flushing file data on sda1-8",
+        "host_acpi_error":
+            "ACPI BIOS Error (bug): Could not resolve symbol "
+            "[\\_SB.PCI0.GPP0], AE_NOT_FOUND",
+        "host_firmware_bug":
+            "[Firmware Bug]: TSC doesn't count with P0 frequency!",
+        "host_clocksource_unstable":
+            "clocksource: timekeeping watchdog on CPU1: Marking "
+            "clocksource 'tsc' as unstable because the skew is too large:",
+        "host_tcp_oom":
+            "TCP: out of memory -- consider tuning tcp_mem",
+        "host_conntrack_full":
+            "nf_conntrack: nf_conntrack: table full, dropping packet",
     })
     missing = [d.name for d in CATALOG if d.name not in representatives]
     assert not missing, f"entries without representative lines: {missing}"
