@@ -284,3 +284,66 @@ def test_gpm_per_xcc_busy(mock_core):
     assert "accelerator_amd_gpm_xcc_busy_percent" in text
     assert "accelerator_amd_gpm_xcc_busy_spread_percent" in text
     assert "accelerator_amd_gpm_mm_activity_percent" in text
+
+
+def test_power_at_limit_degraded(mock_core):
+    """Sustained draw at >=98% of the enforced limit flips power to
+    Degraded (reference: power component's usage-vs-limit gauges —
+    components/accelerator/nvidia/power/component.go)."""
+    backend = mock_core.smi_instance._b
+    backend.state[0]["power_w"] = 1390  # limit 1400 -> 99.3%
+    mock_core.shared_snapshots.refresh()
+    cr = _check(mock_core, "accelerator-amd-power")
+    assert cr.health == HealthStateType.DEGRADED
+    assert "power limit" in cr.reason
+    backend.state[0]["power_w"] = 620
+    mock_core.shared_snapshots.refresh()
+    cr = _check(mock_core, "accelerator-amd-power")
+    assert cr.health == HealthStateType.HEALTHY
+
+
+def test_power_management_disabled_degraded(mock_core, monkeypatch):
+    """power-management mirrors the reference persistence-mode check:
+    informational Degraded when the mode is off (persistence-mode
+    component.go analog)."""
+    backend = mock_core.smi_instance._b
+    monkeypatch.setattr(
+        type(backend), "power_management_enabled", lambda self, i: i != 1
+    )
+    cr = _check(mock_core, "accelerator-amd-power-management")
+    assert cr.health == HealthStateType.DEGRADED
+    assert "disabled" in cr.reason
+
+
+def test_network_latency_unreachable_unhealthy(mock_core):
+    """All targets unreachable -> Unhealthy; one in-threshold target ->
+    Healthy (reference: network-latency global thresholds,
+    components/network/latency/component.go)."""
+    from gpud_amd.components.host.network_latency import NetworkLatencyComponent
+
+    comp = NetworkLatencyComponent(mock_core.gpud_instance)
+    comp.targets = [("198.51.100.1", 443), ("198.51.100.2", 443)]
+    comp.probe = lambda host, port: -1.0
+    cr = comp.check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert cr.extra_info["198.51.100.1:443"] == "unreachable"
+    comp.probe = lambda host, port: 3.5 if host.endswith(".1") else -1.0
+    cr = comp.check()
+    assert cr.health == HealthStateType.HEALTHY
+    assert "1/2 targets" in cr.reason
+
+
+def test_peer_mem_degraded_without_providers_or_dmabuf(mock_core, monkeypatch):
+    """RDMA NICs present but no PeerDirect provider and no DMABUF kernel
+    support -> Degraded (reference: peermem ko/lsmod check,
+    components/accelerator/nvidia/peermem/component.go)."""
+    comp = mock_core.registry.get("accelerator-amd-peer-mem")
+    monkeypatch.setattr(comp, "has_rdma_nics", lambda: True)
+    monkeypatch.setattr(comp, "get_providers", lambda: [])
+    monkeypatch.setattr(comp, "has_dmabuf", lambda: False)
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.DEGRADED
+    assert "bounce buffers" in cr.reason
+    monkeypatch.setattr(comp, "get_providers", lambda: ["amdgpu_peerdirect"])
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.HEALTHY
