@@ -21,4 +21,4 @@ components/registry.go) while staying idiomatic Python; the hot paths
 (SMI polling, diag kernels) are native C++/HIP.
 """
 
-__version__ = "0.2.0"
+__version__ = "0.3.0"
