@@ -1638,6 +1638,130 @@ __global__ __launch_bounds__(512, 2) void gemm_bf16_v8_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// GEMM v9: TWO BLOCKS PER CU. The register-resident stress reaches 97.5%
+// MfmaUtil because co-resident waves cover each other's stalls; v7P is
+// capped at ~70% because its single 512-thread block (128 KiB LDS,
+// 242 VGPR) owns the whole CU, so every barrier arrival and boundary
+// drain parks all 8 waves together. v9 halves the tile to 128x128x64
+// (32 KiB per LDS buffer, 64 KiB per block) and squeezes wave state to
+// ~104 VGPRs — wave tile 64x32: acc[4][2] + 24 fragment regs —
+// so two blocks co-reside (__launch_bounds__(512, 4) caps the
+// allocation at 128 VGPRs): when one block drains/barriers, the other
+// block's two waves per SIMD keep the XDL pipes fed. Costs measured
+// into the A/B: 2x L2/LLC operand traffic (tile reuse halves) and
+// double the boundary frequency per MFMA. LDS array headroom is known:
+// v7 measures 25% SQ_LDS_IDX_ACTIVE, so 2x DS traffic stays far from
+// the array limit. Body generated by scripts/gen_v9_asm.py (E+F
+// schedule: kh-grouped reads, incremental lgkm waits, all 4 glds
+// interleaved into the kh0 burst).
+// ---------------------------------------------------------------------------
+
+#include "gemm_v9_body.h"
+
+__global__ __launch_bounds__(512, 4) void gemm_bf16_v9_kernel(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ Bt,
+    float* __restrict__ C, int M, int N, int K) {
+  constexpr int TM = 128, TN = 128, TK = 64;
+  __shared__ __hip_bfloat16 lds[2][(TM + TN) * TK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wave_m = wave >> 2;  // 0..1 — 64-row band
+  const int wave_n = wave & 3;   // 0..3 — 32-col band
+  const int tiles_n = N / TN;
+  const int brow = (blockIdx.x / tiles_n) * TM;
+  const int bcol = (blockIdx.x % tiles_n) * TN;
+  const int ntiles = K / TK;
+
+  const uint32_t lds0 = (uint32_t)(uintptr_t)&lds[0][0];
+  const int lane_raw = (lane & 15) * 128 + ((lane >> 4) * 16);
+  const uint32_t swz_lane_k0 = (uint32_t)swz(lane_raw);
+  const uint32_t swz_lane_k1 = (uint32_t)swz(lane_raw + 64);
+
+  // staging: A image = 16 slots (1 KiB, 8 rows of 128 B), B image = 16
+  // slots at +16 KiB. Waves 0-3 stage A slots wave*4..+4, waves 4-7
+  // stage B slots (wave-4)*4..+4 — 4 glds per wave per tile, uniform.
+  const int glds_img = wave >> 2;
+  const int s_img0 = glds_img == 0 ? wave * 4 : (wave - 4) * 4;
+  const int D0 = s_img0 * 1024 + lane * 16;
+  const int L0 = swz(D0);
+  const int row0 = L0 >> 7;
+  const int kcol0 = (L0 & 127) >> 1;
+  const __hip_bfloat16* gptr =
+      glds_img == 0 ? A + (size_t)brow * K : Bt + (size_t)bcol * K;
+  uint64_t gbase;
+  {
+    const uint64_t p = (uint64_t)(uintptr_t)gptr;
+    const uint32_t lo = __builtin_amdgcn_readfirstlane((uint32_t)p);
+    const uint32_t hi = __builtin_amdgcn_readfirstlane((uint32_t)(p >> 32));
+    gbase = ((uint64_t)hi << 32) | lo;
+  }
+  const uint32_t gstride =
+      __builtin_amdgcn_readfirstlane((uint32_t)(8u * (uint32_t)K * 2u));
+  const uint32_t voff_lane = ((uint32_t)row0 * (uint32_t)K + kcol0) * 2u;
+  const uint32_t gdest_base = __builtin_amdgcn_readfirstlane(
+      lds0 + (uint32_t)glds_img * (uint32_t)(TM * TK * 2) +
+      (uint32_t)s_img0 * 1024u);
+
+  accfrag_t acc[4][2] = {};
+  bf16x8 afrag[4][2] = {};
+  bf16x8 bfrag[2][2] = {};
+
+  // prologue: stage tile 0, drain once
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int s_img = s_img0 + j;
+    const int D = s_img * 1024 + lane * 16;
+    const int L = swz(D);
+    const int row = L >> 7;
+    const int k = (L & 127) >> 1;
+    const __hip_bfloat16* g = gptr + (size_t)row * K + k;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(
+            &lds[0][glds_img * TM * TK] + s_img * 512),
+        16, 0, 0);
+  }
+  __builtin_amdgcn_s_waitcnt(0x3F70);  // vmcnt(0)
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int cur = kt & 1;
+    const int nxt = cur ^ 1;
+    const int kt_next = (kt + 1 < ntiles) ? kt + 1 : 0;
+    const uint32_t abase =
+        lds0 + (uint32_t)cur * 32768u + (uint32_t)wave_m * 8192u;
+    const uint32_t bbase = lds0 + (uint32_t)cur * 32768u + 16384u +
+                           (uint32_t)wave_n * 4096u;
+    uint32_t aaddr0 = abase + swz_lane_k0;
+    uint32_t aaddr1 = abase + swz_lane_k1;
+    uint32_t baddr0 = bbase + swz_lane_k0;
+    uint32_t baddr1 = bbase + swz_lane_k1;
+    uint32_t voff = voff_lane + (uint32_t)kt_next * 128u;
+    uint32_t gdest = gdest_base + (uint32_t)nxt * 32768u;
+    uint32_t mscratch;
+    V9_TILE(acc, afrag, bfrag, aaddr0, aaddr1, baddr0, baddr1, voff, gdest,
+            mscratch, gbase, gstride);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: 16x16x32 C/D layout — col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int rr = 0; rr < 4; ++rr) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = brow + wave_m * 64 + rr * 16 + (lane >> 4) * 4 + reg;
+        const int col = bcol + wave_n * 32 + c * 16 + (lane & 15);
+        C[(size_t)row * N + col] = acc[rr][c][reg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM v5: register-staged K-loop — NO LDS, NO barriers. Every wave loads
 // its own MFMA fragments straight from global memory (both operands are
 // K-contiguous, so a 16x16x32 fragment is 16 contiguous bytes per lane:
@@ -2148,6 +2272,70 @@ py::dict gemm_stress_bf16_v7_nosp(int size, int iters) {
   return gemm_stress_bf16_v7_impl(size, iters, false);  // alias of default
 }
 
+py::dict gemm_stress_bf16_v9(int size, int iters) {
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 128) * (N / 128);
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  hipLaunchKernelGGL(gemm_bf16_v9_kernel, dim3(blocks), dim3(512), 0, 0, d_a,
+                     d_bt, d_c, M, N, K);  // warmup
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipEventRecord(t0));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(gemm_bf16_v9_kernel, dim3(blocks), dim3(512), 0, 0,
+                       d_a, d_bt, d_c, M, N, K);
+  }
+  HIP_CHECK(hipEventRecord(t1));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int s = 0; s < sample; ++s)
+      idx[s] = ((size_t)s * 2654435761u) % ((size_t)M * N);
+    for (int s = 0; s < sample; ++s) {
+      HIP_CHECK(hipMemcpy(&host[s], d_c + idx[s], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[s] / N), j = (int)(idx[s] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[s] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "128sq-2blocks-per-cu";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_bf16_v8(int size, int iters) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -2585,6 +2773,10 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v7", &gemm_stress_bf16_v7, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
+  m.def("gemm_stress_bf16_v9", &gemm_stress_bf16_v9, py::arg("size") = 8192,
+        py::arg("iters") = 8,
+        "2-blocks-per-CU bf16 GEMM stress (v9): 128sq tile, cross-block "
+        "stall cover");
   m.def("gemm_stress_bf16_v8", &gemm_stress_bf16_v8, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "3-buffer glds-ring bf16 GEMM stress (v8): counted vmcnt boundary, "
