@@ -113,9 +113,12 @@ def test_diag_gemm_bf16_verified():
     res2x = _diag.gemm_stress_bf16_v7(size=4096, iters=2)
     assert res2x["verified"], res2x
     assert res2x["tflops"] > 950, res2x
-    # the 3-buffer ring (v8) stays as a structural seam: correct, slower
+    # the structural seams stay correct: v8 (3-buffer ring) and v9
+    # (2 blocks/CU) both verify even where they lose the A/B
     res8 = _diag.gemm_stress_bf16_v8(size=1024, iters=2)
     assert res8["verified"], res8
+    res9 = _diag.gemm_stress_bf16_v9(size=1024, iters=2)
+    assert res9["verified"], res9
     res3 = _diag.gemm_stress_mxfp8(size=4096, iters=2)
     assert res3["verified"], res3
     assert res3["tflops"] > 1400, res3
