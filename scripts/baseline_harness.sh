@@ -26,8 +26,8 @@ out["mfma_bf16"] = _diag.mfma_stress_bf16(iters=2048, workgroups=1024)
 out["mfma_fp8"] = _diag.mfma_stress_fp8(iters=2048, workgroups=1024)
 out["mfma_mxfp8"] = _diag.mfma_stress_mxfp8(iters=2048, workgroups=1024)
 out["mfma_mxfp4"] = _diag.mfma_stress_mxfp4(iters=2048, workgroups=1024)
-out["gemm_bf16_v2_4096"] = _diag.gemm_stress_bf16_v2(size=4096, iters=5)
-out["gemm_bf16_v2_8192"] = _diag.gemm_stress_bf16_v2(size=8192, iters=5)
+out["gemm_bf16_v7p_4096"] = _diag.gemm_stress_bf16_v7(size=4096, iters=5)
+out["gemm_bf16_v7p_8192"] = _diag.gemm_stress_bf16_v7(size=8192, iters=5)
 out["gemm_mxfp8_8192"] = _diag.gemm_stress_mxfp8(size=8192, iters=5)
 out["hbm"] = _diag.hbm_bandwidth(buffer_gb=4.0, iters=8)
 out["lds"] = _diag.lds_bandwidth(iters=20000, workgroups=512)
