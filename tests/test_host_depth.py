@@ -371,3 +371,49 @@ def mock_core(monkeypatch, tmp_path):
     )
     yield core
     core.close()
+
+
+def test_memory_bpf_jit_and_vmalloc(tmp_path):
+    """vmallocinfo bpf_jit summing + meminfo Vmalloc fields (reference:
+    components/memory/bpf.go:40-60 and component.go:192-193)."""
+    from gpud_amd.components.host.memory import (
+        read_bpf_jit_buffer_bytes,
+        read_vmalloc_meminfo,
+    )
+
+    vmi = tmp_path / "vmallocinfo"
+    vmi.write_text(
+        "0xffffc90000000000-0xffffc90000005000   20480 "
+        "irq_init_percpu_irqstack+0x176/0x1c0 vmap\n"
+        "0xffffc900000b0000-0xffffc900000b2000    8192 "
+        "bpf_jit_alloc_exec+0xe/0x20 pages=1 vmalloc N0=1\n"
+        "0xffffc900000c0000-0xffffc900000c3000   12288 "
+        "bpf_jit_alloc_exec+0xe/0x20 pages=2 vmalloc N0=2\n"
+        "garbage line\n"
+    )
+    assert read_bpf_jit_buffer_bytes(str(vmi)) == 8192 + 12288
+    assert read_bpf_jit_buffer_bytes(str(tmp_path / "missing")) is None
+
+    mi = tmp_path / "meminfo"
+    mi.write_text(
+        "MemTotal:       2113558860 kB\n"
+        "VmallocTotal:   34359738367 kB\n"
+        "VmallocUsed:        2344700 kB\n"
+    )
+    total, used = read_vmalloc_meminfo(str(mi))
+    assert total == 34359738367 * 1024
+    assert used == 2344700 * 1024
+
+
+def test_memory_component_reports_bpf_jit(mock_core, tmp_path):
+    from gpud_amd.components.host.memory import MemoryComponent
+
+    comp = mock_core.registry.get("memory")
+    vmi = tmp_path / "vmallocinfo"
+    vmi.write_text(
+        "0xffffc900000b0000-0xffffc900000b2000    8192 "
+        "bpf_jit_alloc_exec+0xe/0x20 pages=1 vmalloc N0=1\n"
+    )
+    comp.vmallocinfo_path = str(vmi)
+    cr = comp.trigger_check()
+    assert cr.extra_info["bpf_jit_buffer_bytes"] == "8192"
