@@ -3,10 +3,14 @@
 Reference: components/pci (lspci ACS status on bridges; ACS-on hurts GPU
 P2P — pci/component.go:26). On MI355X nodes ACS on the bridges above the
 GPUs forces P2P/xGMI-adjacent PCIe DMA through the root complex.
+The check is virtualization-gated like the reference
+(pci/component.go:159-168): inside KVM guests the hypervisor owns the
+topology and ACS state is expected, so the check is skipped there.
 """
 
 from __future__ import annotations
 
+import shutil
 import subprocess
 from typing import Callable, List, Optional
 
@@ -56,6 +60,25 @@ def bridges_with_acs_enabled(lspci_command: str = "") -> Optional[List[str]]:
     return parse_acs_bridges(out.stdout)
 
 
+def detect_virt_env() -> str:
+    """`systemd-detect-virt --vm` output: "none" on bare metal (and in
+    plain containers on bare metal), "kvm"/"qemu"/"vmware"/... inside VM
+    guests, "" when undeterminable (reference:
+    pkg/host/virtualization_environment.go:21 — the VM field drives the
+    ACS skip, a container runtime alone does not)."""
+    path = shutil.which("systemd-detect-virt")
+    if path is None:
+        return ""
+    try:
+        out = subprocess.run(
+            [path, "--vm"], capture_output=True, text=True, timeout=10
+        )
+        # exit code 1 means "none" — stdout still carries the answer
+        return (out.stdout or "").strip()
+    except (OSError, subprocess.TimeoutExpired):
+        return ""
+
+
 class PCIComponent(TickerComponent):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
@@ -63,6 +86,7 @@ class PCIComponent(TickerComponent):
         self.get_acs_bridges: Callable = lambda: bridges_with_acs_enabled(
             self._lspci_command
         )
+        self.get_virt_env: Callable = detect_virt_env
 
     @property
     def name(self) -> str:
@@ -72,6 +96,13 @@ class PCIComponent(TickerComponent):
         return [NAME]
 
     def check(self) -> CheckResult:
+        virt = self.get_virt_env()
+        if virt and virt != "none":
+            return CheckResult(
+                NAME,
+                reason=f"host virt env is {virt} (no need to check ACS)",
+                extra_info={"virt_env": virt},
+            )
         bridges = self.get_acs_bridges()
         if bridges is None:
             return CheckResult(NAME, reason="lspci unavailable; ACS check skipped")

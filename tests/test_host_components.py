@@ -46,6 +46,7 @@ def test_pci_acs_parse():
 
 def test_pci_component_degraded_on_acs(mock_core):
     comp = mock_core.registry.get("pci")
+    comp.get_virt_env = lambda: "none"  # ACS check is virt-gated
     comp.get_acs_bridges = lambda: ["00:01.0"]
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.DEGRADED

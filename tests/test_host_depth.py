@@ -417,3 +417,41 @@ def test_memory_component_reports_bpf_jit(mock_core, tmp_path):
     comp.vmallocinfo_path = str(vmi)
     cr = comp.trigger_check()
     assert cr.extra_info["bpf_jit_buffer_bytes"] == "8192"
+
+
+def test_pci_acs_skipped_in_vm(mock_core):
+    """ACS is expected inside VM guests — the check is virt-gated like
+    the reference (pci/component.go:159-168)."""
+    comp = mock_core.registry.get("pci")
+    comp.get_virt_env = lambda: "kvm"
+    comp.get_acs_bridges = lambda: ["00:01.1"]
+    cr = comp.trigger_check()
+    assert cr.health == "Healthy"
+    assert "kvm" in cr.reason
+    comp.get_virt_env = lambda: "none"
+    cr = comp.trigger_check()
+    assert cr.health == "Degraded"
+
+
+def test_tailscale_service_and_backend_states(mock_core):
+    """Installed + inactive service -> Unhealthy; active but BackendState
+    != Running -> Unhealthy; Running -> Healthy (reference:
+    tailscale/component.go:114-145)."""
+    from gpud_amd.components.host.tailscale import TailscaleComponent
+
+    comp = TailscaleComponent(mock_core.gpud_instance)
+    comp.get_version = lambda: "1.62.0"
+    comp.get_service_active = lambda: False
+    comp.get_backend_state = lambda: None
+    cr = comp.check()
+    assert cr.health == "Unhealthy" and "not active" in cr.reason
+    comp.get_service_active = lambda: True
+    comp.get_backend_state = lambda: "NeedsLogin"
+    cr = comp.check()
+    assert cr.health == "Unhealthy" and "NeedsLogin" in cr.reason
+    comp.get_backend_state = lambda: "Running"
+    cr = comp.check()
+    assert cr.health == "Healthy" and "Running" in cr.reason
+    # not installed stays Healthy (reference semantics)
+    comp.get_version = lambda: None
+    assert comp.check().health == "Healthy"
