@@ -18,9 +18,16 @@ Reference: components/disk + pkg/disk. Mechanism parity items:
 
 Every external probe is a function field so tests inject fakes (the
 reference's injected-function-field pattern).
+
+Disk-health kernel messages land in THIS component's event bucket
+(reference: disk/kmsg_matcher.go — RAID array failure, filesystem
+remounted read-only, NVMe path/timeout/disable, beyond-end-of-device,
+buffer I/O and superblock write errors).
 """
 
 from __future__ import annotations
+
+import re
 
 import json
 import subprocess
@@ -30,10 +37,43 @@ from typing import Callable, Dict, List, Optional
 import psutil
 
 from ...apiv1.types import HealthStateType
+from ...pkg.kmsg.syncer import MatchResult, Syncer
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 from ..metrics_util import ComponentGauges
 
 NAME = "disk"
+
+# kmsg matcher (reference: components/disk/kmsg_matcher.go constants)
+_KMSG_RULES = (
+    ("raid_array_failure",
+     re.compile(r"md/raid.*: Disk failure on .* detected, failing array"),
+     "Critical"),
+    ("filesystem_read_only",
+     re.compile(r"Remounting filesystem read-only"), "Critical"),
+    ("nvme_path_failure",
+     re.compile(r"block nvme.*: no available path - failing I/O"),
+     "Critical"),
+    ("nvme_controller_timeout",
+     re.compile(r"nvme nvme\d+: I/O .* timeout, reset controller"),
+     "Warning"),
+    ("nvme_device_disabled",
+     re.compile(r"nvme nvme\d+: Disabling device after reset failure"),
+     "Critical"),
+    ("beyond_end_of_device",
+     re.compile(r"attempt to access beyond end of device"), "Warning"),
+    ("buffer_io_error",
+     re.compile(r"Buffer I/O error on dev \S+, logical block \d+"),
+     "Warning"),
+    ("superblock_write_error",
+     re.compile(r"I/O error while writing superblock"), "Critical"),
+)
+
+
+def match_disk_kmsg(line):
+    for name, rx, event_type in _KMSG_RULES:
+        if rx.search(line):
+            return MatchResult(name=name, event_type=event_type, message=line)
+    return None
 
 DEFAULT_USED_PERCENT_DEGRADED = 90.0
 DEFAULT_USED_PERCENT_UNHEALTHY = 98.0
@@ -185,6 +225,12 @@ class DiskComponent(TickerComponent):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
+        self._bucket = (
+            inst.event_store.bucket(NAME)
+            if inst.event_store is not None else None
+        )
+        self._kmsg = inst.kmsg_reader
+        self._syncer = None
         self.mount_points = list(inst.mount_points or ["/"])
         self.mount_targets = list(inst.mount_targets or [])
         self._lsblk_command = inst.lsblk_command
@@ -210,6 +256,16 @@ class DiskComponent(TickerComponent):
 
     def tags(self) -> list:
         return [NAME]
+
+    def start(self) -> None:
+        # disk-health kmsg events into this component's bucket
+        # (reference: disk/kmsg_matcher.go wiring in disk/component.go)
+        if self._kmsg is not None and self._bucket is not None:
+            self._syncer = Syncer(self._kmsg, match_disk_kmsg, self._bucket)
+        super().start()
+
+    def events(self, since):
+        return self._bucket.get(since) if self._bucket is not None else []
 
     def check(self) -> CheckResult:
         degraded, unhealthy, missing = [], [], []

@@ -1,8 +1,11 @@
-"""nfs — NFS member-visibility checker.
+"""nfs — NFS member-visibility checker + client-health kernel events.
 
 Reference: components/nfs + pkg/nfs-checker (write/read a per-machine file
 in each configured NFS group dir, verify member visibility; configs pushed
-from the control plane — nfs/component.go:31).
+from the control plane — nfs/component.go:31). NFS client kernel
+messages land in this component's bucket (reference:
+nfs/kmsg_matcher.go — server not responding / recovered, lock-reclaim
+failure, writeback-path hang stack frames).
 """
 
 from __future__ import annotations
@@ -11,11 +14,40 @@ import os
 from dataclasses import dataclass
 from typing import List
 
+import re
+
 from ...apiv1.types import HealthStateType
 from ...pkg import host as pkghost
+from ...pkg.kmsg.syncer import MatchResult, Syncer
 from ..base import CheckResult, Component, GPUdInstance, TickerComponent
 
 NAME = "nfs"
+
+# kmsg matcher (reference: components/nfs/kmsg_matcher.go)
+_KMSG_RULES = (
+    ("nfs_server_not_responding",
+     re.compile(r"nfs: server (?P<server>\S+) not responding"), "Warning"),
+    ("nfs_server_ok",
+     re.compile(r"nfs: server (?P<server>\S+) OK"), "Info"),
+    ("nfs_lock_reclaim_failed",
+     re.compile(r"nfs4_reclaim_open_state: Lock reclaim failed"),
+     "Warning"),
+    ("nfs_writeback_hang",
+     re.compile(r"(?:^|\s)(?:nfs_lock_and_join_requests|nfs_wb_all|"
+                r"nfs_page_async_flush|nfs_writepages_callback)"
+                r"\+0x[0-9a-f]+"),
+     "Critical"),
+)
+
+
+def match_nfs_kmsg(line):
+    for name, rx, event_type in _KMSG_RULES:
+        m = rx.search(line)
+        if m:
+            extra = {k: v for k, v in m.groupdict().items() if v}
+            return MatchResult(name=name, event_type=event_type,
+                               message=line, extra_info=extra or None)
+    return None
 
 
 @dataclass
@@ -30,6 +62,12 @@ class GroupConfig:
 class NFSComponent(TickerComponent):
     def __init__(self, inst: GPUdInstance):
         super().__init__()
+        self._bucket = (
+            inst.event_store.bucket(NAME)
+            if inst.event_store is not None else None
+        )
+        self._kmsg = inst.kmsg_reader
+        self._syncer = None
         self.configs: List[GroupConfig] = [
             c if isinstance(c, GroupConfig) else GroupConfig(**c)
             for c in (inst.nfs_checker_configs or [])
@@ -39,6 +77,14 @@ class NFSComponent(TickerComponent):
     @property
     def name(self) -> str:
         return NAME
+
+    def start(self) -> None:
+        if self._kmsg is not None and self._bucket is not None:
+            self._syncer = Syncer(self._kmsg, match_nfs_kmsg, self._bucket)
+        super().start()
+
+    def events(self, since):
+        return self._bucket.get(since) if self._bucket is not None else []
 
     def tags(self) -> list:
         return [NAME]

@@ -455,3 +455,70 @@ def test_tailscale_service_and_backend_states(mock_core):
     # not installed stays Healthy (reference semantics)
     comp.get_version = lambda: None
     assert comp.check().health == "Healthy"
+
+
+def test_disk_kmsg_matcher_families():
+    """Disk-health kernel messages are attributed to the disk component's
+    bucket (reference: disk/kmsg_matcher.go event names)."""
+    from gpud_amd.components.host.disk import match_disk_kmsg
+
+    cases = {
+        "md/raid1:md0: Disk failure on sdb1 detected, failing array":
+            "raid_array_failure",
+        "EXT4-fs (sda1): Remounting filesystem read-only":
+            "filesystem_read_only",
+        "block nvme0n1: no available path - failing I/O":
+            "nvme_path_failure",
+        "nvme nvme0: I/O 564 QID 7 timeout, reset controller":
+            "nvme_controller_timeout",
+        "nvme nvme0: Disabling device after reset failure: -19":
+            "nvme_device_disabled",
+        "attempt to access beyond end of device sda1":
+            "beyond_end_of_device",
+        "Buffer I/O error on dev sda1, logical block 1234, lost async page "
+        "write": "buffer_io_error",
+        "EXT4-fs (sda1): I/O error while writing superblock":
+            "superblock_write_error",
+    }
+    for line, expect in cases.items():
+        res = match_disk_kmsg(line)
+        assert res is not None and res.name == expect, (line, res)
+    assert match_disk_kmsg("usb 1-1: new device") is None
+
+
+def test_nfs_kmsg_matcher_families():
+    from gpud_amd.components.host.nfs import match_nfs_kmsg
+
+    res = match_nfs_kmsg("nfs: server fileserver01 not responding, "
+                         "still trying")
+    assert res.name == "nfs_server_not_responding"
+    assert res.extra_info == {"server": "fileserver01"}
+    assert match_nfs_kmsg("nfs: server fileserver01 OK").name == \
+        "nfs_server_ok"
+    assert match_nfs_kmsg(
+        "nfs4_reclaim_open_state: Lock reclaim failed!").name == \
+        "nfs_lock_reclaim_failed"
+    assert match_nfs_kmsg(
+        " nfs_wb_all+0x1c/0x120 [nfs]").name == "nfs_writeback_hang"
+    assert match_nfs_kmsg("nfs: mounted ok") is None
+
+
+def test_disk_kmsg_events_reach_component_bucket(mock_core):
+    """Replay a disk-failure line through the disk component's syncer and
+    read it back via the component events() surface."""
+    import datetime
+
+    from gpud_amd.apiv1.types import utcnow
+    from gpud_amd.components.host.disk import match_disk_kmsg
+    from gpud_amd.pkg.kmsg.syncer import Syncer
+    from gpud_amd.pkg.kmsg.watcher import Message, Watcher
+
+    comp = mock_core.registry.get("disk")
+    assert comp._bucket is not None
+    syn = Syncer(Watcher(path="/nonexistent"), match_disk_kmsg, comp._bucket)
+    now = utcnow()
+    syn.replay([Message(
+        message="Buffer I/O error on dev sda1, logical block 99, lost "
+                "async page write", time=now)])
+    evs = comp.events(now - datetime.timedelta(minutes=1))
+    assert evs and evs[0].name == "buffer_io_error"
