@@ -40,6 +40,39 @@ _KMSG_RULES = (
 )
 
 
+def collect_nfs_hang_events(events):
+    """(hang_events, reason) from a window of this component's kmsg
+    events: every lock-reclaim failure, every writeback-path hang trace,
+    and every server whose LAST response event is still not-responding
+    (a later "server OK" resolves it) — reference:
+    nfs/hang_evaluator.go collectNFSHangEvents."""
+    lock = [e for e in events if e.name == "nfs_lock_reclaim_failed"]
+    wb = [e for e in events if e.name == "nfs_writeback_hang"]
+    by_server = {}
+    server_rx = re.compile(r"nfs: server (\S+)")
+    for e in events:
+        if e.name in ("nfs_server_not_responding", "nfs_server_ok"):
+            m = server_rx.search(e.message or "")
+            server = m.group(1) if m else "?"
+            by_server.setdefault(server, []).append(e)
+    hang = list(lock) + list(wb)
+    parts = []
+    if lock:
+        parts.append(f"{len(lock)} lock reclaim failure(s)")
+    unresolved = []
+    for server, evs in sorted(by_server.items()):
+        evs_sorted = sorted(evs, key=lambda e: e.time)
+        if evs_sorted and evs_sorted[-1].name == "nfs_server_not_responding":
+            unresolved.append(server)
+            hang.extend(e for e in evs_sorted
+                        if e.name == "nfs_server_not_responding")
+    if unresolved:
+        parts.append("server(s) not responding: " + ", ".join(unresolved))
+    if wb:
+        parts.append(f"{len(wb)} writeback-path hang trace(s)")
+    return hang, "; ".join(parts)
+
+
 def match_nfs_kmsg(line):
     for name, rx, event_type in _KMSG_RULES:
         m = rx.search(line)
@@ -124,6 +157,20 @@ class NFSComponent(TickerComponent):
                 reason="; ".join(problems),
                 extra_info=extra,
             )
+        if self._bucket is not None:
+            import datetime as _dt
+
+            window = _dt.datetime.now(_dt.timezone.utc) - _dt.timedelta(
+                minutes=10)
+            hang, hang_reason = collect_nfs_hang_events(
+                self._bucket.get(window))
+            if hang:
+                return CheckResult(
+                    NAME,
+                    health=HealthStateType.DEGRADED,
+                    reason="NFS client hang indicators: " + hang_reason,
+                    extra_info=extra,
+                )
         return CheckResult(
             NAME,
             reason=f"all {len(self.configs)} NFS group dir(s) writable and readable",
