@@ -879,6 +879,15 @@ CATALOG: List[Detail] = [
         EventType.WARNING,
     ),
     _d(
+        "host_vfs_file_max_reached",
+        r"VFS: file-max limit \d+ reached",
+        "System-wide open-file-handle limit reached — new opens fail "
+        "everywhere until handles are released or fs.file-max is raised "
+        "(reference: os/kmsg_matcher.go VFS file-max event)",
+        EventType.CRITICAL,
+        _APP,
+    ),
+    _d(
         "host_tcp_oom",
         r"TCP: out of memory",
         "TCP stack out of memory — socket buffers exhausted under load; "

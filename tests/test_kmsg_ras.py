@@ -680,6 +680,8 @@ def test_every_catalog_entry_matches_a_representative_line():
         "host_clocksource_unstable":
             "clocksource: timekeeping watchdog on CPU1: Marking "
             "clocksource 'tsc' as unstable because the skew is too large:",
+        "host_vfs_file_max_reached":
+            "VFS: file-max limit 9223372036854775807 reached",
         "host_tcp_oom":
             "TCP: out of memory -- consider tuning tcp_mem",
         "host_conntrack_full":
