@@ -2272,6 +2272,83 @@ py::dict gemm_stress_bf16_v7_nosp(int size, int iters) {
   return gemm_stress_bf16_v7_impl(size, iters, false);  // alias of default
 }
 
+py::dict gemm_stress_bf16_v7_graph(int size, int iters) {
+  // v7P timed through ONE hipGraph of `iters` back-to-back launches —
+  // removes the per-launch submission gaps from the timed region
+  // (identical kernels and work; the guide's §6 launch-bound-loop lever)
+  if (size % 256 != 0 || size < 512 || size > 16384)
+    throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
+  if (iters <= 0 || iters > 100) throw std::invalid_argument("iters");
+  const int M = size, N = size, K = size;
+  __hip_bfloat16 *d_a = nullptr, *d_bt = nullptr;
+  float* d_c = nullptr;
+  HIP_CHECK(hipMalloc(&d_a, (size_t)M * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_bt, (size_t)N * K * sizeof(__hip_bfloat16)));
+  HIP_CHECK(hipMalloc(&d_c, (size_t)M * N * sizeof(float)));
+  hipLaunchKernelGGL(gemm_fill_kernel, dim3(2048), dim3(256), 0, 0, d_a, d_bt,
+                     M, N, K);
+  HIP_CHECK(hipGetLastError());
+  HIP_CHECK(hipDeviceSynchronize());
+  const int blocks = (M / 256) * (N / 256);
+  auto* kern = gemm_bf16_v7p_kernel<false>;
+  hipStream_t s;
+  HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeGlobal));
+  for (int i = 0; i < iters; ++i) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(512), 0, s, d_a, d_bt, d_c,
+                       M, N, K);
+  }
+  hipGraph_t graph = nullptr;
+  HIP_CHECK(hipStreamEndCapture(s, &graph));
+  hipGraphExec_t ge = nullptr;
+  HIP_CHECK(hipGraphInstantiate(&ge, graph, nullptr, nullptr, 0));
+  HIP_CHECK(hipGraphLaunch(ge, s));  // warmup replay
+  HIP_CHECK(hipStreamSynchronize(s));
+  hipEvent_t t0, t1;
+  HIP_CHECK(hipEventCreate(&t0));
+  HIP_CHECK(hipEventCreate(&t1));
+  HIP_CHECK(hipEventRecord(t0, s));
+  HIP_CHECK(hipGraphLaunch(ge, s));
+  HIP_CHECK(hipEventRecord(t1, s));
+  HIP_CHECK(hipEventSynchronize(t1));
+  float ms = 0.f;
+  HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+  size_t bad = 0;
+  {
+    const int sample = 509;
+    std::vector<float> host(sample);
+    std::vector<size_t> idx(sample);
+    for (int ss = 0; ss < sample; ++ss)
+      idx[ss] = ((size_t)ss * 2654435761u) % ((size_t)M * N);
+    for (int ss = 0; ss < sample; ++ss) {
+      HIP_CHECK(hipMemcpy(&host[ss], d_c + idx[ss], sizeof(float),
+                          hipMemcpyDeviceToHost));
+      const int i = (int)(idx[ss] / N), j = (int)(idx[ss] % N);
+      const float expect =
+          (float)K * (0.25f * ((i % 5) + 1)) * (0.125f * ((j % 7) + 1));
+      if (host[ss] != expect) bad++;
+    }
+  }
+  HIP_CHECK(hipGraphExecDestroy(ge));
+  HIP_CHECK(hipGraphDestroy(graph));
+  HIP_CHECK(hipStreamDestroy(s));
+  HIP_CHECK(hipFree(d_a));
+  HIP_CHECK(hipFree(d_bt));
+  HIP_CHECK(hipFree(d_c));
+  HIP_CHECK(hipEventDestroy(t0));
+  HIP_CHECK(hipEventDestroy(t1));
+  const double flops = (double)iters * 2.0 * M * (double)N * K;
+  py::dict d;
+  d["dtype"] = "bf16";
+  d["size"] = size;
+  d["structure"] = "256sq-asm-kloop-xbarrier-pipelined-hipgraph";
+  d["tflops"] = flops / (ms * 1e-3) / 1e12;
+  d["seconds_per_gemm"] = ms * 1e-3 / iters;
+  d["verify_failures"] = (long)bad;
+  d["verified"] = (bad == 0);
+  return d;
+}
+
 py::dict gemm_stress_bf16_v9(int size, int iters) {
   if (size % 256 != 0 || size < 512 || size > 16384)
     throw std::invalid_argument("size must be a multiple of 256 in [512,16384]");
@@ -2773,6 +2850,10 @@ PYBIND11_MODULE(_diag, m) {
   m.def("gemm_stress_bf16_v7", &gemm_stress_bf16_v7, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "hand-scheduled asm K-loop bf16 GEMM stress (v7)");
+  m.def("gemm_stress_bf16_v7_graph", &gemm_stress_bf16_v7_graph,
+        py::arg("size") = 8192, py::arg("iters") = 8,
+        "v7P timed through one hipGraph of iters launches (no submission "
+        "gaps in the timed region)");
   m.def("gemm_stress_bf16_v9", &gemm_stress_bf16_v9, py::arg("size") = 8192,
         py::arg("iters") = 8,
         "2-blocks-per-CU bf16 GEMM stress (v9): 128sq tile, cross-block "
