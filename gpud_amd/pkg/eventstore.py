@@ -60,6 +60,11 @@ class Bucket:
 
     def _create_table(self) -> None:
         t = self.table
+        # the v0_4_0-era table is unused — dropped at open like the
+        # reference (eventstore/database.go:97-103), so a gpud.state file
+        # carried over from an old reference install opens cleanly
+        legacy = default_table_name(self.component_name, "v0_4_0")
+        self._db_rw.executescript(f"DROP TABLE IF EXISTS {legacy};")
         self._db_rw.executescript(
             f"""
 CREATE TABLE IF NOT EXISTS {t} (

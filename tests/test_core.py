@@ -323,3 +323,21 @@ def test_metrics_store_component_and_since_filters(mem_db):
     both = store.read(since=now - datetime.timedelta(minutes=30),
                       components=["memory"])
     assert set(both) == {"memory"}
+
+
+def test_eventstore_drops_legacy_v0_4_0_table(mem_db):
+    """A reference-era v0_4_0 table in a carried-over gpud.state is
+    dropped at bucket open (reference: eventstore/database.go:97-103)."""
+    from gpud_amd.pkg.eventstore import Store, default_table_name
+
+    rw, ro = mem_db
+    legacy = default_table_name("error-ras", "v0_4_0")
+    rw.executescript(f"CREATE TABLE {legacy} (timestamp INTEGER);")
+    store = Store(rw, ro)
+    store.bucket("error-ras", disable_purge=True)
+    rows = ro.query(
+        "SELECT name FROM sqlite_master WHERE type='table' AND name=?",
+        (legacy,),
+    )
+    assert rows == []
+    store.close()
