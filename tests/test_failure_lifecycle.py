@@ -172,3 +172,52 @@ def test_per_event_escalation_threshold(monkeypatch, tmp_path):
         ], cr.suggested_actions
     finally:
         core.close()
+
+
+def test_disk_and_nfs_kmsg_through_file_seam(tmp_path, monkeypatch):
+    """Injected disk/NFS kernel lines travel the full seam path —
+    FileSeamWriter -> watcher poll-follow -> per-component syncer ->
+    component event bucket -> component events() — proving the new
+    per-component matchers are wired into the daemon, not just unit-level
+    (reference: disk/nfs kmsg_matcher wiring in their components)."""
+    import time
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.apiv1.types import utcnow
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg.config import Config
+    from gpud_amd.pkg.fault_injector import KernelMessage, Request
+
+    seam = tmp_path / "kmsg-seam"
+    seam.write_text("")
+    cfg = Config(data_dir=str(tmp_path / "data"), kmsg_path=str(seam))
+    core = build_core(cfg, in_memory_db=True)
+    try:
+        disk = core.registry.get("disk")
+        nfs = core.registry.get("nfs")
+        disk.start()
+        nfs.start()
+        core.kmsg_watcher.start(from_start=True)
+        since = utcnow() - datetime.timedelta(minutes=1)
+        for msg in (
+            "Buffer I/O error on dev sda1, logical block 77, lost async "
+            "page write",
+            "nfs: server fileserver01 not responding, still trying",
+        ):
+            err = core.fault_injector.inject(
+                Request(kernel_message=KernelMessage(message=msg, priority=2))
+            )
+            assert err is None
+        deadline = time.time() + 10
+        got_disk = got_nfs = False
+        while time.time() < deadline and not (got_disk and got_nfs):
+            got_disk = any(
+                e.name == "buffer_io_error" for e in disk.events(since))
+            got_nfs = any(
+                e.name == "nfs_server_not_responding"
+                for e in nfs.events(since))
+            time.sleep(0.2)
+        assert got_disk, "disk kmsg event did not reach the disk bucket"
+        assert got_nfs, "nfs kmsg event did not reach the nfs bucket"
+    finally:
+        core.close()
