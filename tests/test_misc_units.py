@@ -566,3 +566,67 @@ def test_machine_info_wire_keys(monkeypatch):
     assert gi["architecture"].startswith("gfx950")
     assert len(gi["gpus"]) == 8  # mock default
     assert d["cudaVersion"]  # carries the ROCm version in the wire slot
+
+
+def test_component_gauges_cache_and_labels():
+    """ComponentGauges: gpud_component label curried on every metric
+    (reference metric naming), child cache stable
+    across repeated sets, no-registry mode is a no-op."""
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from gpud_amd.components.metrics_util import ComponentGauges
+
+    reg = CollectorRegistry()
+    g = ComponentGauges("accelerator-amd-temperature", reg)
+    g.set("accelerator_amd_temperature_current_celsius",
+          "Current temp", 45.0, uuid="gpu-0")
+    g.set("accelerator_amd_temperature_current_celsius",
+          "Current temp", 46.0, uuid="gpu-0")  # same child, updated
+    g.set("accelerator_amd_temperature_current_celsius",
+          "Current temp", 50.0, uuid="gpu-1")
+    g.set("accelerator_amd_temperature_limit_celsius",
+          "Limit", 110.0)  # no extra label
+    text = generate_latest(reg).decode()
+    assert 'gpud_component="accelerator-amd-temperature"' in text
+    assert 'uuid="gpu-0"' in text and 'uuid="gpu-1"' in text
+    assert "46.0" in text and "50.0" in text and "45.0" not in text
+    assert len(g._children) == 3
+    # registry-less mode: silent no-op (components run without metrics)
+    g2 = ComponentGauges("x", None)
+    g2.set("anything", "d", 1.0)
+
+
+def test_recorder_self_telemetry(mem_db):
+    """Recorder exports gpud's own FD count, DB size and sqlite op
+    latencies (reference: pkg/metrics/recorder/gpud_metrics.go:120-160)
+    plus the per-check duration histogram the reference lacks."""
+    from prometheus_client import CollectorRegistry, generate_latest
+
+    from gpud_amd.pkg.metrics.recorder import Recorder
+
+    rw, _ro = mem_db
+    reg = CollectorRegistry()
+    rec = Recorder(reg, db_rw=rw)
+    rec.record_once()
+    rec.observe_check_duration("cpu", 0.012)
+    rec.observe_check_duration("cpu", 0.5)
+    text = generate_latest(reg).decode()
+    assert "gpud_file_descriptor_usage" in text
+    assert "gpud_component_check_duration_seconds" in text
+    assert 'gpud_component="cpu"' in text
+
+
+def test_process_runner_timeout_kills_group(tmp_path):
+    """RunUntilCompletion analog: a hung script is process-group killed
+    at the timeout (reference: pkg/process/runner.go)."""
+    import time
+
+    from gpud_amd.pkg.process_runner import run_bash
+
+    t0 = time.time()
+    res = run_bash("sleep 30 & wait", timeout_seconds=1.5)
+    assert time.time() - t0 < 10
+    assert res.timed_out
+    res2 = run_bash("echo out; echo err >&2; exit 3", timeout_seconds=10)
+    assert res2.exit_code == 3
+    assert "out" in res2.output
