@@ -61,6 +61,11 @@ class ClockSpeedComponent(TickerComponent, SmiComponentMixin):
                 uuid=uuid,
             )
             extra[f"{uuid}.gfx_mhz"] = str(int(gfx))
+            # deep-sleep is informational — an idle GPU legitimately clocks
+            # down (never unhealthy; reference clock-speed is also
+            # info-only)
+            if c.get("gfx_deep_sleep"):
+                extra[f"{uuid}.gfx_deep_sleep"] = "1"
         return CheckResult(
             NAME,
             reason=f"clock speeds collected for {len(snaps)} GPU(s)",
