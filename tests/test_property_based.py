@@ -155,3 +155,69 @@ def test_plugin_spec_loader_never_crashes(tmp_path_factory, doc):
         assert isinstance(specs, list)
     except ValueError:
         pass  # clean rejection is fine; anything else propagates and fails
+
+
+@settings(max_examples=300, deadline=None)
+@given(line=st.text(max_size=400))
+def test_ras_catalog_match_total(line):
+    """The 99-signature catalog must never raise on arbitrary text and
+    must return a catalog Detail when it matches."""
+    from gpud_amd.pkg.ras_catalog import CATALOG, match
+
+    res = match(line)
+    if res is not None:
+        detail, groups = res
+        assert any(d.name == detail.name for d in CATALOG)
+        assert isinstance(groups, dict)
+
+
+@settings(max_examples=150, deadline=None)
+@given(text=st.text(max_size=600))
+def test_component_kmsg_matchers_total(text):
+    """disk/nfs/memory component matchers never raise on arbitrary
+    input."""
+    from gpud_amd.components.host.disk import match_disk_kmsg
+    from gpud_amd.components.host.memory import match_memory_kmsg
+    from gpud_amd.components.host.nfs import match_nfs_kmsg
+
+    for fn in (match_disk_kmsg, match_nfs_kmsg, match_memory_kmsg):
+        res = fn(text)
+        assert res is None or res.name
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    rows=st.lists(
+        st.tuples(
+            st.sampled_from(
+                ["bpf_jit_alloc_exec+0xe/0x20", "vmap", "irq_init", ""]),
+            st.integers(min_value=0, max_value=2**40),
+        ),
+        max_size=30,
+    ),
+    garbage=st.text(max_size=80),
+)
+def test_bpf_jit_sum_total(tmp_path_factory, rows, garbage):
+    """vmallocinfo summing: totals exactly the bpf_jit rows, survives
+    arbitrary garbage lines."""
+    from gpud_amd.components.host.memory import read_bpf_jit_buffer_bytes
+
+    p = tmp_path_factory.mktemp("vmi") / "vmallocinfo"
+    lines = []
+    expect = 0
+    for tag, size in rows:
+        if tag.startswith("bpf_jit"):
+            expect += size
+        lines.append(f"0xdead-0xbeef {size} {tag} pages=1")
+    lines.append(garbage.replace("\n", " "))
+    p.write_text("\n".join(lines) + "\n")
+    assert read_bpf_jit_buffer_bytes(str(p)) == expect
+
+
+@settings(max_examples=120, deadline=None)
+@given(text=st.text(max_size=2000))
+def test_lspci_acs_parser_total(text):
+    from gpud_amd.components.host.pci import parse_acs_bridges
+
+    out = parse_acs_bridges(text)
+    assert isinstance(out, list)
