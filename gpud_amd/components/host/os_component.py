@@ -157,6 +157,22 @@ class BlockedProcessTracker:
             }
 
 
+# FD / PID pressure thresholds (reference: os/component.go:51-54 —
+# defaultMaxAllocatedFileHandlesPctDegraded/Unhealthy, same for PIDs)
+FD_USAGE_PCT_DEGRADED = 90.0
+FD_USAGE_PCT_UNHEALTHY = 95.0
+PID_USAGE_PCT_DEGRADED = 90.0
+PID_USAGE_PCT_UNHEALTHY = 95.0
+
+
+def pid_max() -> int:
+    try:
+        with open("/proc/sys/kernel/pid_max") as f:
+            return int(f.read().strip())
+    except (OSError, ValueError):
+        return 0
+
+
 def file_nr() -> tuple:
     """(allocated, maximum) file handles from /proc/sys/fs/file-nr
     (reference: os component fd usage)."""
@@ -232,6 +248,8 @@ class OSComponent(TickerComponent):
             else list(DEFAULT_DSTATE_NAME_REGEXES)
         )
         self.get_time_now: Callable[[], float] = time.time
+        self.get_file_nr: Callable = file_nr
+        self.get_pid_max: Callable = pid_max
 
     def _dstate_name_matches(self, name: str) -> bool:
         return any(re.search(rx, name) for rx in self.dstate_name_regexes)
@@ -272,8 +290,10 @@ class OSComponent(TickerComponent):
         self._gauges.set(
             "os_uptime_seconds", "Seconds since boot", pkghost.uptime_seconds()
         )
-        fd_alloc, fd_max = file_nr()
+        fd_alloc, fd_max = self.get_file_nr()
+        fd_pct = 0.0
         if fd_max > 0:
+            fd_pct = 100.0 * fd_alloc / fd_max
             self._gauges.set(
                 "os_file_handles_allocated", "System-wide allocated file handles",
                 fd_alloc,
@@ -281,7 +301,16 @@ class OSComponent(TickerComponent):
             self._gauges.set(
                 "os_file_handles_usage_percent",
                 "Allocated file handles as percent of the system maximum",
-                100.0 * fd_alloc / fd_max,
+                fd_pct,
+            )
+        pmax = self.get_pid_max()
+        pid_pct = 0.0
+        if pmax > 0:
+            pid_pct = 100.0 * states["total"] / pmax
+            self._gauges.set(
+                "os_running_pids_usage_percent",
+                "Running PIDs as percent of kernel.pid_max",
+                pid_pct,
             )
         # pstore kernel-panic scan (new findings become Fatal events)
         panic_findings = []
@@ -385,6 +414,28 @@ class OSComponent(TickerComponent):
                 health=HealthStateType.DEGRADED,
                 reason=f"{len(persistent)} persistent D-state process(es): "
                 + ", ".join(f"{b.name}[{b.pid}]" for b in persistent),
+                extra_info=extra,
+            )
+        if fd_pct >= FD_USAGE_PCT_UNHEALTHY or pid_pct >= PID_USAGE_PCT_UNHEALTHY:
+            what = (f"file handles {fd_pct:.1f}% of file-max"
+                    if fd_pct >= FD_USAGE_PCT_UNHEALTHY
+                    else f"PIDs {pid_pct:.1f}% of pid_max")
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason=f"system resource exhaustion imminent: {what}",
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="find and stop the leaking workload",
+                    repair_actions=[RepairActionType.CHECK_USER_APP_AND_GPU],
+                ),
+            )
+        if fd_pct >= FD_USAGE_PCT_DEGRADED or pid_pct >= PID_USAGE_PCT_DEGRADED:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.DEGRADED,
+                reason=f"system resource pressure: file handles "
+                f"{fd_pct:.1f}% / PIDs {pid_pct:.1f}% of their limits",
                 extra_info=extra,
             )
         if states["zombies"] >= self.zombie_unhealthy:

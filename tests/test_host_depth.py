@@ -567,3 +567,26 @@ def test_nfs_hang_evaluator(mock_core):
         # use one tmp group config so the group check passes
     # direct evaluator coverage above is the contract; the component path
     # is covered by test_disk_kmsg_events_reach_component_bucket's pattern
+
+
+def test_os_fd_and_pid_pressure_thresholds(mock_core):
+    """System FD/PID usage vs file-max/pid_max thresholds (reference:
+    os/component.go defaultMaxAllocatedFileHandlesPct* rules)."""
+    comp = mock_core.registry.get("os")
+    comp.get_process_states = lambda: {
+        "total": 100, "zombies": 0, "dstate": 0, "blocked": []}
+    comp.get_file_nr = lambda: (960_000, 1_000_000)  # 96% -> Unhealthy
+    comp.get_pid_max = lambda: 4_000_000
+    cr = comp.trigger_check()
+    assert cr.health == "Unhealthy" and "file handles" in cr.reason
+    comp.get_file_nr = lambda: (910_000, 1_000_000)  # 91% -> Degraded
+    cr = comp.trigger_check()
+    assert cr.health == "Degraded" and "pressure" in cr.reason
+    comp.get_file_nr = lambda: (10_000, 1_000_000)
+    comp.get_process_states = lambda: {
+        "total": 3_900_000, "zombies": 0, "dstate": 0, "blocked": []}
+    cr = comp.trigger_check()  # 97.5% of pid_max
+    assert cr.health == "Unhealthy" and "PIDs" in cr.reason
+    comp.get_process_states = lambda: {
+        "total": 100, "zombies": 0, "dstate": 0, "blocked": []}
+    assert comp.trigger_check().health == "Healthy"
