@@ -9,6 +9,8 @@ compare with the amdsmi enumeration and the configured expected count.
 
 from __future__ import annotations
 
+import os
+
 import subprocess
 from typing import Callable, Optional
 
@@ -41,6 +43,23 @@ def count_amd_gpus_lspci(lspci_command: str = "") -> Optional[int]:
     return n
 
 
+def count_render_nodes(dri_dir: str = "/dev/dri") -> Optional[int]:
+    """amdgpu render nodes (/dev/dri/renderD*) — the DCGM-style /dev
+    triangulation source (reference: pkg/nvidia/dev/device_count.go
+    counting /dev/nvidia[0-9]+). On partitioned (DPX/QPX/CPX) boards one
+    physical GPU exposes several render nodes, so this is compared
+    against the amdsmi enumeration (which is also per-partition), not
+    the PCI count."""
+    try:
+        import re as _re
+
+        return sum(
+            1 for n in os.listdir(dri_dir) if _re.fullmatch(r"renderD\d+", n)
+        )
+    except OSError:
+        return None
+
+
 class GPUCountsComponent(TickerComponent, SmiComponentMixin):
     # The PCI topology changes only on hotplug/driver rebind, and a GPU
     # dropping off the driver is caught every cycle by the amdsmi
@@ -57,6 +76,7 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
         self._lspci_cache: Optional[int] = None
         self._lspci_cached_at = 0.0
         self.count_lspci: Callable[[], Optional[int]] = self._cached_lspci
+        self.count_render: Callable[[], Optional[int]] = count_render_nodes
 
     def _cached_lspci(self) -> Optional[int]:
         import time
@@ -102,9 +122,21 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
                 "AMD accelerators on the PCI bus",
                 pci_count,
             )
+        render_count = None
+        from ... import smi as smi_pkg
+
+        if not smi_pkg.mock_enabled():
+            render_count = self.count_render()
+            if render_count is not None:
+                self._gauges.set(
+                    "accelerator_amd_gpu_counts_render_nodes",
+                    "amdgpu render nodes under /dev/dri",
+                    render_count,
+                )
         extra = {
             "smi_count": str(smi_count),
             "lspci_count": "" if pci_count is None else str(pci_count),
+            "render_nodes": "" if render_count is None else str(render_count),
             "expected": str(self.expected),
         }
         if smi_err:
@@ -131,6 +163,21 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
                         RepairActionType.REBOOT_SYSTEM,
                         RepairActionType.HARDWARE_INSPECTION,
                     ],
+                ),
+            )
+        if render_count is not None and smi_count and render_count < smi_count:
+            return CheckResult(
+                NAME,
+                health=HealthStateType.UNHEALTHY,
+                reason=(
+                    f"only {render_count} render node(s) under /dev/dri for "
+                    f"{smi_count} amdsmi device(s) — device files missing "
+                    "(container device-cgroup or udev problem)"
+                ),
+                extra_info=extra,
+                suggested_actions=SuggestedActions(
+                    description="GPU device files missing",
+                    repair_actions=[RepairActionType.CHECK_USER_APP_AND_GPU],
                 ),
             )
         if pci_count is not None and smi_count and pci_count < smi_count:

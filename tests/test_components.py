@@ -347,3 +347,22 @@ def test_peer_mem_degraded_without_providers_or_dmabuf(mock_core, monkeypatch):
     monkeypatch.setattr(comp, "get_providers", lambda: ["amdgpu_peerdirect"])
     cr = comp.trigger_check()
     assert cr.health == HealthStateType.HEALTHY
+
+
+def test_gpu_counts_render_node_triangulation(mock_core, monkeypatch):
+    """/dev/dri renderD* is the third count source (reference:
+    pkg/nvidia/dev/device_count.go): fewer render nodes than amdsmi
+    devices means device files are missing."""
+    import gpud_amd.smi as smi_pkg
+
+    comp = mock_core.registry.get("accelerator-amd-gpu-counts")
+    monkeypatch.setattr(smi_pkg, "mock_enabled", lambda: False)
+    comp.count_lspci = lambda: 4
+    comp.count_render = lambda: 2  # 2 nodes for 4 GPUs
+    cr = comp.trigger_check()
+    assert cr.health == HealthStateType.UNHEALTHY
+    assert "render node" in cr.reason
+    comp.count_render = lambda: 4
+    assert comp.trigger_check().health == HealthStateType.HEALTHY
+    comp.count_render = lambda: None  # /dev/dri unreadable: no claim
+    assert comp.trigger_check().health == HealthStateType.HEALTHY
