@@ -119,6 +119,9 @@ def test_diag_gemm_bf16_verified():
     assert res8["verified"], res8
     res9 = _diag.gemm_stress_bf16_v9(size=1024, iters=2)
     assert res9["verified"], res9
+    # the sustained (hipGraph replay) form must verify too
+    resg = _diag.gemm_stress_bf16_v7_graph(size=1024, iters=2)
+    assert resg["verified"], resg
     res3 = _diag.gemm_stress_mxfp8(size=4096, iters=2)
     assert res3["verified"], res3
     assert res3["tflops"] > 1400, res3
