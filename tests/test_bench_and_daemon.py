@@ -470,3 +470,34 @@ def test_bench_four_ranks_gloo_mock():
     assert d["scaling"] == "weak"
     assert d["config"]["gpus_per_rank"] == 1 or d["data"] == "synthetic"
     assert d["value"] > 0
+
+
+def test_mock_8gpu_daemon_shape(monkeypatch):
+    """The deployment shape: one core, 8 GPUs — every per-GPU metric must
+    fan out to 8 uuid series and the poll stays single-process
+    (SURVEY §7 'overhead flat to 8 GPUs')."""
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK_GPUS", "8")
+    from prometheus_client import generate_latest
+
+    from gpud_amd.bootstrap import build_core
+
+    core = build_core(in_memory_db=True, kmsg_writable=False,
+                      record_reboot=False)
+    try:
+        for name in ("accelerator-amd-temperature", "accelerator-amd-power",
+                     "accelerator-amd-utilization", "accelerator-amd-ecc",
+                     "accelerator-amd-memory", "accelerator-amd-xgmi"):
+            comp = core.registry.get(name)
+            cr = comp.trigger_check()
+            assert cr.health in ("Healthy", "Degraded"), (name, cr.reason)
+        text = generate_latest(core.metrics_registry).decode()
+        import re
+
+        for metric in ("accelerator_amd_temperature_hotspot_celsius",
+                       "accelerator_amd_power_usage_watts"):
+            uuids = set(re.findall(
+                rf'{metric}{{[^}}]*uuid="([^"]+)"', text))
+            assert len(uuids) == 8, (metric, len(uuids))
+    finally:
+        core.close()
