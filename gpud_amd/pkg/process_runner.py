@@ -12,6 +12,7 @@ import os
 import signal
 import subprocess
 import tempfile
+import time
 import threading
 from dataclasses import dataclass
 from typing import Optional
@@ -97,6 +98,66 @@ def run_bash(
             error="script timed out" if timed_out else "",
         )
     finally:
+        try:
+            os.unlink(path)
+        except OSError:
+            pass
+
+
+def stream_bash(
+    bash_script: str,
+    timeout_seconds: float = 600.0,
+    env: Optional[dict] = None,
+):
+    """Yield combined-output lines from a bash script as they appear
+    (reference: pkg/process Process.StdoutReader streaming — used for
+    long-running package installs). The process group is killed when the
+    deadline passes or the consumer abandons the generator."""
+    with tempfile.NamedTemporaryFile(
+        "w", suffix=".sh", prefix="gpud-", delete=False
+    ) as f:
+        f.write(bash_script)
+        path = f.name
+    merged_env = dict(os.environ)
+    if env:
+        merged_env.update(env)
+    proc = subprocess.Popen(
+        ["bash", path],
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        start_new_session=True,
+        env=merged_env,
+        text=True,
+    )
+    deadline = time.monotonic() + timeout_seconds
+
+    def _kill():
+        try:
+            os.killpg(proc.pid, signal.SIGKILL)
+        except ProcessLookupError:
+            pass
+
+    try:
+        assert proc.stdout is not None
+        for line in proc.stdout:
+            yield line.rstrip("\n")
+            if time.monotonic() > deadline:
+                _kill()
+                raise TimeoutError("script timed out")
+        proc.wait(timeout=max(0.1, deadline - time.monotonic()))
+    except GeneratorExit:
+        _kill()
+        raise
+    except subprocess.TimeoutExpired:
+        _kill()
+        raise TimeoutError("script timed out") from None
+    finally:
+        try:
+            proc.stdout and proc.stdout.close()
+        except OSError:
+            pass
+        if proc.poll() is None:
+            _kill()
         try:
             os.unlink(path)
         except OSError:

@@ -630,3 +630,19 @@ def test_process_runner_timeout_kills_group(tmp_path):
     res2 = run_bash("echo out; echo err >&2; exit 3", timeout_seconds=10)
     assert res2.exit_code == 3
     assert "out" in res2.output
+
+
+def test_stream_bash_lines_and_abandon():
+    """stream_bash yields lines live; abandoning the generator kills the
+    process group (reference: pkg/process streaming reader)."""
+    from gpud_amd.pkg.process_runner import stream_bash
+
+    lines = list(stream_bash("echo a; echo b; echo c", timeout_seconds=10))
+    assert lines == ["a", "b", "c"]
+
+    # abandon mid-stream: the GeneratorExit path kills the process group
+    gen = stream_bash("echo first; sleep 60; echo never", timeout_seconds=120)
+    assert next(gen) == "first"
+    gen.close()
+    # the runner stays usable after an abandoned stream
+    assert list(stream_bash("echo z", timeout_seconds=5)) == ["z"]
