@@ -45,7 +45,7 @@ from typing import Dict, List, Optional
 from ..apiv1.types import PackagePhase, PackageStatus
 from .config import Config
 from .log import logger
-from .process_runner import run_bash
+from .process_runner import RunResult, run_bash, stream_bash
 
 SYNC_PERIOD_SECONDS = 3.0  # reference: package_controller.go syncPeriod
 
@@ -172,10 +172,39 @@ def discover_packages(packages_dir: str) -> Dict[str, str]:
 
 
 def _run_pkg(init_sh: str, subcommand: str, timeout: float = 300.0,
-             log_output: bool = True):
+             log_output: bool = True, stream: bool = False):
     """Run one init.sh subcommand; mirror the reference's per-subcommand
     log file (<pkgdir>/<arg>.log) so operators can inspect what each
-    lifecycle step printed."""
+    lifecycle step printed. ``stream=True`` (long installs/upgrades)
+    writes the log LIVE line-by-line like the reference's streaming
+    process reader, so a wedged install is inspectable mid-run."""
+    if stream and log_output:
+        log_path = os.path.join(os.path.dirname(init_sh), f"{subcommand}.log")
+        lines = []
+        timed_out = False
+        holder = {}
+        try:
+            with open(log_path, "w") as f:
+                try:
+                    for line in stream_bash(
+                        f'bash "{init_sh}" {subcommand}',
+                        timeout_seconds=timeout,
+                        result_holder=holder,
+                    ):
+                        lines.append(line)
+                        f.write(line + "\n")
+                        f.flush()
+                except TimeoutError:
+                    timed_out = True
+        except OSError:
+            # log dir unwritable: fall back to the buffered path
+            return _run_pkg(init_sh, subcommand, timeout, log_output=False)
+        return RunResult(
+            exit_code=-1 if timed_out else holder.get("exit_code", -1),
+            output="\n".join(lines),
+            timed_out=timed_out,
+            error="script timed out" if timed_out else "",
+        )
     res = run_bash(f'bash "{init_sh}" {subcommand}', timeout_seconds=timeout)
     if log_output:
         try:
@@ -307,7 +336,7 @@ class PackageController:
         return True
 
     def _do_install(self, pkg: ManagedPackage) -> None:
-        res = _run_pkg(pkg.script_path, "install")
+        res = _run_pkg(pkg.script_path, "install", stream=True)
         if res.exit_code == 0:
             start = _run_pkg(pkg.script_path, "start")
             if start.exit_code != 0:
@@ -353,7 +382,7 @@ class PackageController:
                 ).start()
 
     def _do_upgrade(self, pkg: ManagedPackage) -> None:
-        res = _run_pkg(pkg.script_path, "upgrade")
+        res = _run_pkg(pkg.script_path, "upgrade", stream=True)
         if res.exit_code != 0:
             logger.error("package %s upgrade failed (%d)", pkg.name,
                          res.exit_code)

@@ -108,11 +108,14 @@ def stream_bash(
     bash_script: str,
     timeout_seconds: float = 600.0,
     env: Optional[dict] = None,
+    result_holder: Optional[dict] = None,
 ):
     """Yield combined-output lines from a bash script as they appear
     (reference: pkg/process Process.StdoutReader streaming — used for
     long-running package installs). The process group is killed when the
-    deadline passes or the consumer abandons the generator."""
+    deadline passes or the consumer abandons the generator.
+    ``result_holder`` (a dict) receives ``exit_code`` once the process
+    finishes — generators cannot hand a return value to a for-loop."""
     with tempfile.NamedTemporaryFile(
         "w", suffix=".sh", prefix="gpud-", delete=False
     ) as f:
@@ -145,6 +148,8 @@ def stream_bash(
                 _kill()
                 raise TimeoutError("script timed out")
         proc.wait(timeout=max(0.1, deadline - time.monotonic()))
+        if result_holder is not None:
+            result_holder["exit_code"] = proc.returncode
     except GeneratorExit:
         _kill()
         raise

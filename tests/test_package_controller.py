@@ -242,3 +242,27 @@ def test_admin_statuses_shape(tmp_path):
     assert st["total_time"] == 120 * 10**9  # Go duration ns
     assert st["target_version"] == "3.0"
     assert st["current_version"] == "3.0"
+
+
+def test_install_streams_log_and_exit_code(tmp_path):
+    """install/upgrade stream their per-subcommand log live and surface
+    the script's real exit code (reference: process streaming in the
+    package controllers)."""
+    from gpud_amd.pkg.gpud_manager import _run_pkg
+
+    pkg = tmp_path / "pkgA"
+    pkg.mkdir()
+    init = pkg / "init.sh"
+    init.write_text(
+        "#!/bin/bash\n"
+        'case "$1" in\n'
+        "  install) echo step1; echo step2; exit 0 ;;\n"
+        "  upgrade) echo bad; exit 9 ;;\n"
+        "esac\n"
+    )
+    res = _run_pkg(str(init), "install", stream=True)
+    assert res.exit_code == 0
+    assert (pkg / "install.log").read_text() == "step1\nstep2\n"
+    res = _run_pkg(str(init), "upgrade", stream=True)
+    assert res.exit_code == 9
+    assert "bad" in (pkg / "upgrade.log").read_text()
