@@ -72,6 +72,7 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
         self._smi = inst.smi
         self._gauges = ComponentGauges(NAME, inst.metrics_registry)
         self.expected = inst.expected_gpu_count
+        self._cfg = inst.config
         self._lspci_command = inst.lspci_command
         self._lspci_cache: Optional[int] = None
         self._lspci_cached_at = 0.0
@@ -103,6 +104,13 @@ class GPUCountsComponent(TickerComponent, SmiComponentMixin):
         return True  # meaningful even when SMI is missing (counts mismatch!)
 
     def check(self) -> CheckResult:
+        # pick up control-plane updateConfig changes live (the shared
+        # Config object is mutated in place)
+        cfg_expected = (
+            getattr(self._cfg, "expected_gpu_count", 0) if self._cfg else 0
+        )
+        if cfg_expected > 0:
+            self.expected = cfg_expected
         smi_count = 0
         smi_err = ""
         if self._smi is not None and self._smi.exists:

@@ -73,9 +73,10 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
         self._bucket = (
             inst.event_store.bucket(NAME) if inst.event_store is not None else None
         )
-        cfg = inst.config
+        self._cfg = inst.config
         self.expected_links = (
-            getattr(cfg, "expected_xgmi_link_count", 0) if cfg else 0
+            getattr(self._cfg, "expected_xgmi_link_count", 0)
+            if self._cfg else 0
         )
         self._last_states: Dict[str, List[int]] = {}
         self._last_traffic: Dict = {}
@@ -141,7 +142,12 @@ class XGMIComponent(TickerComponent, SmiComponentMixin):
         missing_links: List[str] = []
         extra: Dict[str, str] = {}
         any_links = False
-        expected = self.expected_links
+        # control-plane updateConfig mutates the shared Config live
+        # (reference: SetDefault* setters re-read per check)
+        expected = (
+            getattr(self._cfg, "expected_xgmi_link_count", 0)
+            if self._cfg else 0
+        ) or self.expected_links
         if expected <= 0:
             expected = expected_links_for_product(
                 getattr(self._smi, "product_name", "") or "", len(snaps)

@@ -740,3 +740,20 @@ def test_update_pinned_root_key_enforced(tmp_path):
         install_dir=str(tmp_path / "i3"),
     )
     assert err is not None and "FAILED" in err
+
+
+def test_update_config_expected_counts_apply_live(mock_core):
+    """updateConfig's expected_gpu_count / expected_xgmi_link_count reach
+    the running components without a restart (reference: SetDefault*
+    setters re-read per check)."""
+    cfg = mock_core.config
+    comp = mock_core.registry.get("accelerator-amd-gpu-counts")
+    base = comp.trigger_check()
+    assert base.health == "Healthy"
+    cfg.expected_gpu_count = 99  # what updateConfig's setattr does
+    cr = comp.trigger_check()
+    assert cr.health == "Unhealthy"
+    assert "expected 99" in cr.reason
+    cfg.expected_gpu_count = 0
+    comp.expected = 0
+    assert comp.trigger_check().health == "Healthy"
