@@ -489,6 +489,18 @@ class Session:
                 applied.append(key)
             else:
                 rejected.append(key)
+        # poll interval reaches running tickers immediately — the loop
+        # re-reads component.poll_interval every cycle (reference:
+        # updateConfig pushes SetDefault* setters live)
+        if "poll_interval_seconds" in applied and self.core.registry:
+            try:
+                iv = float(cfg.poll_interval_seconds)
+                if iv > 0:
+                    for comp in self.core.registry.all_components():
+                        if hasattr(comp, "poll_interval"):
+                            comp.poll_interval = iv
+            except (TypeError, ValueError):
+                pass
         out = {"applied": applied}
         if rejected:
             out["rejected"] = rejected

@@ -757,3 +757,13 @@ def test_update_config_expected_counts_apply_live(mock_core):
     cfg.expected_gpu_count = 0
     comp.expected = 0
     assert comp.trigger_check().health == "Healthy"
+
+
+def test_update_config_poll_interval_reaches_tickers(mock_core):
+    sess = _session(mock_core)
+    comp = mock_core.registry.get("cpu")
+    before = comp.poll_interval
+    out = sess._m_updateConfig({"poll_interval_seconds": 7})
+    assert "poll_interval_seconds" in out["applied"]
+    assert comp.poll_interval == 7.0
+    sess._m_updateConfig({"poll_interval_seconds": before})
