@@ -492,11 +492,12 @@ class Session:
         # poll interval reaches running tickers immediately — the loop
         # re-reads component.poll_interval every cycle (reference:
         # updateConfig pushes SetDefault* setters live)
-        if "poll_interval_seconds" in applied and self.core.registry:
+        registry = getattr(self.core, "registry", None)
+        if "poll_interval_seconds" in applied and registry is not None:
             try:
                 iv = float(cfg.poll_interval_seconds)
                 if iv > 0:
-                    for comp in self.core.registry.all_components():
+                    for comp in registry.all_components():
                         if hasattr(comp, "poll_interval"):
                             comp.poll_interval = iv
             except (TypeError, ValueError):
