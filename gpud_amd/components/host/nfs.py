@@ -11,10 +11,9 @@ failure, writeback-path hang stack frames).
 from __future__ import annotations
 
 import os
+import re
 from dataclasses import dataclass
 from typing import List
-
-import re
 
 from ...apiv1.types import HealthStateType
 from ...pkg import host as pkghost
