@@ -1284,7 +1284,7 @@ PYBIND11_MODULE(_amdsmi, m) {
     try {
       if (p) std::rethrow_exception(p);
     } catch (const SmiError& e) {
-      exc(e.what());
+      py::set_error(exc, e.what());
     }
   });
   m.def("init", &smi_init, "Initialize amdsmi and enumerate GPUs");
