@@ -221,3 +221,46 @@ def test_disk_and_nfs_kmsg_through_file_seam(tmp_path, monkeypatch):
         assert got_nfs, "nfs kmsg event did not reach the nfs bucket"
     finally:
         core.close()
+
+
+def test_full_daemon_restart_state_survival(tmp_path, monkeypatch):
+    """Checkpoint/resume end-to-end: a daemon core with a FILE-backed
+    state DB accumulates RAS events, link history, metadata and a
+    set-healthy tombstone; a freshly-built core on the same data dir
+    sees all of it (reference: single gpud.state SQLite — SURVEY §5
+    checkpoint/resume row)."""
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    from gpud_amd.apiv1.types import Event, utcnow
+    from gpud_amd.bootstrap import build_core
+    from gpud_amd.pkg import metadata
+    from gpud_amd.pkg.config import Config
+
+    data = tmp_path / "data"
+    cfg = Config(data_dir=str(data))
+    now = utcnow()
+
+    core1 = build_core(cfg, kmsg_writable=False, record_reboot=False)
+    try:
+        bucket = core1.event_store.bucket("accelerator-amd-error-ras",
+                                          disable_purge=True)
+        bucket.insert(Event(time=now, name="amdgpu_ring_timeout",
+                            type="Critical", message="ring gfx timeout"))
+        metadata.set_value(core1.db_rw, "machine_id", "m-123")
+        xgmi = core1.registry.get("accelerator-amd-xgmi")
+        if getattr(xgmi, "_store", None) is not None:
+            xgmi._store.set_tombstone(now.timestamp())
+    finally:
+        core1.close()
+
+    core2 = build_core(cfg, kmsg_writable=False, record_reboot=False)
+    try:
+        bucket = core2.event_store.bucket("accelerator-amd-error-ras",
+                                          disable_purge=True)
+        evs = bucket.get(now - datetime.timedelta(minutes=5))
+        assert any(e.name == "amdgpu_ring_timeout" for e in evs)
+        assert metadata.get_value(core2.db_ro, "machine_id") == "m-123"
+        xgmi = core2.registry.get("accelerator-amd-xgmi")
+        if getattr(xgmi, "_store", None) is not None:
+            assert xgmi._store.get_tombstone() >= int(now.timestamp())
+    finally:
+        core2.close()
