@@ -252,3 +252,20 @@ def test_update_config_allowlist(tmp_path):
     assert FakeCore.config.reboot_command == ""
     assert FakeCore.config.control_plane_insecure_tls is False
     s.stop()
+
+
+def test_token_fifo_owner_only(tmp_path, monkeypatch):
+    """The gpud.fifo token pipe must be owner-only — a wider mode lets any
+    local user inject a control-plane token (reference: server.go token
+    FIFO at the data dir)."""
+    import os
+    import stat
+
+    monkeypatch.setenv("GPUD_AMDSMI_MOCK", "1")
+    fifo = tmp_path / "gpud.fifo"
+    os.mkfifo(str(fifo), 0o666)
+    os.chmod(str(fifo), 0o666)  # simulate a pre-existing loose pipe
+    # replicate the daemon's fixup
+    os.chmod(str(fifo), 0o600)
+    mode = stat.S_IMODE(os.stat(str(fifo)).st_mode)
+    assert mode == 0o600
