@@ -80,6 +80,13 @@ def open_rw(path: str) -> Conn:
         os.makedirs(os.path.dirname(os.path.abspath(path)) or ".", exist_ok=True)
     conn = sqlite3.connect(path, check_same_thread=False, timeout=5.0)
     _apply_pragmas(conn, readonly=False)
+    if path != ":memory:":
+        # the state DB carries the control-plane token and machine
+        # credentials (metadata table) — owner-only regardless of umask
+        try:
+            os.chmod(path, 0o600)
+        except OSError:
+            pass
     return Conn(conn, readonly=False)
 
 

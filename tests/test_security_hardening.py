@@ -269,3 +269,19 @@ def test_token_fifo_owner_only(tmp_path, monkeypatch):
     os.chmod(str(fifo), 0o600)
     mode = stat.S_IMODE(os.stat(str(fifo)).st_mode)
     assert mode == 0o600
+
+
+def test_state_db_owner_only(tmp_path):
+    """gpud.state carries the CP token + machine credentials — the file
+    must be created owner-only."""
+    import os
+    import stat
+
+    from gpud_amd.pkg.sqlite_util import open_rw
+
+    path = tmp_path / "gpud.state"
+    conn = open_rw(str(path))
+    try:
+        assert stat.S_IMODE(os.stat(str(path)).st_mode) == 0o600
+    finally:
+        conn.close()
