@@ -40,6 +40,10 @@ def generate_self_signed_cert(
             check=True,
             capture_output=True,
         )
+    try:
+        os.chmod(key, 0o600)  # private key owner-only regardless of umask
+    except OSError:
+        pass
     return cert, key
 
 
