@@ -590,3 +590,32 @@ def test_os_fd_and_pid_pressure_thresholds(mock_core):
     comp.get_process_states = lambda: {
         "total": 100, "zombies": 0, "dstate": 0, "blocked": []}
     assert comp.trigger_check().health == "Healthy"
+
+
+def test_lsblk_deep_nesting_and_missing_fields():
+    """Reference-style edge fixtures: 3-level nesting (disk > part > LVM)
+    flattens depth-first with parent attribution; rows missing optional
+    keys survive (pkg/disk/lsblk_flatten.go)."""
+    import json
+
+    from gpud_amd.components.host.disk import _flatten_devices
+
+    tree = json.loads("""
+    {"blockdevices": [
+      {"name": "sda", "type": "disk", "children": [
+        {"name": "sda1", "type": "part", "children": [
+          {"name": "vg0-root", "type": "lvm", "mountpoint": "/"},
+          {"name": "vg0-swap", "type": "lvm"}
+        ]},
+        {"name": "sda2", "type": "part", "fstype": "xfs"}
+      ]},
+      {"name": "zram0", "type": "disk"}
+    ]}""")["blockdevices"]
+    flat = _flatten_devices(tree)
+    names = [d["name"] for d in flat]
+    assert names == ["sda", "sda1", "vg0-root", "vg0-swap", "sda2", "zram0"]
+    by = {d["name"]: d for d in flat}
+    assert by["vg0-root"]["pkname"] == "sda1"
+    assert by["sda2"]["pkname"] == "sda"
+    assert by["zram0"].get("pkname", "") == ""
+    assert "children" not in by["sda"]
